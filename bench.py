@@ -1,0 +1,157 @@
+"""Flagship benchmark: ResNet-18 CIFAR-10 DDP training throughput (images/sec,
+whole node) on MI355X — the BASELINE.json north-star metric.
+
+Contract (driver): `python bench.py --gpus N --steps K --warmup W`; for N>1
+the driver launches it under torch.distributed.run with one rank per GPU
+(RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* in env). W untimed warmup steps, then
+exactly K timed steps bracketed by barrier + torch.cuda.synchronize on both
+sides; elapsed is MAX over ranks; rank 0 prints ONE JSON line.
+
+Synthetic CIFAR-shaped data (no network for the real set), random-init
+weights, bf16 compute via the mi355x HIP kernels, full training step
+(zero_grad, fwd, loss, bwd, bucket all-reduce overlap, fused SGD step).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from mi355x import optim
+from mi355x.models import build_model
+from mi355x.ops import cross_entropy
+from mi355x.parallel import DistributedDataParallel, comm
+from mi355x.parallel.flat import FlatState
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=30)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--batch", type=int,
+                    default=int(os.environ.get("MI355X_BENCH_BATCH", "256")),
+                    help="per-GPU batch size")
+    ap.add_argument("--model", default=os.environ.get("MI355X_BENCH_MODEL",
+                                                      "resnet18"))
+    ap.add_argument("--size", type=int,
+                    default=int(os.environ.get("MI355X_BENCH_SIZE", "32")),
+                    help="image side (32 CIFAR, 224 config 4)")
+    ap.add_argument("--classes", type=int, default=None)
+    ap.add_argument("--sync-bn", action="store_true",
+                    default=os.environ.get("MI355X_SYNC_BN", "0") == "1")
+    ap.add_argument("--fp16", action="store_true",
+                    default=os.environ.get("MI355X_FP16", "0") == "1")
+    return ap.parse_args()
+
+
+def main():
+    args = parse_args()
+    use_cuda = torch.cuda.is_available()
+    world = comm.env_world_size()
+    distributed = world > 1
+    if distributed:
+        rank, world, local = comm.init_process_group()
+    else:
+        rank, local = 0, 0
+    device = torch.device("cuda", local) if use_cuda else torch.device("cpu")
+    if args.fp16:
+        from mi355x import amp
+        amp.set_compute_dtype(torch.float16)
+
+    num_classes = args.classes or (1000 if args.size >= 224 else 10)
+    model_name = args.model if args.size < 224 or args.model != "resnet18" \
+        else "resnet18_imagenet"
+    torch.manual_seed(1234)
+    net = build_model(model_name, num_classes=num_classes).to(device)
+    if args.sync_bn and distributed:
+        from mi355x.parallel import sync_bn
+        sync_bn.enable(net)
+    if distributed:
+        net = DistributedDataParallel(net)
+        flat = net.flat
+        gscale = net.grad_scale
+    else:
+        flat = FlatState(net)
+        gscale = 1.0
+    optimizer = optim.SGD(flat, lr=0.1, momentum=0.9, grad_scale=gscale)
+
+    # synthetic on-device data pool (new batch each step, cycling)
+    g = torch.Generator(device="cpu").manual_seed(4321 + rank)
+    n_pool = 8
+    pool_x = [torch.randn(args.batch, 3, args.size, args.size, generator=g)
+              .to(device) for _ in range(n_pool)]
+    pool_y = [torch.randint(0, num_classes, (args.batch,), generator=g)
+              .to(device) for _ in range(n_pool)]
+
+    def step(i):
+        x, y = pool_x[i % n_pool], pool_y[i % n_pool]
+        optimizer.zero_grad()
+        out = net(x)
+        loss = cross_entropy(out, y)
+        loss.backward()
+        if distributed:
+            net.finish_grad_sync()
+        optimizer.step()
+        return loss
+
+    for i in range(args.warmup):
+        step(i)
+    if distributed:
+        comm.barrier()
+    if use_cuda:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        loss = step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if distributed:
+        comm.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if (distributed and use_cuda) else "cpu")
+    if distributed:
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    n_gpus = world if use_cuda else 0
+    global_batch = args.batch * world
+    images_per_sec = global_batch * args.steps / elapsed
+    if rank == 0:
+        print(json.dumps({
+            "metric": "images/sec (whole node), ResNet-18 CIFAR-10",
+            "value": round(images_per_sec, 1),
+            "unit": "images/sec",
+            "n_gpus": n_gpus if use_cuda else world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "fp16" if args.fp16 else "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": global_batch,
+                "image_size": args.size,
+                "num_classes": num_classes,
+                "parallelism": f"dp{world}",
+                "sync_bn": bool(args.sync_bn),
+                "loss_final": round(float(loss.item()), 4),
+            },
+        }))
+    if distributed:
+        comm.destroy()
+
+
+if __name__ == "__main__":
+    main()

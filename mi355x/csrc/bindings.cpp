@@ -1,0 +1,62 @@
+// pybind bindings for the mi355x gfx950 kernel library.
+#include <torch/extension.h>
+
+#include <vector>
+
+at::Tensor relu_bwd(at::Tensor dy, at::Tensor y);
+void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr, double mu,
+              double wd, double gscale);
+at::Tensor cast_to_16(at::Tensor src, at::Tensor like);
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                      long stride, long pad, long act);
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, long stride, long pad,
+                        long H, long W);
+at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
+                        long stride, long pad);
+
+at::Tensor bn_stats(at::Tensor x);
+at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
+                    at::Tensor gamma, at::Tensor beta, at::Tensor res,
+                    long act);
+at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor mean,
+                         at::Tensor invstd);
+at::Tensor bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor mean,
+                     at::Tensor invstd, at::Tensor gamma, at::Tensor dgamma,
+                     at::Tensor dbeta, double m_total);
+
+std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
+                                    long pad);
+at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
+                       long kernel, long stride, long pad);
+at::Tensor global_avg_pool(at::Tensor x);
+
+at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act);
+at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w);
+at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy);
+
+std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
+                                          at::Tensor target);
+at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
+                             at::Tensor lse, double scale);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("relu_bwd", &relu_bwd);
+  m.def("sgd_step", &sgd_step);
+  m.def("cast_to_16", &cast_to_16);
+  m.def("conv2d_fwd", &conv2d_fwd);
+  m.def("conv2d_dgrad", &conv2d_dgrad);
+  m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("bn_stats", &bn_stats);
+  m.def("bn_apply", &bn_apply);
+  m.def("bn_bwd_reduce", &bn_bwd_reduce);
+  m.def("bn_bwd_dx", &bn_bwd_dx);
+  m.def("maxpool_fwd", &maxpool_fwd);
+  m.def("maxpool_bwd", &maxpool_bwd);
+  m.def("global_avg_pool", &global_avg_pool);
+  m.def("linear_fwd", &linear_fwd);
+  m.def("linear_dgrad", &linear_dgrad);
+  m.def("linear_wgrad", &linear_wgrad);
+  m.def("cross_entropy_fwd", &cross_entropy_fwd);
+  m.def("cross_entropy_bwd", &cross_entropy_bwd);
+}

@@ -1,0 +1,213 @@
+// NHWC conv2d: forward, dgrad, wgrad (SURVEY.md N5 — the reference's
+// cuDNN/MIOpen implicit-GEMM conv, rebuilt for CDNA4).
+//
+// v0: direct kernels, correctness-first, laid out so the C (channel-minor)
+// dimension is the coalesced/vectorized axis everywhere. The MFMA
+// implicit-GEMM path (conv_mfma.hip) replaces these for the hot shapes;
+// these remain the fallback for odd geometries and the numerics reference
+// on-device.
+#include "common.h"
+
+namespace {
+
+template <typename T16>
+__global__ void conv_fwd_direct(const T16* __restrict__ x,
+                                const T16* __restrict__ w,
+                                const float* __restrict__ bias,
+                                T16* __restrict__ y, int N, int H, int W,
+                                int C, int K, int R, int S, int P, int Q,
+                                int stride, int pad, int act, int has_bias) {
+  const long total = (long)N * P * Q * K;
+  const bool cvec = (C % 8 == 0);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(t % K);
+    long npq = t / K;
+    const int q = (int)(npq % Q);
+    long np = npq / Q;
+    const int p = (int)(np % P);
+    const int n = (int)(np / P);
+    float acc = has_bias ? bias[k] : 0.f;
+    const T16* wk = w + (long)k * R * S * C;
+    for (int r = 0; r < R; ++r) {
+      const int ih = p * stride - pad + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        const int iw = q * stride - pad + s;
+        if (iw < 0 || iw >= W) continue;
+        const T16* xp = x + (((long)n * H + ih) * W + iw) * C;
+        const T16* wp = wk + (long)(r * S + s) * C;
+        if (cvec) {
+          for (int c = 0; c < C; c += 8) {
+            short8 xv = *reinterpret_cast<const short8*>(xp + c);
+            short8 wv = *reinterpret_cast<const short8*>(wp + c);
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+              acc += s16_to_f32<T16>(xv[u]) * s16_to_f32<T16>(wv[u]);
+          }
+        } else {
+          for (int c = 0; c < C; ++c)
+            acc += F16<T16>::to_f32(xp[c]) * F16<T16>::to_f32(wp[c]);
+        }
+      }
+    }
+    if (act == 1) acc = fmaxf(acc, 0.f);
+    y[t] = F16<T16>::from_f32(acc);
+  }
+}
+
+template <typename T16>
+__global__ void conv_dgrad_direct(const T16* __restrict__ dy,
+                                  const T16* __restrict__ w,
+                                  T16* __restrict__ dx, int N, int H, int W,
+                                  int C, int K, int R, int S, int P, int Q,
+                                  int stride, int pad) {
+  const long total = (long)N * H * W * C;
+  const long wkstride = (long)R * S * C;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int c = (int)(t % C);
+    long nhw = t / C;
+    const int iw = (int)(nhw % W);
+    long nh = nhw / W;
+    const int ih = (int)(nh % H);
+    const int n = (int)(nh / H);
+    float acc = 0.f;
+    for (int r = 0; r < R; ++r) {
+      const int ph = ih + pad - r;
+      if (ph < 0 || ph % stride) continue;
+      const int p = ph / stride;
+      if (p >= P) continue;
+      for (int s = 0; s < S; ++s) {
+        const int qw = iw + pad - s;
+        if (qw < 0 || qw % stride) continue;
+        const int q = qw / stride;
+        if (q >= Q) continue;
+        const T16* dyp = dy + (((long)n * P + p) * Q + q) * K;
+        const T16* wp = w + (long)(r * S + s) * C + c;
+        for (int k = 0; k < K; ++k)
+          acc += F16<T16>::to_f32(dyp[k]) *
+                 F16<T16>::to_f32(wp[(long)k * wkstride]);
+      }
+    }
+    dx[t] = F16<T16>::from_f32(acc);
+  }
+}
+
+template <typename T16>
+__global__ void conv_wgrad_direct(const T16* __restrict__ x,
+                                  const T16* __restrict__ dy,
+                                  float* __restrict__ dw, int N, int H, int W,
+                                  int C, int K, int R, int S, int P, int Q,
+                                  int stride, int pad, long m_per_chunk) {
+  const long total_w = (long)K * R * S * C;
+  const long M = (long)N * P * Q;
+  const long m0 = (long)blockIdx.y * m_per_chunk;
+  const long m1 = min(M, m0 + m_per_chunk);
+  const int nchunks = gridDim.y;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total_w;
+       t += (long)gridDim.x * blockDim.x) {
+    const int c = (int)(t % C);
+    const int s = (int)((t / C) % S);
+    const int r = (int)((t / ((long)C * S)) % R);
+    const int k = (int)(t / ((long)C * S * R));
+    float acc = 0.f;
+    for (long m = m0; m < m1; ++m) {
+      const int q = (int)(m % Q);
+      long np = m / Q;
+      const int p = (int)(np % P);
+      const int n = (int)(np / P);
+      const int ih = p * stride - pad + r;
+      const int iw = q * stride - pad + s;
+      if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
+      acc += F16<T16>::to_f32(dy[m * K + k]) *
+             F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * C + c]);
+    }
+    if (nchunks == 1)
+      dw[t] = acc;
+    else
+      atomicAdd(dw + t, acc);
+  }
+}
+
+inline dim3 conv_grid(long total, int block = 256, int cap = 4096) {
+  return dim3((unsigned)std::min<long>(cdiv_l(total, block), cap));
+}
+
+}  // namespace
+
+static inline int out_dim(int in, int k, int stride, int pad) {
+  return (in + 2 * pad - k) / stride + 1;
+}
+
+at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                      long stride, long pad, long act) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_16BIT(x);
+  CHECK_CONTIG(w);
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int K = w.size(0), R = w.size(1), S = w.size(2);
+  TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  auto y = at::empty({N, P, Q, K}, x.options());
+  const long total = (long)N * P * Q * K;
+  const int has_bias = bias.numel() > 0;
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL(conv_fwd_direct<T16>, conv_grid(total), dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)w.data_ptr(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       (T16*)y.data_ptr(), N, H, W, C, K, R, S, P, Q,
+                       (int)stride, (int)pad, (int)act, has_bias);
+  });
+  return y;
+}
+
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, long stride, long pad,
+                        long H, long W) {
+  CHECK_GPU(dy);
+  CHECK_CONTIG(dy);
+  CHECK_16BIT(dy);
+  const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
+  const int R = w.size(1), S = w.size(2), C = w.size(3);
+  auto dx = at::empty({N, H, W, (long)C}, dy.options());
+  const long total = (long)N * H * W * C;
+  DISPATCH_16(dy, T16, {
+    hipLaunchKernelGGL(conv_dgrad_direct<T16>, conv_grid(total), dim3(256), 0,
+                       cur_stream(), (const T16*)dy.data_ptr(),
+                       (const T16*)w.data_ptr(), (T16*)dx.data_ptr(), N,
+                       (int)H, (int)W, C, K, R, S, P, Q, (int)stride,
+                       (int)pad);
+  });
+  return dx;
+}
+
+at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
+                        long stride, long pad) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_CONTIG(dy);
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int P = dy.size(1), Q = dy.size(2), K = dy.size(3);
+  const long M = (long)N * P * Q;
+  const long total_w = (long)K * R * S * C;
+  // split the NPQ reduction so small filters still fill the chip
+  int nchunks = (int)std::min<long>(cdiv_l(M, 4096),
+                                    std::max<long>(1, (256L * 2048) / std::max(total_w, 1L)));
+  nchunks = std::max(nchunks, 1);
+  const long m_per_chunk = cdiv_l(M, nchunks);
+  auto dw = nchunks == 1
+                ? at::empty({K, R, S, (long)C},
+                            x.options().dtype(at::kFloat))
+                : at::zeros({K, R, S, (long)C}, x.options().dtype(at::kFloat));
+  dim3 grid((unsigned)std::min<long>(cdiv_l(total_w, 256), 4096), nchunks);
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL(conv_wgrad_direct<T16>, grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)dy.data_ptr(), dw.data_ptr<float>(), N, H,
+                       W, C, K, (int)R, (int)S, P, Q, (int)stride, (int)pad,
+                       m_per_chunk);
+  });
+  return dw;
+}

@@ -1,0 +1,128 @@
+// Linear layer GEMMs. The classifier GEMMs in this
+// workload are small ([B,512]x[512,10] for ResNet-18/CIFAR); v0 uses
+// simple coalesced direct kernels. The MFMA tile path (gemm_mfma.hip)
+// takes over for large shapes.
+#include "common.h"
+
+namespace {
+
+// y[M,N] = x[M,K] @ w[N,K]^T + b
+template <typename T16>
+__global__ void linear_fwd_kernel(const T16* __restrict__ x,
+                                  const T16* __restrict__ w,
+                                  const float* __restrict__ bias,
+                                  T16* __restrict__ y, int M, int N, int K,
+                                  int act, int has_bias) {
+  const long total = (long)M * N;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int nn = (int)(t % N);
+    const int m = (int)(t / N);
+    const T16* xp = x + (long)m * K;
+    const T16* wp = w + (long)nn * K;
+    float acc = has_bias ? bias[nn] : 0.f;
+    int k = 0;
+    if (K % 8 == 0) {
+      for (; k < K; k += 8) {
+        short8 xv = *reinterpret_cast<const short8*>(xp + k);
+        short8 wv = *reinterpret_cast<const short8*>(wp + k);
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          acc += s16_to_f32<T16>(xv[u]) * s16_to_f32<T16>(wv[u]);
+      }
+    } else {
+      for (; k < K; ++k)
+        acc += F16<T16>::to_f32(xp[k]) * F16<T16>::to_f32(wp[k]);
+    }
+    if (act == 1) acc = fmaxf(acc, 0.f);
+    y[t] = F16<T16>::from_f32(acc);
+  }
+}
+
+// dx[M,K] = dy[M,N] @ w[N,K]
+template <typename T16>
+__global__ void linear_dgrad_kernel(const T16* __restrict__ dy,
+                                    const T16* __restrict__ w,
+                                    T16* __restrict__ dx, int M, int N,
+                                    int K) {
+  const long total = (long)M * K;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(t % K);
+    const int m = (int)(t / K);
+    float acc = 0.f;
+    for (int nn = 0; nn < N; ++nn)
+      acc += F16<T16>::to_f32(dy[(long)m * N + nn]) *
+             F16<T16>::to_f32(w[(long)nn * K + k]);
+    dx[t] = F16<T16>::from_f32(acc);
+  }
+}
+
+// dw[N,K] (f32) = dy[M,N]^T @ x[M,K]
+template <typename T16>
+__global__ void linear_wgrad_kernel(const T16* __restrict__ x,
+                                    const T16* __restrict__ dy,
+                                    float* __restrict__ dw, int M, int N,
+                                    int K) {
+  const long total = (long)N * K;
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int k = (int)(t % K);
+    const int nn = (int)(t / K);
+    float acc = 0.f;
+    for (int m = 0; m < M; ++m)
+      acc += F16<T16>::to_f32(dy[(long)m * N + nn]) *
+             F16<T16>::to_f32(x[(long)m * K + k]);
+    dw[t] = acc;
+  }
+}
+
+inline int ggrid(long n) { return (int)std::min<long>(cdiv_l(n, 256), 4096); }
+
+}  // namespace
+
+at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_16BIT(x);
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(w.size(1) == K, "linear weight/input mismatch");
+  auto y = at::empty({M, N}, x.options());
+  const int has_bias = bias.numel() > 0;
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL(linear_fwd_kernel<T16>, dim3(ggrid((long)M * N)),
+                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)w.data_ptr(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       (T16*)y.data_ptr(), M, N, K, (int)act, has_bias);
+  });
+  return y;
+}
+
+at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w) {
+  CHECK_GPU(dy);
+  CHECK_CONTIG(dy);
+  const int M = dy.size(0), N = dy.size(1), K = w.size(1);
+  auto dx = at::empty({M, K}, dy.options());
+  DISPATCH_16(dy, T16, {
+    hipLaunchKernelGGL(linear_dgrad_kernel<T16>, dim3(ggrid((long)M * K)),
+                       dim3(256), 0, cur_stream(), (const T16*)dy.data_ptr(),
+                       (const T16*)w.data_ptr(), (T16*)dx.data_ptr(), M, N,
+                       K);
+  });
+  return dx;
+}
+
+at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  const int M = x.size(0), K = x.size(1), N = dy.size(1);
+  auto dw = at::empty({N, K}, x.options().dtype(at::kFloat));
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL(linear_wgrad_kernel<T16>, dim3(ggrid((long)N * K)),
+                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)dy.data_ptr(), dw.data_ptr<float>(), M, N,
+                       K);
+  });
+  return dw;
+}

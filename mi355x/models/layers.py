@@ -1,0 +1,100 @@
+"""nn.Module wrappers over the mi355x ops layer.
+
+Parameters keep the torch-conventional shapes so checkpoints interchange
+with the reference scripts' state_dicts; forward runs NHWC through the
+mi355x kernels on GPU (bf16/fp16) and the plain-torch fp32 path on CPU.
+"""
+
+from __future__ import annotations
+
+import math
+
+import torch
+from torch import nn
+
+from mi355x import ops
+
+
+def to_model_layout(x: torch.Tensor) -> torch.Tensor:
+    """NCHW input (reference contract) -> NHWC activation, 16-bit on GPU."""
+    x = x.permute(0, 2, 3, 1)
+    if x.is_cuda:
+        x = x.to(ops.compute_dtype())
+    return x.contiguous()
+
+
+class Conv2d(nn.Module):
+    def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+                 padding=0, bias=True, act=None):
+        super().__init__()
+        self.stride, self.padding, self.act = stride, padding, act
+        self.weight = nn.Parameter(
+            torch.empty(out_channels, in_channels, kernel_size, kernel_size))
+        self.bias = nn.Parameter(torch.empty(out_channels)) if bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+        if self.bias is not None:
+            fan_in = self.weight.shape[1] * self.weight.shape[2] * self.weight.shape[3]
+            bound = 1 / math.sqrt(fan_in)
+            nn.init.uniform_(self.bias, -bound, bound)
+
+    def forward(self, x):
+        return ops.conv2d(x, self.weight, self.bias, self.stride,
+                          self.padding, self.act)
+
+
+class Linear(nn.Module):
+    def __init__(self, in_features, out_features, bias=True, act=None):
+        super().__init__()
+        self.act = act
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.empty(out_features)) if bias else None
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+        if self.bias is not None:
+            bound = 1 / math.sqrt(self.weight.shape[1])
+            nn.init.uniform_(self.bias, -bound, bound)
+
+    def forward(self, x):
+        return ops.linear(x, self.weight, self.bias, self.act)
+
+
+class BatchNorm2d(nn.Module):
+    """BN over NHWC with optional fused residual-add + ReLU epilogue.
+
+    `sync` marks the layer for SyncBatchNorm: when a process group is
+    installed (mi355x.parallel.sync_bn.convert / enable), batch statistics
+    are all-reduced across ranks (BASELINE config 5).
+    """
+
+    def __init__(self, num_features, momentum=0.1, act=None):
+        super().__init__()
+        self.momentum, self.act = momentum, act
+        self.weight = nn.Parameter(torch.ones(num_features))
+        self.bias = nn.Parameter(torch.zeros(num_features))
+        self.register_buffer("running_mean", torch.zeros(num_features))
+        self.register_buffer("running_var", torch.ones(num_features))
+        self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        self.process_group = None  # set by sync_bn.enable()
+
+    def forward(self, x, residual=None):
+        if self.training:
+            self.num_batches_tracked += 1
+        return ops.batch_norm(x, self.weight, self.bias, self.running_mean,
+                              self.running_var, self.training, self.momentum,
+                              residual, self.act, self.process_group if self.training else None)
+
+
+class MaxPool2d(nn.Module):
+    def __init__(self, kernel_size, stride=None, padding=0):
+        super().__init__()
+        self.kernel_size = kernel_size
+        self.stride = stride or kernel_size
+        self.padding = padding
+
+    def forward(self, x):
+        return ops.max_pool2d(x, self.kernel_size, self.stride, self.padding)

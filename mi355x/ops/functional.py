@@ -1,0 +1,349 @@
+"""Functional op layer: HIP/CDNA4 kernels on GPU, plain-torch fp32 on CPU.
+
+Activation layout on GPU is NHWC end to end (coalesced channel-minor access
+for CDNA4 implicit-GEMM conv; see mi355x/csrc/). Parameters are stored in
+the PyTorch-conventional shapes/dtypes ([K,C,R,S] fp32 conv weight, etc.) so
+state_dicts interchange with the reference scripts
+(/root/reference/cifar_example.py:92-93 checkpoint format); 16-bit KRSC
+copies for the kernels are cached per parameter version.
+
+CPU tensors run a plain differentiable fp32 torch path — this is both the
+BASELINE config-1 (CPU) execution mode and the numerics reference that the
+GPU kernels are unit-tested against. CUDA tensors REQUIRE the mi355x._C
+extension (no silent PyTorch fallback on GPU).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn.functional as F
+
+from ._ext import ext
+
+BN_EPS = 1e-5
+
+_ACT_NONE = 0
+_ACT_RELU = 1
+
+
+def compute_dtype() -> torch.dtype:
+    """Current 16-bit compute dtype for GPU kernels (bf16 unless amp says fp16)."""
+    from mi355x import amp
+
+    return amp.get_compute_dtype()
+
+
+# ---------------------------------------------------------------------------
+# cached 16-bit weight copies
+# ---------------------------------------------------------------------------
+
+# Bumped by the optimizer after each step (flat-param in-place updates do not
+# advance the per-parameter ._version, so the cache keys on both).
+_cache_epoch = 0
+
+
+def bump_cache_epoch() -> None:
+    global _cache_epoch
+    _cache_epoch += 1
+
+
+def _w16_conv(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """fp32 [K,C,R,S] parameter -> cached 16-bit [K,R,S,C] kernel copy."""
+    key = (weight._version, _cache_epoch, dtype)
+    cache = getattr(weight, "_mi355x_w16", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    w16 = weight.detach().to(dtype).permute(0, 2, 3, 1).contiguous()
+    weight._mi355x_w16 = (key, w16)
+    return w16
+
+
+def _w16_linear(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """fp32 [N,K] parameter -> cached 16-bit [N,K] kernel copy."""
+    key = (weight._version, _cache_epoch, dtype)
+    cache = getattr(weight, "_mi355x_w16", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    w16 = weight.detach().to(dtype).contiguous()
+    weight._mi355x_w16 = (key, w16)
+    return w16
+
+
+def _relu_mask_bwd(dy: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
+    """dy masked by y>0, via the extension's elementwise kernel."""
+    return ext().relu_bwd(dy.contiguous(), y)
+
+
+# ---------------------------------------------------------------------------
+# conv2d (NHWC implicit GEMM)
+# ---------------------------------------------------------------------------
+
+
+class _ConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride, padding, act):
+        dtype = x.dtype
+        w16 = _w16_conv(weight, dtype)
+        b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
+        y = ext().conv2d_fwd(x, w16, b32, stride, padding, act)
+        ctx.save_for_backward(x, w16, y)
+        ctx.conf = (stride, padding, act, bias is not None, weight.shape)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w16, y = ctx.saved_tensors
+        stride, padding, act, has_bias, wshape = ctx.conf
+        dy = dy.contiguous()
+        if act == _ACT_RELU:
+            dy = _relu_mask_bwd(dy, y)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = ext().conv2d_dgrad(dy, w16, stride, padding, x.shape[1], x.shape[2])
+        dw = None
+        if ctx.needs_input_grad[1]:
+            dw_krsc = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride, padding)
+            dw = dw_krsc.permute(0, 3, 1, 2).contiguous()  # fp32 [K,C,R,S]
+        db = None
+        if has_bias and ctx.needs_input_grad[2]:
+            db = dy.float().sum(dim=(0, 1, 2))
+        return dx, dw, db, None, None, None
+
+
+def conv2d(x, weight, bias=None, stride=1, padding=0, act=None):
+    """2-D convolution. x: NHWC; weight: fp32 [K,C,R,S] (torch layout).
+
+    act: None | 'relu' fused into the epilogue (and into the kernel backward
+    mask). Reference call sites: /root/reference/cifar_example.py:20-29.
+    """
+    a = _ACT_RELU if act == "relu" else _ACT_NONE
+    if x.is_cuda:
+        return _ConvFn.apply(x, weight, bias, stride, padding, a)
+    y = F.conv2d(x.permute(0, 3, 1, 2), weight, bias, stride, padding)
+    y = y.permute(0, 2, 3, 1)
+    if a == _ACT_RELU:
+        y = F.relu(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# linear (MFMA GEMM)
+# ---------------------------------------------------------------------------
+
+
+class _LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, act):
+        w16 = _w16_linear(weight, x.dtype)
+        b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
+        y = ext().linear_fwd(x, w16, b32, act)
+        ctx.save_for_backward(x, w16, y)
+        ctx.conf = (act, bias is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w16, y = ctx.saved_tensors
+        act, has_bias = ctx.conf
+        dy = dy.contiguous()
+        if act == _ACT_RELU:
+            dy = _relu_mask_bwd(dy, y)
+        dx = ext().linear_dgrad(dy, w16) if ctx.needs_input_grad[0] else None
+        dw = ext().linear_wgrad(x, dy) if ctx.needs_input_grad[1] else None
+        db = dy.float().sum(dim=0) if (has_bias and ctx.needs_input_grad[2]) else None
+        return dx, dw, db, None
+
+
+def linear(x, weight, bias=None, act=None):
+    """y = x @ W^T + b. x: [M,K]; weight: fp32 [N,K]."""
+    a = _ACT_RELU if act == "relu" else _ACT_NONE
+    if x.is_cuda:
+        return _LinearFn.apply(x, weight, bias, a)
+    y = F.linear(x, weight, bias)
+    if a == _ACT_RELU:
+        y = F.relu(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# batch norm (+ residual add + relu, fused)
+# ---------------------------------------------------------------------------
+
+
+class _BNFn(torch.autograd.Function):
+    """Training-mode batch norm over NHWC with optional fused residual add
+    and ReLU. Stats in fp32. Optionally SyncBN: partial (sum,sumsq) and the
+    backward (sum_dy, sum_dy_xhat) reductions are all-reduced over the
+    process group (BASELINE config 5)."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, residual, running_mean, running_var,
+                momentum, act, process_group):
+        N, H, W, C = x.shape
+        m_local = N * H * W
+        s = ext().bn_stats(x)  # f32 [2,C]: sum, sumsq
+        m_total = m_local
+        if process_group is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(s, group=process_group)
+            m_total = m_local * dist.get_world_size(process_group)
+        mean = s[0] / m_total
+        var = s[1] / m_total - mean * mean
+        invstd = torch.rsqrt(var + BN_EPS)
+        if running_mean is not None:
+            with torch.no_grad():
+                unbiased = var * (m_total / max(m_total - 1, 1))
+                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
+                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
+        y = ext().bn_apply(x, mean, invstd, gamma.detach().float(),
+                           beta.detach().float(), res, act)
+        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        ctx.conf = (act, residual is not None, m_total, process_group)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, y, mean, invstd, gamma = ctx.saved_tensors
+        act, has_res, m_total, pg = ctx.conf
+        dy = dy.contiguous()
+        if act == _ACT_RELU:
+            dy = _relu_mask_bwd(dy, y)
+        r = ext().bn_bwd_reduce(x, dy, mean, invstd)  # f32 [2,C]: sum_dy_xhat, sum_dy
+        if pg is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(r, group=pg)
+        dgamma, dbeta = r[0], r[1]
+        dx = ext().bn_bwd_dx(x, dy, mean, invstd, gamma.detach().float(),
+                             dgamma, dbeta, float(m_total))
+        dres = dy if has_res else None
+        return dx, dgamma, dbeta, dres, None, None, None, None, None
+
+
+def batch_norm(x, gamma, beta, running_mean, running_var, training,
+               momentum=0.1, residual=None, act=None, process_group=None):
+    """BatchNorm over NHWC with optional fused residual-add + ReLU.
+
+    Matches torch BN semantics (momentum convention, unbiased running var).
+    process_group != None => SyncBatchNorm (stats all-reduced over ranks).
+    """
+    a = _ACT_RELU if act == "relu" else _ACT_NONE
+    if x.is_cuda:
+        if training:
+            return _BNFn.apply(x, gamma, beta, residual, running_mean,
+                               running_var, momentum, a, process_group)
+        mean = running_mean.float()
+        invstd = torch.rsqrt(running_var.float() + BN_EPS)
+        res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
+        return ext().bn_apply(x.contiguous(), mean, invstd, gamma.detach().float(),
+                              beta.detach().float(), res, a)
+    xc = x.permute(0, 3, 1, 2)
+    y = F.batch_norm(xc, running_mean, running_var, gamma, beta, training,
+                     momentum, BN_EPS).permute(0, 2, 3, 1)
+    if residual is not None:
+        y = y + residual
+    if a == _ACT_RELU:
+        y = F.relu(y)
+    return y
+
+
+# ---------------------------------------------------------------------------
+# pooling
+# ---------------------------------------------------------------------------
+
+
+class _MaxPoolFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, kernel, stride, padding):
+        y, idx = ext().maxpool_fwd(x, kernel, stride, padding)
+        ctx.save_for_backward(idx)
+        ctx.conf = (x.shape, kernel, stride, padding)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (idx,) = ctx.saved_tensors
+        xshape, kernel, stride, padding = ctx.conf
+        dx = ext().maxpool_bwd(dy.contiguous(), idx, xshape[1], xshape[2],
+                               kernel, stride, padding)
+        return dx, None, None, None
+
+
+def max_pool2d(x, kernel, stride=None, padding=0):
+    """Max pooling over NHWC. Reference: /root/reference/cifar_example.py:21."""
+    stride = stride or kernel
+    if x.is_cuda:
+        return _MaxPoolFn.apply(x, kernel, stride, padding)
+    y = F.max_pool2d(x.permute(0, 3, 1, 2), kernel, stride, padding)
+    return y.permute(0, 2, 3, 1)
+
+
+class _GAPFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.conf = x.shape
+        return ext().global_avg_pool(x)
+
+    @staticmethod
+    def backward(ctx, dy):
+        N, H, W, C = ctx.conf
+        dx = (dy / (H * W)).reshape(N, 1, 1, C).expand(N, H, W, C).contiguous()
+        return dx
+
+
+def global_avg_pool(x):
+    """NHWC [N,H,W,C] -> [N,C] mean over H,W."""
+    if x.is_cuda:
+        return _GAPFn.apply(x)
+    return x.mean(dim=(1, 2))
+
+
+# ---------------------------------------------------------------------------
+# loss
+# ---------------------------------------------------------------------------
+
+
+class _CEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        loss, lse = ext().cross_entropy_fwd(logits, target)
+        ctx.save_for_backward(logits, target, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, target, lse = ctx.saved_tensors
+        scale = float(dloss) / logits.shape[0]
+        dlogits = ext().cross_entropy_bwd(logits, target, lse, scale)
+        return dlogits, None
+
+
+def cross_entropy(logits, target):
+    """Mean cross-entropy. logits [B,C] (16-bit on GPU), target int64 [B].
+    Reference call site: /root/reference/cifar_example.py:63,78."""
+    if logits.is_cuda:
+        return _CEFn.apply(logits, target)
+    return F.cross_entropy(logits.float(), target)
+
+
+# ---------------------------------------------------------------------------
+# fused optimizer step
+# ---------------------------------------------------------------------------
+
+
+def sgd_step(flat_param, flat_grad, flat_momentum, lr, momentum,
+             weight_decay=0.0, grad_scale=1.0):
+    """v = mu*v + (g*gs + wd*p); p -= lr*v — one kernel over the flat fp32
+    buffers (reference semantics: optim.SGD(lr,momentum),
+    /root/reference/cifar_example.py:64)."""
+    if flat_param.is_cuda:
+        ext().sgd_step(flat_param, flat_grad, flat_momentum, lr, momentum,
+                       weight_decay, grad_scale)
+        return
+    g = flat_grad.mul(grad_scale)
+    if weight_decay != 0.0:
+        g = g.add(flat_param, alpha=weight_decay)
+    flat_momentum.mul_(momentum).add_(g)
+    flat_param.add_(flat_momentum, alpha=-lr)

@@ -1,0 +1,209 @@
+"""GPU kernel numerics: every HIP kernel vs the plain-torch fp32 CPU
+reference on randomized shapes. bf16 tolerances."""
+
+import pytest
+import torch
+
+import mi355x.ops as ops
+from mi355x.ops import functional as fn
+
+pytestmark = pytest.mark.gpu
+
+RTOL, ATOL = 3e-2, 3e-2
+
+
+def _pair(shape, requires_grad=False, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    cpu = torch.randn(*shape, generator=g)
+    gpu = cpu.to("cuda").to(torch.bfloat16)
+    cpu.requires_grad_(requires_grad)
+    gpu.requires_grad_(requires_grad)
+    return cpu, gpu
+
+
+def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
+    torch.testing.assert_close(gpu_t.float().cpu(), cpu_t.float(),
+                               rtol=rtol, atol=atol)
+
+
+@pytest.mark.parametrize("shape,K,ksz,stride,pad,bias,act", [
+    ((2, 32, 32, 3), 6, 5, 1, 0, True, "relu"),   # Net conv1
+    ((2, 14, 14, 6), 16, 5, 1, 0, True, None),    # Net conv2
+    ((2, 32, 32, 16), 32, 3, 1, 1, False, None),  # ResNet 3x3
+    ((2, 16, 16, 32), 64, 3, 2, 1, False, None),  # stride-2 downsample
+    ((2, 16, 16, 32), 64, 1, 2, 0, False, None),  # 1x1 shortcut
+    ((2, 33, 33, 8), 16, 7, 2, 3, False, None),   # 7x7/2 stem, odd size
+])
+def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
+    C = shape[-1]
+    xc, xg = _pair(shape, requires_grad=True)
+    g = torch.Generator().manual_seed(1)
+    w = torch.randn(K, C, ksz, ksz, generator=g) * 0.2
+    b = torch.randn(K, generator=g) * 0.1 if bias else None
+    wc = w.clone().requires_grad_(True)
+    wg = w.clone().cuda().requires_grad_(True)
+    bc = b.clone().requires_grad_(True) if bias else None
+    bg = b.clone().cuda().requires_grad_(True) if bias else None
+
+    yc = fn.conv2d(xc, wc, bc, stride, pad, act)
+    yg = fn.conv2d(xg, wg, bg, stride, pad, act)
+    _close(yg, yc)
+
+    dy = torch.randn(*yc.shape, generator=g)
+    yc.backward(dy)
+    yg.backward(dy.cuda().to(torch.bfloat16))
+    _close(xg.grad, xc.grad)
+    _close(wg.grad, wc.grad, rtol=5e-2, atol=5e-2)
+    if bias:
+        _close(bg.grad, bc.grad, rtol=5e-2, atol=5e-2)
+
+
+@pytest.mark.parametrize("M,N,K,bias,act", [
+    (8, 10, 400, True, None),
+    (16, 120, 400, True, "relu"),
+    (4, 84, 120, True, "relu"),
+    (32, 10, 512, True, None),
+    (8, 33, 100, False, None),  # K % 8 != 0 scalar path
+])
+def test_linear_fwd_bwd(M, N, K, bias, act):
+    xc, xg = _pair((M, K), requires_grad=True)
+    g = torch.Generator().manual_seed(2)
+    w = torch.randn(N, K, generator=g) * 0.1
+    b = torch.randn(N, generator=g) * 0.1 if bias else None
+    wc = w.clone().requires_grad_(True)
+    wg = w.clone().cuda().requires_grad_(True)
+    bc = b.clone().requires_grad_(True) if bias else None
+    bg = b.clone().cuda().requires_grad_(True) if bias else None
+
+    yc = fn.linear(xc, wc, bc, act)
+    yg = fn.linear(xg, wg, bg, act)
+    _close(yg, yc)
+
+    dy = torch.randn(*yc.shape, generator=g)
+    yc.backward(dy)
+    yg.backward(dy.cuda().to(torch.bfloat16))
+    _close(xg.grad, xc.grad)
+    _close(wg.grad, wc.grad)
+    if bias:
+        _close(bg.grad, bc.grad)
+
+
+@pytest.mark.parametrize("shape,res,act", [
+    ((4, 8, 8, 16), False, "relu"),
+    ((4, 8, 8, 16), True, "relu"),
+    ((2, 5, 7, 32), False, None),
+])
+def test_batch_norm_train_fwd_bwd(shape, res, act):
+    C = shape[-1]
+    xc, xg = _pair(shape, requires_grad=True)
+    resc = resg = None
+    if res:
+        resc, resg = _pair(shape, requires_grad=True, seed=9)
+    g = torch.Generator().manual_seed(3)
+    gamma = torch.rand(C, generator=g) + 0.5
+    beta = torch.randn(C, generator=g) * 0.1
+    gc, bc = gamma.clone().requires_grad_(True), beta.clone().requires_grad_(True)
+    gg, bg = gamma.clone().cuda().requires_grad_(True), beta.clone().cuda().requires_grad_(True)
+    rm_c, rv_c = torch.zeros(C), torch.ones(C)
+    rm_g, rv_g = rm_c.clone().cuda(), rv_c.clone().cuda()
+
+    yc = fn.batch_norm(xc, gc, bc, rm_c, rv_c, True, 0.1, resc, act)
+    yg = fn.batch_norm(xg, gg, bg, rm_g, rv_g, True, 0.1, resg, act)
+    _close(yg, yc)
+    _close(rm_g, rm_c, rtol=2e-2, atol=2e-2)  # running stats updated alike
+    _close(rv_g, rv_c, rtol=2e-2, atol=2e-2)
+
+    dy = torch.randn(*yc.shape, generator=g)
+    yc.backward(dy)
+    yg.backward(dy.cuda().to(torch.bfloat16))
+    _close(xg.grad, xc.grad, rtol=5e-2, atol=5e-2)
+    _close(gg.grad, gc.grad, rtol=5e-2, atol=5e-2)
+    _close(bg.grad, bc.grad, rtol=5e-2, atol=5e-2)
+    if res:
+        _close(resg.grad, resc.grad)
+
+
+def test_batch_norm_eval():
+    C = 8
+    xc, xg = _pair((2, 4, 4, C))
+    rm = torch.randn(C) * 0.1
+    rv = torch.rand(C) + 0.5
+    gamma, beta = torch.rand(C) + 0.5, torch.randn(C) * 0.1
+    yc = fn.batch_norm(xc, gamma, beta, rm, rv, False)
+    yg = fn.batch_norm(xg, gamma.cuda(), beta.cuda(), rm.cuda(), rv.cuda(),
+                       False)
+    _close(yg, yc)
+
+
+@pytest.mark.parametrize("shape,k,s,p", [
+    ((2, 28, 28, 6), 2, 2, 0),   # Net pool
+    ((2, 10, 10, 16), 2, 2, 0),
+    ((2, 56, 56, 8), 3, 2, 1),   # ResNet stem pool (overlapping)
+])
+def test_maxpool_fwd_bwd(shape, k, s, p):
+    xc, xg = _pair(shape, requires_grad=True)
+    yc = fn.max_pool2d(xc, k, s, p)
+    yg = fn.max_pool2d(xg, k, s, p)
+    _close(yg, yc)
+    g = torch.Generator().manual_seed(4)
+    dy = torch.randn(*yc.shape, generator=g)
+    yc.backward(dy)
+    yg.backward(dy.cuda().to(torch.bfloat16))
+    _close(xg.grad, xc.grad)
+
+
+def test_global_avg_pool():
+    xc, xg = _pair((3, 4, 4, 512), requires_grad=True)
+    yc = fn.global_avg_pool(xc)
+    yg = fn.global_avg_pool(xg)
+    _close(yg, yc)
+    g = torch.Generator().manual_seed(5)
+    dy = torch.randn(*yc.shape, generator=g)
+    yc.backward(dy)
+    yg.backward(dy.cuda().to(torch.bfloat16))
+    _close(xg.grad, xc.grad)
+
+
+@pytest.mark.parametrize("B,C", [(8, 10), (64, 1000)])
+def test_cross_entropy_fwd_bwd(B, C):
+    xc, xg = _pair((B, C), requires_grad=True)
+    g = torch.Generator().manual_seed(6)
+    t = torch.randint(0, C, (B,), generator=g)
+    lc = fn.cross_entropy(xc, t)
+    lg = fn.cross_entropy(xg, t.cuda())
+    _close(lg, lc, rtol=2e-2, atol=2e-2)
+    lc.backward()
+    lg.backward()
+    _close(xg.grad, xc.grad, rtol=2e-2, atol=1e-3)
+
+
+def test_relu_bwd_kernel():
+    g = torch.Generator().manual_seed(7)
+    y = torch.randn(1000, generator=g)
+    dy = torch.randn(1000, generator=g)
+    out = ops.ext().relu_bwd(dy.cuda().to(torch.bfloat16),
+                             y.cuda().to(torch.bfloat16))
+    ref = dy * (y > 0)
+    _close(out, ref.to(torch.bfloat16).float())
+
+
+def test_sgd_step_kernel():
+    g = torch.Generator().manual_seed(8)
+    n = 10001  # odd size: exercises the scalar tail
+    p = torch.randn(n, generator=g)
+    grad = torch.randn(n, generator=g)
+    m = torch.randn(n, generator=g)
+    pg, gg, mg = p.clone().cuda(), grad.clone().cuda(), m.clone().cuda()
+    fn.sgd_step(pg, gg, mg, 0.1, 0.9, 1e-4, 0.5)
+    fn.sgd_step(p, grad, m, 0.1, 0.9, 1e-4, 0.5)
+    torch.testing.assert_close(pg.cpu(), p, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(mg.cpu(), m, rtol=1e-5, atol=1e-6)
+
+
+def test_fp16_dtype_path():
+    from mi355x import amp
+    with amp.autocast(torch.float16):
+        x = torch.randn(4, 8, 8, 16).cuda().half()
+        w = torch.randn(16, 16, 3, 3) * 0.2
+        y = fn.conv2d(x, w.cuda(), None, 1, 1, None)
+        assert y.dtype == torch.float16
