@@ -140,6 +140,15 @@ static inline int out_dim(int in, int k, int stride, int pad) {
   return (in + 2 * pad - k) / stride + 1;
 }
 
+// conv_mfma.hip — MFMA implicit-GEMM path for C%64==0 && K%64==0
+bool conv_mfma_supported(long CI, long KO);
+void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                          at::Tensor y, long stride, long pad, long act);
+void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
+                            long R, long S, long stride, long pad);
+void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+                            long R, long S, long stride, long pad);
+
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act) {
   CHECK_GPU(x);
@@ -151,6 +160,10 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
   auto y = at::empty({N, P, Q, K}, x.options());
+  if (conv_mfma_supported(C, K)) {
+    conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act);
+    return y;
+  }
   const long total = (long)N * P * Q * K;
   const int has_bias = bias.numel() > 0;
   DISPATCH_16(x, T16, {
@@ -164,14 +177,18 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   return y;
 }
 
-at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, long stride, long pad,
-                        long H, long W) {
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wflip,
+                        long stride, long pad, long H, long W) {
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
   CHECK_16BIT(dy);
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
   const int R = w.size(1), S = w.size(2), C = w.size(3);
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
+  if (wflip.numel() > 0 && conv_mfma_supported(K, C)) {
+    conv_dgrad_mfma_launch(dy, wflip, dx, R, S, stride, pad);
+    return dx;
+  }
   const long total = (long)N * H * W * C;
   DISPATCH_16(dy, T16, {
     hipLaunchKernelGGL(conv_dgrad_direct<T16>, conv_grid(total), dim3(256), 0,
@@ -192,6 +209,11 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
   const int P = dy.size(1), Q = dy.size(2), K = dy.size(3);
   const long M = (long)N * P * Q;
   const long total_w = (long)K * R * S * C;
+  if (conv_mfma_supported(C, K)) {
+    auto dw = at::zeros({K, R, S, (long)C}, x.options().dtype(at::kFloat));
+    conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad);
+    return dw;
+  }
   // split the NPQ reduction so small filters still fill the chip
   int nchunks = (int)std::min<long>(cdiv_l(M, 4096),
                                     std::max<long>(1, (256L * 2048) / std::max(total_w, 1L)));

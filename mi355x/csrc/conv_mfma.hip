@@ -1,0 +1,353 @@
+// MFMA implicit-GEMM conv for CDNA4 (gfx950) — the hot path for every
+// ResNet conv (C%64==0, K%64==0). One gather-GEMM kernel serves forward
+// (TRANS=false) and dgrad (TRANS=true; host pre-flips the weight to
+// [R,S,C,K] so dgrad is the same GEMM with transposed-window gather).
+//
+// GEMM view: OUT[M=N*Ho*Wo, KO] = A[M, R*S*CI] * B,
+// iterated as (r,s,c-chunk) k-slices of BK=64. A k-slice of A is a
+// contiguous 64-channel read of the input window row (NHWC), staged
+// through registers into LDS with zero-fill for padding/out-of-window
+// rows (predication the glds path can't do). Structure follows the
+// canonical CDNA GEMM anatomy (cdna_hip_programming.md §5): 128x64 block
+// tile, 4 waves x (32M x 64N) each as 2 x mfma_f32_32x32x16, LDS tiles
+// [row][BK+8] padded against bank conflicts (G4), T14-style staging
+// (write after barrier, next global load issued under the MFMA phase).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) _Float16 half8;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+namespace {
+
+template <typename T16>
+struct Mfma32 {};
+template <>
+struct Mfma32<__hip_bfloat16> {
+  static DEV_INLINE f32x16 run(short8 a, short8 b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct Mfma32<__half> {
+  static DEV_INLINE f32x16 run(short8 a, short8 b, f32x16 c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16((half8)a, (half8)b, c, 0, 0,
+                                                  0);
+  }
+};
+
+constexpr int BM = 128, BN = 64, BK = 64;
+constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
+
+// TRANS=false (fwd): src row = ho*stride - pad + r, in [0,Hi)
+// TRANS=true (dgrad): src row = (ho + pad - r), valid iff %stride==0, /stride in [0,Hi)
+template <typename T16, bool TRANS>
+__global__ __launch_bounds__(256) void conv_gather_gemm(
+    const T16* __restrict__ in,    // [N, Hi, Wi, CI]
+    const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
+    const float* __restrict__ bias,  // [KO] or null
+    T16* __restrict__ out,         // [N*Ho*Wo, KO]
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    const int Ho, const int Wo, const int R, const int S, const int stride,
+    const int pad, const long b_row_stride, const long b_rs_stride,
+    const int act, const int has_bias) {
+  __shared__ T16 lds[BM * LDK + BN * LDK];
+  T16* ldsA = lds;
+  T16* ldsB = lds + BM * LDK;
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const long bm0 = (long)blockIdx.x * BM;
+  const int k0 = blockIdx.y * BN;
+
+  // ---- A staging coords (2 threads per m-row, 32 channels each) ----
+  const int sa_m = tid >> 1;
+  const int sa_c = (tid & 1) * (BK / 2);
+  const long m_a = bm0 + sa_m;
+  const bool m_ok = m_a < Mtot;
+  int n_ = 0, p_ = 0, q_ = 0;
+  if (m_ok) {
+    n_ = (int)(m_a / ((long)Ho * Wo));
+    const int pq = (int)(m_a % ((long)Ho * Wo));
+    p_ = pq / Wo;
+    q_ = pq % Wo;
+  }
+  // fwd: top-left of the receptive field; dgrad: p_,q_ used directly
+  const int ih0 = TRANS ? p_ : p_ * stride - pad;
+  const int iw0 = TRANS ? q_ : q_ * stride - pad;
+
+  // ---- B staging coords (4 threads per n-row, 16 elements each) ----
+  const int sb_n = tid >> 2;
+  const int sb_c = (tid & 3) * (BK / 4);
+  const T16* wrow = wgt + (long)(k0 + sb_n) * b_row_stride + sb_c;
+
+  // ---- wave/lane fragment coords ----
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;  // k-half: lane holds k = kk + kh*8 + e
+  const int wm = wave * 32;  // wave's m-offset inside the block tile
+
+  f32x16 acc0 = {}, acc1 = {};
+
+  const int cchunks = CI / BK;
+  const int ksteps = R * S * cchunks;
+
+  short8 sa[4];  // 32 channels = 4 x 16B
+  short8 sb[2];  // 16 elements = 2 x 16B
+
+  auto load_step = [&](int j) {
+    const int c0 = (j % cchunks) * BK;
+    const int s_ = (j / cchunks) % S;
+    const int r_ = j / (cchunks * S);
+    bool va = m_ok;
+    long ioff = 0;
+    if constexpr (TRANS) {
+      const int ph = ih0 + pad - r_;
+      const int qw = iw0 + pad - s_;
+      const int pp = ph / stride, qq = qw / stride;
+      va = va && ph >= 0 && qw >= 0 && (ph % stride) == 0 &&
+           (qw % stride) == 0 && pp < Hi && qq < Wi;
+      if (va) ioff = (((long)n_ * Hi + pp) * Wi + qq) * CI;
+    } else {
+      const int ih = ih0 + r_;
+      const int iw = iw0 + s_;
+      va = va && (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
+      if (va) ioff = (((long)n_ * Hi + ih) * Wi + iw) * CI;
+    }
+    if (va) {
+      const T16* xp = in + ioff + c0 + sa_c;
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        sa[i] = *reinterpret_cast<const short8*>(xp + 8 * i);
+    } else {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) sa[i] = short8{};
+    }
+    const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
+    sb[0] = *reinterpret_cast<const short8*>(wp);
+    sb[1] = *reinterpret_cast<const short8*>(wp + 8);
+  };
+
+  load_step(0);
+  for (int j = 0; j < ksteps; ++j) {
+    __syncthreads();  // previous MFMA phase done reading LDS
+    {
+      short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
+      short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
+      *reinterpret_cast<short8*>(pb) = sb[0];
+      *reinterpret_cast<short8*>(pb + 8) = sb[1];
+    }
+    __syncthreads();
+    if (j + 1 < ksteps) load_step(j + 1);  // HBM latency hides under MFMA
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsA + (wm + li) * LDK + kk + kh * 8);
+      const short8 bf0 = *reinterpret_cast<const short8*>(
+          ldsB + li * LDK + kk + kh * 8);
+      const short8 bf1 = *reinterpret_cast<const short8*>(
+          ldsB + (32 + li) * LDK + kk + kh * 8);
+      acc0 = Mfma32<T16>::run(af, bf0, acc0);
+      acc1 = Mfma32<T16>::run(af, bf1, acc1);
+    }
+  }
+
+  // ---- epilogue: bias + act + store ----
+  const float b0 = has_bias ? bias[k0 + li] : 0.f;
+  const float b1 = has_bias ? bias[k0 + 32 + li] : 0.f;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mtot) {
+      float v0 = acc0[reg] + b0;
+      float v1 = acc1[reg] + b1;
+      if (act == 1) {
+        v0 = fmaxf(v0, 0.f);
+        v1 = fmaxf(v1, 0.f);
+      }
+      out[m_out * KO + k0 + li] = F16<T16>::from_f32(v0);
+      out[m_out * KO + k0 + 32 + li] = F16<T16>::from_f32(v1);
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// wgrad: dw[KO, kg] = sum_m dy[m, KO] * A_im2col[m, kg]. MFMA reduces over
+// its register-minor k dim, which here is m — both operands are m-major in
+// memory, so both tiles are staged TRANSPOSED into LDS ([k][m] / [kg][m]),
+// packing 4 m-values per ds_write_b64 during staging. Each block owns a
+// 64(KO) x 64(kg at one (r,s,c-chunk)) output tile and an m-chunk;
+// partials are atomicAdd'ed in fp32 (split-M fills the chip for the small
+// late-layer filter counts).
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int WGM = 64;   // m per step
+constexpr int LDM = 72;   // m-minor row length (+16B: alignment + banks)
+
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
+    const T16* __restrict__ x,    // [N, Hi, Wi, CI]
+    const T16* __restrict__ dy,   // [M, KO]
+    float* __restrict__ dw,       // [KO, R*S*CI]
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    const int Ho, const int Wo, const int R, const int S, const int stride,
+    const int pad, const long m_per_chunk, const int nchunks) {
+  __shared__ T16 lds[2 * 64 * LDM];
+  T16* ldsDyT = lds;            // [64 k][LDM m]
+  T16* ldsXT = lds + 64 * LDM;  // [64 kg][LDM m]
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const int k0 = blockIdx.x * 64;
+  const int cchunks = CI / BK;
+  const int c0 = (blockIdx.y % cchunks) * BK;
+  const int s_ = (blockIdx.y / cchunks) % S;
+  const int r_ = blockIdx.y / (cchunks * S);
+  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+
+  // staging assignment: 128 threads per tile, 4 m-rows x 8 elems each
+  const int st_x = tid >= 128;           // which tile this thread stages
+  const int t = tid & 127;
+  const int sm = (t & 15) * 4;           // m offset (4 rows)
+  const int sk = (t >> 4) * 8;           // element offset within the 64-row
+  T16* ldsT = st_x ? ldsXT : ldsDyT;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int i0 = (wave & 1) * 32;   // KO half
+  const int j0 = (wave >> 1) * 32;  // kg half
+
+  f32x16 acc = {};
+
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
+    // ---- gather + transpose-stage 4 m-rows of 8 elems ----
+    short8 v[4];
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const long m = m0 + sm + mi;
+      bool ok = m < m_end;
+      long off = 0;
+      if (ok) {
+        if (st_x) {
+          const int n_ = (int)(m / ((long)Ho * Wo));
+          const int pq = (int)(m % ((long)Ho * Wo));
+          const int ih = (pq / Wo) * stride - pad + r_;
+          const int iw = (pq % Wo) * stride - pad + s_;
+          ok = (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
+          if (ok) off = (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
+        } else {
+          off = m * KO + k0 + sk;
+        }
+      }
+      v[mi] = ok ? *reinterpret_cast<const short8*>((st_x ? x : dy) + off)
+                 : short8{};
+    }
+    __syncthreads();  // previous MFMA phase done
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
+      *reinterpret_cast<short4v*>(
+          reinterpret_cast<short*>(ldsT + (sk + e) * LDM + sm)) = pk;
+    }
+    __syncthreads();
+    // ---- MFMA over the 64-m step ----
+#pragma unroll
+    for (int kk = 0; kk < WGM; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsDyT + (i0 + li) * LDM + kk + kh * 8);
+      const short8 bf = *reinterpret_cast<const short8*>(
+          ldsXT + (j0 + li) * LDM + kk + kh * 8);
+      acc = Mfma32<T16>::run(af, bf, acc);
+    }
+  }
+
+  // ---- scatter the 32x32 fp32 tile ----
+  const long kg_base = (long)(r_ * S + s_) * CI + c0;
+  const long RSC = (long)R * S * CI;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;  // KO row
+    float* p = dw + (long)(k0 + i0 + i) * RSC + kg_base + j0 + li;
+    if (nchunks == 1)
+      *p = acc[reg];
+    else
+      atomicAdd(p, acc[reg]);
+  }
+}
+
+}  // namespace
+
+bool conv_mfma_supported(long CI, long KO) {
+  return CI % 64 == 0 && KO % 64 == 0;
+}
+
+void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+                            long R, long S, long stride, long pad) {
+  const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
+  const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
+  const long M = (long)N * Ho * Wo;
+  const long blocks_xy = (KO / 64) * (R * S * (CI / 64));
+  int nchunks = (int)std::min<long>(std::max<long>(2048 / blocks_xy, 1),
+                                    cdiv_l(M, 512));
+  nchunks = std::max(nchunks, 1);
+  long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
+  nchunks = (int)cdiv_l(M, m_per_chunk);
+  dim3 grid(KO / 64, R * S * (CI / 64), nchunks);
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL(conv_wgrad_mfma_kernel<T16>, grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)dy.data_ptr(), dw.data_ptr<float>(), N, Hi,
+                       Wi, CI, KO, Ho, Wo, (int)R, (int)S, (int)stride,
+                       (int)pad, m_per_chunk, nchunks);
+  });
+}
+
+// fwd: in = x[N,Hi,Wi,CI], wgt = w16 [KO, R,S,CI] row-major
+void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                          at::Tensor y, long stride, long pad, long act) {
+  const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
+  const int KO = w.size(0), R = w.size(1), S = w.size(2);
+  const int Ho = y.size(1), Wo = y.size(2);
+  const long M = (long)N * Ho * Wo;
+  dim3 grid((unsigned)cdiv_l(M, BM), KO / BN);
+  const int has_bias = bias.numel() > 0;
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL((conv_gather_gemm<T16, false>), grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)w.data_ptr(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
+                       (int)stride, (int)pad, (long)R * S * CI, (long)CI,
+                       (int)act, has_bias);
+  });
+}
+
+// dgrad: in = dy[N,P,Q,KO], wflip = [R,S,CI,KO] (w[k,R-1-r,S-1-s,c]),
+// out = dx[N,H,W,CI]
+void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
+                            long R, long S, long stride, long pad) {
+  const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
+  const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
+  const long M = (long)N * H * W;
+  dim3 grid((unsigned)cdiv_l(M, BM), CI / BN);
+  auto empty_bias = at::Tensor();
+  DISPATCH_16(dy, T16, {
+    hipLaunchKernelGGL((conv_gather_gemm<T16, true>), grid, dim3(256), 0,
+                       cur_stream(), (const T16*)dy.data_ptr(),
+                       (const T16*)wflip.data_ptr(), nullptr,
+                       (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
+                       (int)S, (int)stride, (int)pad, (long)KO,
+                       (long)CI * KO, 0, 0);
+  });
+}

