@@ -1,0 +1,97 @@
+"""One-trip GPU kernel debug harness: per-op max-diff breakdown + r50
+repro. Run: python benchmarks/debug_kernels.py [probe|conv|r50|all]"""
+
+import sys
+
+import torch
+
+import mi355x.ops as ops
+from mi355x.ops import functional as fn
+
+
+def q(t):
+    return t.to(torch.bfloat16).float()
+
+
+def md(a, b):
+    return (a.float().cpu() - b.float()).abs().max().item()
+
+
+def probe():
+    g = torch.Generator().manual_seed(11)
+    A = q(torch.randn(32, 16, generator=g))
+    B = q(torch.randn(16, 32, generator=g))
+    D = ops.ext().mfma_probe32(A.cuda(), B.cuda())
+    print("mfma probe maxdiff vs A@B:", md(D, A @ B))
+    I = torch.zeros(32, 16)
+    I[:16, :16] = torch.eye(16)
+    D2 = ops.ext().mfma_probe32(I.cuda(), B.cuda())
+    print("mfma identity rows maxdiff:", md(D2[:16], B),
+          " lower-rows-zero:", D2[16:].abs().max().item())
+    torch.cuda.synchronize()
+
+
+CASES = [
+    ((2, 32, 32, 3), 6, 5, 1, 0),
+    ((2, 14, 14, 6), 16, 5, 1, 0),
+    ((2, 32, 32, 16), 32, 3, 1, 1),
+    ((2, 16, 16, 32), 64, 3, 2, 1),
+    ((2, 8, 8, 64), 64, 3, 1, 1),
+    ((2, 8, 8, 128), 128, 1, 1, 0),
+    ((3, 7, 7, 64), 192, 3, 2, 1),
+    ((4, 16, 16, 64), 128, 3, 2, 1),
+    ((2, 9, 9, 256), 512, 1, 2, 0),
+]
+
+
+def conv_cases():
+    for shape, K, ksz, stride, pad in CASES:
+        C = shape[-1]
+        g = torch.Generator().manual_seed(0)
+        x = q(torch.randn(*shape, generator=g))
+        w = q(torch.randn(K, C, ksz, ksz, generator=g) * 0.2)
+        xc = x.clone().requires_grad_(True)
+        xg = x.cuda().to(torch.bfloat16).requires_grad_(True)
+        wc = w.clone().requires_grad_(True)
+        wg = w.cuda().requires_grad_(True)
+        yc = fn.conv2d(xc, wc, None, stride, pad, None)
+        yg = fn.conv2d(xg, wg, None, stride, pad, None)
+        dy = q(torch.randn(*yc.shape, generator=g))
+        yc.backward(dy)
+        yg.backward(dy.cuda().to(torch.bfloat16))
+        torch.cuda.synchronize()
+        print(f"conv C={C} K={K} k={ksz} s={stride} p={pad}: "
+              f"fwd={md(yg, yc):.4f} dx={md(xg.grad, xc.grad):.4f} "
+              f"dw={md(wg.grad, wc.grad):.4f}")
+
+
+def r50():
+    from mi355x import optim
+    from mi355x.models import build_model
+    from mi355x.parallel.flat import FlatState
+
+    torch.manual_seed(0)
+    net = build_model("resnet50").cuda()
+    flat = FlatState(net)
+    opt = optim.SGD(flat, lr=0.01, momentum=0.9)
+    x = torch.randn(4, 3, 64, 64, device="cuda")
+    y = torch.randint(0, 1000, (4,), device="cuda")
+    for i in range(3):
+        opt.zero_grad()
+        loss = fn.cross_entropy(net(x), y)
+        print("fwd done", i, flush=True)
+        loss.backward()
+        torch.cuda.synchronize()
+        print("bwd done", i, flush=True)
+        opt.step()
+        print("r50 step", i, "loss", loss.item(), flush=True)
+
+
+if __name__ == "__main__":
+    what = sys.argv[1] if len(sys.argv) > 1 else "all"
+    if what in ("probe", "all"):
+        probe()
+    if what in ("conv", "all"):
+        conv_cases()
+    if what in ("r50", "all"):
+        r50()

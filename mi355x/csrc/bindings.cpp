@@ -4,6 +4,7 @@
 #include <vector>
 
 at::Tensor relu_bwd(at::Tensor dy, at::Tensor y);
+at::Tensor mfma_probe32(at::Tensor A, at::Tensor B);
 void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr, double mu,
               double wd, double gscale);
 at::Tensor cast_to_16(at::Tensor src, at::Tensor like);
@@ -40,8 +41,12 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
                              at::Tensor lse, double scale);
 
+void register_rccl(py::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_rccl(m);
   m.def("relu_bwd", &relu_bwd);
+  m.def("mfma_probe32", &mfma_probe32);
   m.def("sgd_step", &sgd_step);
   m.def("cast_to_16", &cast_to_16);
   m.def("conv2d_fwd", &conv2d_fwd);

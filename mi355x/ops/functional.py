@@ -59,15 +59,15 @@ def _w16_conv(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
 
 
 def _w16_conv_flip(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
-    """fp32 [K,C,R,S] parameter -> cached 16-bit [R,S,C,K] spatially-flipped
-    copy (wflip[r,s,c,k] = w[k,R-1-r,S-1-s,c]) — the dgrad MFMA kernel's
-    B-operand layout (transposed implicit GEMM, conv_mfma.hip)."""
+    """fp32 [K,C,R,S] parameter -> cached 16-bit [R,S,C,K] transposed copy
+    (wt[r,s,c,k] = w[k,c,r,s]) — the dgrad MFMA kernel's B-operand layout.
+    No spatial flip: the kernel's transposed-window gather pairs source row
+    (h+pad-r)/stride directly with weight tap r (conv_mfma.hip)."""
     key = (weight._version, _cache_epoch, dtype)
     cache = getattr(weight, "_mi355x_wflip", None)
     if cache is not None and cache[0] == key:
         return cache[1]
-    wf = (weight.detach().to(dtype).permute(2, 3, 1, 0).flip(0, 1)
-          .contiguous())
+    wf = weight.detach().to(dtype).permute(2, 3, 1, 0).contiguous()
     weight._mi355x_wflip = (key, wf)
     return wf
 

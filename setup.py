@@ -19,10 +19,12 @@ SRC = [
     "mi355x/csrc/elementwise.hip",
     "mi355x/csrc/conv.hip",
     "mi355x/csrc/conv_mfma.hip",
+    "mi355x/csrc/mfma_probe.hip",
     "mi355x/csrc/bn.hip",
     "mi355x/csrc/pool.hip",
     "mi355x/csrc/gemm.hip",
     "mi355x/csrc/loss.hip",
+    "mi355x/csrc/rccl_comm.cpp",
 ]
 
 setup(
@@ -37,6 +39,7 @@ setup(
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17"],
             },
+            libraries=["rccl"],
         )
     ],
     cmdclass={"build_ext": BuildExtension.with_options(no_python_abi_suffix=False)},

@@ -14,11 +14,18 @@ RTOL, ATOL = 3e-2, 3e-2
 
 def _pair(shape, requires_grad=False, seed=0):
     g = torch.Generator().manual_seed(seed)
-    cpu = torch.randn(*shape, generator=g)
-    gpu = cpu.to("cuda").to(torch.bfloat16)
+    gpu = torch.randn(*shape, generator=g).to("cuda").to(torch.bfloat16)
+    # the fp32 reference runs on the SAME bf16-quantized values the kernel
+    # sees, so only accumulation-order rounding separates the two paths
+    cpu = gpu.float().cpu()
     cpu.requires_grad_(requires_grad)
     gpu.requires_grad_(requires_grad)
     return cpu, gpu
+
+
+def _qt(t):
+    """bf16-quantize an fp32 reference tensor (what the kernel consumes)."""
+    return t.to(torch.bfloat16).float()
 
 
 def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
@@ -41,7 +48,7 @@ def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
     C = shape[-1]
     xc, xg = _pair(shape, requires_grad=True)
     g = torch.Generator().manual_seed(1)
-    w = torch.randn(K, C, ksz, ksz, generator=g) * 0.2
+    w = _qt(torch.randn(K, C, ksz, ksz, generator=g) * 0.2)
     b = torch.randn(K, generator=g) * 0.1 if bias else None
     wc = w.clone().requires_grad_(True)
     wg = w.clone().cuda().requires_grad_(True)
@@ -79,7 +86,7 @@ def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
 def test_linear_fwd_bwd(M, N, K, bias, act):
     xc, xg = _pair((M, K), requires_grad=True)
     g = torch.Generator().manual_seed(2)
-    w = torch.randn(N, K, generator=g) * 0.1
+    w = _qt(torch.randn(N, K, generator=g) * 0.1)
     b = torch.randn(N, generator=g) * 0.1 if bias else None
     wc = w.clone().requires_grad_(True)
     wg = w.clone().cuda().requires_grad_(True)
@@ -243,3 +250,18 @@ def test_fp16_dtype_path():
         w = torch.randn(16, 16, 3, 3) * 0.2
         y = fn.conv2d(x, w.cuda(), None, 1, 1, None)
         assert y.dtype == torch.float16
+
+
+def test_mfma_fragment_layout():
+    """Asymmetric A/B elementwise check of the 32x32x16 fragment maps the
+    conv/gemm kernels assume (guide §3: symmetric inputs hide transposes)."""
+    g = torch.Generator().manual_seed(11)
+    A = _qt(torch.randn(32, 16, generator=g))
+    B = _qt(torch.randn(16, 32, generator=g))
+    D = ops.ext().mfma_probe32(A.cuda(), B.cuda())
+    torch.testing.assert_close(D.cpu(), A @ B, rtol=1e-2, atol=1e-2)
+    # identity checks pin the exact row/col mapping
+    I16 = torch.zeros(32, 16)
+    I16[:16, :16] = torch.eye(16)
+    D2 = ops.ext().mfma_probe32(I16.cuda(), B.cuda())
+    torch.testing.assert_close(D2.cpu()[:16], B, rtol=1e-2, atol=1e-2)
