@@ -1,7 +1,10 @@
 """One-trip GPU kernel debug harness: per-op max-diff breakdown + r50
 repro. Run: python benchmarks/debug_kernels.py [probe|conv|r50|all]"""
 
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
@@ -87,11 +90,50 @@ def r50():
         print("r50 step", i, "loss", loss.item(), flush=True)
 
 
+def r50_fwd_steps():
+    """Step r50 forward submodule-by-submodule with syncs to localize a
+    device fault."""
+    from mi355x.models import build_model
+    from mi355x.models.layers import to_model_layout
+
+    def ck(tag, h):
+        torch.cuda.synchronize()
+        print("ok:", tag, tuple(h.shape), flush=True)
+        return h
+
+    torch.manual_seed(0)
+    net = build_model("resnet50").cuda()
+    net.train()
+    x = torch.randn(4, 3, 64, 64, device="cuda")
+    h = ck("layout", to_model_layout(x))
+    h = ck("conv1", net.conv1(h))
+    h = ck("bn1", net.bn1(h))
+    h = ck("maxpool", net.maxpool(h))
+    for name in ["layer1", "layer2", "layer3", "layer4"]:
+        layer = getattr(net, name)
+        for i, blk in enumerate(layer):
+            ident = h if blk.downsample is None else None
+            if blk.downsample is not None:
+                ident = ck(f"{name}.{i}.ds.conv", blk.downsample.conv(h))
+                ident = ck(f"{name}.{i}.ds.bn", blk.downsample.bn(ident))
+            o = ck(f"{name}.{i}.conv1", blk.conv1(h))
+            o = ck(f"{name}.{i}.bn1", blk.bn1(o))
+            o = ck(f"{name}.{i}.conv2", blk.conv2(o))
+            o = ck(f"{name}.{i}.bn2", blk.bn2(o))
+            o = ck(f"{name}.{i}.conv3", blk.conv3(o))
+            h = ck(f"{name}.{i}.bn3", blk.bn3(o, residual=ident))
+    h = ck("gap", ops.global_avg_pool(h))
+    h = ck("fc", net.fc(h))
+    print("forward complete", flush=True)
+
+
 if __name__ == "__main__":
     what = sys.argv[1] if len(sys.argv) > 1 else "all"
     if what in ("probe", "all"):
         probe()
     if what in ("conv", "all"):
         conv_cases()
+    if what == "r50fwd":
+        r50_fwd_steps()
     if what in ("r50", "all"):
         r50()

@@ -8,14 +8,21 @@ c10d::Reducer does for the reference script
   rank 0 (the reference broadcasts per-bucket), plus the module buffers
   (BN running stats);
 - per-parameter post-accumulate-grad hooks mark bucket readiness during
-  backward; a complete bucket launches an async RCCL all-reduce on the
-  bucket's contiguous flat-grad slice, overlapped with the remaining
-  backward (xGMI is 7 point-to-point links/GPU, so bucket size is a
-  first-class tunable — MI355X_BUCKET_MB env or ctor arg);
-- gradients are reduced as SUMs; the 1/world_size averaging is folded into
-  the fused SGD step (grad_scale) instead of a separate pass.
+  backward; complete buckets launch async all-reduces on contiguous
+  flat-grad slices, overlapped with the remaining backward. Buckets are
+  launched in FIXED index order on every rank (a bucket launches only
+  after all lower-index buckets have) so collectives can never interleave
+  differently across ranks — the same invariant torch's reducer enforces;
+- xGMI is 7 point-to-point links/GPU, so bucket size is a first-class
+  tunable (MI355X_BUCKET_MB env or ctor arg);
+- gradients are reduced as SUMs; the 1/world_size averaging is folded
+  into the fused SGD step (grad_scale) instead of a separate pass.
 
-Works identically over the gloo backend on CPU (multi-process CPU tests).
+Comm backends: on GPU the default is our native RCCL communicator
+(mi355x/parallel/rccl.py + csrc/rccl_comm.cpp) on its own high-priority
+HIP stream; MI355X_COMM=torch selects torch.distributed collectives
+instead, and CPU/gloo always uses torch.distributed (multi-process CPU
+tests run the identical reducer logic).
 """
 
 from __future__ import annotations
@@ -29,11 +36,47 @@ from torch import nn
 from .flat import FlatState
 
 
+class _TorchComm:
+    def __init__(self, process_group):
+        self.pg = process_group
+
+    def broadcast(self, t, root=0):
+        dist.broadcast(t, src=root, group=self.pg)
+
+    def all_reduce_async(self, t):
+        return dist.all_reduce(t, async_op=True, group=self.pg)
+
+    def finish(self, works):
+        for w in works:
+            w.wait()
+
+
+class _NativeComm:
+    def __init__(self):
+        from .rccl import native_comm
+
+        self.c = native_comm()
+
+    def broadcast(self, t, root=0):
+        if t.is_cuda:
+            self.c.broadcast(t.contiguous(), root)
+            self.c.wait()
+        else:  # module buffers may live on CPU in odd setups
+            dist.broadcast(t, src=root)
+
+    def all_reduce_async(self, t):
+        self.c.all_reduce(t)
+        return None
+
+    def finish(self, works):
+        self.c.wait()
+
+
 class DistributedDataParallel(nn.Module):
     def __init__(self, module: nn.Module, flat: FlatState | None = None,
                  bucket_mb: float | None = None,
                  first_bucket_mb: float = 1.0,
-                 process_group=None):
+                 process_group=None, comm: str | None = None):
         super().__init__()
         if bucket_mb is None:
             bucket_mb = float(os.environ.get("MI355X_BUCKET_MB", "25"))
@@ -46,8 +89,14 @@ class DistributedDataParallel(nn.Module):
         )
         self.world_size = dist.get_world_size(process_group) if dist.is_initialized() else 1
         self.grad_scale = 1.0 / self.world_size
+        comm = comm or os.environ.get("MI355X_COMM", "rccl")
+        use_native = (comm == "rccl"
+                      and self.flat.flat_param.is_cuda
+                      and self.world_size > 1)
+        self.comm = _NativeComm() if use_native else _TorchComm(process_group)
         self._works: list = []
         self._ready: dict[int, int] = {}
+        self._launched = 0
         self._hooks = []
         if self.world_size > 1:
             self._broadcast_initial_state()
@@ -58,22 +107,29 @@ class DistributedDataParallel(nn.Module):
 
     # -- init-time sync ----------------------------------------------------
     def _broadcast_initial_state(self):
-        dist.broadcast(self.flat.flat_param, src=0, group=self.process_group)
+        self.comm.broadcast(self.flat.flat_param, 0)
         for buf in self.module.buffers():
-            dist.broadcast(buf, src=0, group=self.process_group)
+            self.comm.broadcast(buf, 0)
 
     # -- per-iteration machinery -------------------------------------------
     def _reset_bucket_state(self):
         self._ready = {b.index: 0 for b in self.flat.buckets}
+        self._launched = 0
         self._works = []
 
     def _mark_ready(self, param: torch.Tensor):
         b = self.flat.bucket_of[id(param)]
         self._ready[b.index] += 1
-        if self._ready[b.index] == len(b.params):
-            w = dist.all_reduce(self.flat.grad_slice(b), async_op=True,
-                                group=self.process_group)
-            self._works.append(w)
+        # launch complete buckets in fixed index order (cross-rank safety)
+        buckets = self.flat.buckets
+        while (self._launched < len(buckets)
+               and self._ready[self._launched]
+               == len(buckets[self._launched].params)):
+            w = self.comm.all_reduce_async(
+                self.flat.grad_slice(buckets[self._launched]))
+            if w is not None:
+                self._works.append(w)
+            self._launched += 1
 
     def forward(self, *args, **kwargs):
         if self.world_size > 1:
@@ -84,8 +140,10 @@ class DistributedDataParallel(nn.Module):
         """Wait for all in-flight bucket all-reduces (call before the
         optimizer step). Gradients are left as SUMS over ranks; consume
         self.grad_scale in the optimizer (or scale explicitly)."""
-        for w in self._works:
-            w.wait()
+        if self.world_size > 1:
+            assert self._launched == len(self.flat.buckets), \
+                "backward did not produce grads for every bucket"
+            self.comm.finish(self._works)
         self._works = []
 
     # -- passthroughs: checkpoint key parity with torch DDP ("module." prefix,
