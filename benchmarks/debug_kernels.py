@@ -74,7 +74,7 @@ def r50():
     from mi355x.parallel.flat import FlatState
 
     torch.manual_seed(0)
-    net = build_model("resnet50").cuda()
+    net = build_model("resnet50", num_classes=1000).cuda()
     flat = FlatState(net)
     opt = optim.SGD(flat, lr=0.01, momentum=0.9)
     x = torch.randn(4, 3, 64, 64, device="cuda")

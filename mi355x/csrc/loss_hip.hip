@@ -28,7 +28,10 @@ __global__ void ce_fwd_kernel(const T16* __restrict__ logits,
   if (lane == 0) {
     const float l = mx + __logf(s);
     lse[b] = l;
-    loss[b] = l - F16<T16>::to_f32(row[target[b]]);
+    const long t = target[b];
+    // out-of-range target: poison the loss (NaN) instead of faulting
+    loss[b] = (t >= 0 && t < C) ? l - F16<T16>::to_f32(row[t])
+                                : __builtin_nanf("");
   }
 }
 

@@ -49,8 +49,8 @@ def test_resnet18_trains_on_gpu():
 
 def test_resnet50_trains_on_gpu():
     torch.manual_seed(0)
-    losses = _train(build_model("resnet50"), batch=4, size=64, classes=1000,
-                    steps=4, lr=0.01)
+    losses = _train(build_model("resnet50", num_classes=1000), batch=4,
+                    size=64, classes=1000, steps=4, lr=0.01)
     assert all(l == l for l in losses), losses
 
 
