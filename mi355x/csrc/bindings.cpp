@@ -39,7 +39,7 @@ at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy);
 std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
                                           at::Tensor target);
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
-                             at::Tensor lse, double scale);
+                             at::Tensor lse, at::Tensor dloss);
 
 void register_rccl(py::module_& m);
 

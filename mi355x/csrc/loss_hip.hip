@@ -39,8 +39,10 @@ template <typename T16>
 __global__ void ce_bwd_kernel(const T16* __restrict__ logits,
                               const long* __restrict__ target,
                               const float* __restrict__ lse,
+                              const float* __restrict__ dloss,  // 0-dim, device
                               T16* __restrict__ dlogits, int B, int C,
-                              float scale) {
+                              float inv_b) {
+  const float scale = (dloss ? *dloss : 1.f) * inv_b;
   const long total = (long)B * C;
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
@@ -71,18 +73,21 @@ std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
   return {loss.mean(), lse};
 }
 
+// dloss: 0-dim fp32 CUDA tensor (the upstream grad) read ON DEVICE so the
+// backward is hipGraph-capturable (no host .item() sync); empty => 1.0.
 at::Tensor cross_entropy_bwd(at::Tensor logits, at::Tensor target,
-                             at::Tensor lse, double scale) {
+                             at::Tensor lse, at::Tensor dloss) {
   CHECK_GPU(logits);
   const int B = logits.size(0), C = logits.size(1);
   auto dlogits = at::empty_like(logits);
   const long total = (long)B * C;
   const int grid = (int)std::min<long>(cdiv_l(total, 256), 2048);
+  const float* dl = dloss.numel() ? dloss.data_ptr<float>() : nullptr;
   DISPATCH_16(logits, T16, {
     hipLaunchKernelGGL(ce_bwd_kernel<T16>, dim3(grid), dim3(256), 0,
                        cur_stream(), (const T16*)logits.data_ptr(),
-                       target.data_ptr<long>(), lse.data_ptr<float>(),
-                       (T16*)dlogits.data_ptr(), B, C, (float)scale);
+                       target.data_ptr<long>(), lse.data_ptr<float>(), dl,
+                       (T16*)dlogits.data_ptr(), B, C, 1.f / B);
   });
   return dlogits;
 }

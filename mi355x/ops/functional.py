@@ -332,8 +332,12 @@ class _CEFn(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dloss):
         logits, target, lse = ctx.saved_tensors
-        scale = float(dloss) / logits.shape[0]
-        dlogits = ext().cross_entropy_bwd(logits, target, lse, scale)
+        # dloss read on-device (hipGraph-capturable: no host sync)
+        if not (torch.is_tensor(dloss) and dloss.is_cuda):
+            dloss = torch.as_tensor(dloss, dtype=torch.float32,
+                                    device=logits.device)
+        dlogits = ext().cross_entropy_bwd(logits, target, lse,
+                                          dloss.float().reshape(()))
         return dlogits, None
 
 
