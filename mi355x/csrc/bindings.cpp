@@ -8,6 +8,8 @@ at::Tensor mfma_probe32(at::Tensor A, at::Tensor B);
 void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr, double mu,
               double wd, double gscale);
 at::Tensor cast_to_16(at::Tensor src, at::Tensor like);
+at::Tensor cast_permute_krsc(at::Tensor w, at::Tensor like);
+at::Tensor cast_permute_rsck(at::Tensor w, at::Tensor like);
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act);
@@ -20,11 +22,13 @@ at::Tensor bn_stats(at::Tensor x);
 at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                     at::Tensor gamma, at::Tensor beta, at::Tensor res,
                     long act);
-at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor mean,
-                         at::Tensor invstd);
-at::Tensor bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor mean,
-                     at::Tensor invstd, at::Tensor gamma, at::Tensor dgamma,
-                     at::Tensor dbeta, double m_total);
+at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
+                         at::Tensor mean, at::Tensor invstd);
+std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
+                                  at::Tensor mean, at::Tensor invstd,
+                                  at::Tensor gamma, at::Tensor dgamma,
+                                  at::Tensor dbeta, double m_total,
+                                  bool want_dres);
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
                                     long pad);
@@ -49,6 +53,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe32", &mfma_probe32);
   m.def("sgd_step", &sgd_step);
   m.def("cast_to_16", &cast_to_16);
+  m.def("cast_permute_krsc", &cast_permute_krsc);
+  m.def("cast_permute_rsck", &cast_permute_rsck);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);

@@ -56,9 +56,12 @@ __global__ void bn_apply_kernel(const T16* __restrict__ x,
   }
 }
 
+// y != null: the forward ended in ReLU; mask dy by y>0 inline (saves a
+// standalone relu_bwd pass over the activation tensor)
 template <typename T16>
 __global__ void bn_bwd_reduce_kernel(const T16* __restrict__ x,
                                      const T16* __restrict__ dy,
+                                     const T16* __restrict__ y,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ invstd,
                                      float* __restrict__ out,  // [2,C]
@@ -70,7 +73,8 @@ __global__ void bn_bwd_reduce_kernel(const T16* __restrict__ x,
   const float mu = mean[c], is = invstd[c];
   float s_dyx = 0.f, s_dy = 0.f;
   for (long m = m0; m < m1; ++m) {
-    const float d = F16<T16>::to_f32(dy[m * C + c]);
+    float d = F16<T16>::to_f32(dy[m * C + c]);
+    if (y && F16<T16>::to_f32(y[m * C + c]) <= 0.f) d = 0.f;
     const float xh = (F16<T16>::to_f32(x[m * C + c]) - mu) * is;
     s_dyx += d * xh;
     s_dy += d;
@@ -84,27 +88,111 @@ __global__ void bn_bwd_reduce_kernel(const T16* __restrict__ x,
   }
 }
 
+// y != null: mask dy by y>0 inline; dres != null: also emit the masked dy
+// (the residual-branch gradient) so no separate relu_bwd pass is needed.
 template <typename T16>
 __global__ void bn_bwd_dx_kernel(const T16* __restrict__ x,
                                  const T16* __restrict__ dy,
+                                 const T16* __restrict__ y,
                                  const float* __restrict__ mean,
                                  const float* __restrict__ invstd,
                                  const float* __restrict__ gamma,
                                  const float* __restrict__ dgamma,
                                  const float* __restrict__ dbeta,
-                                 T16* __restrict__ dx, long M, int C,
+                                 T16* __restrict__ dx,
+                                 T16* __restrict__ dres, long M, int C,
                                  float inv_m) {
   const long total = M * C;
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const int c = (int)(t % C);
     const float mu = mean[c], is = invstd[c];
-    const float d = F16<T16>::to_f32(dy[t]);
+    float d = F16<T16>::to_f32(dy[t]);
+    if (y && F16<T16>::to_f32(y[t]) <= 0.f) d = 0.f;
+    if (dres) dres[t] = F16<T16>::from_f32(d);
     const float xh = (F16<T16>::to_f32(x[t]) - mu) * is;
     const float v =
         gamma[c] * is * (d - dbeta[c] * inv_m - xh * dgamma[c] * inv_m);
     dx[t] = F16<T16>::from_f32(v);
   }
+}
+
+// ---- fast reduction path: C % 8 == 0 && 2048 % C == 0 -------------------
+// Linear sweep of the [M,C] tensor in 16 B/lane vectors (fully coalesced,
+// every thread active); each thread owns a FIXED 8-channel group (c0 =
+// (tid*8) % C, invariant because 2048 % C == 0), accumulates 8 partial
+// pairs in registers, combines within the block through an LDS f32
+// accumulator, then one global atomicAdd per channel per block.
+// BWD=false: (sum x, sum x^2). BWD=true: (sum dy*xhat, sum dy), with the
+// fused ReLU mask from y when given.
+template <typename T16, bool BWD>
+__global__ __launch_bounds__(256) void bn_reduce_fast(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    const T16* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ invstd, float* __restrict__ out,  // [2,C]
+    long E, int C, long e_per_block) {
+  extern __shared__ __attribute__((aligned(16))) float lsum[];  // [2*C]
+  const int tid = threadIdx.x;
+  const int c0 = (tid * 8) % C;
+  float s0[8] = {}, s1[8] = {};
+  float mu[8], is[8];
+  if (BWD) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      mu[u] = mean[c0 + u];
+      is[u] = invstd[c0 + u];
+    }
+  }
+  const long e0 = (long)blockIdx.x * e_per_block;
+  const long e1 = min(E, e0 + e_per_block);
+  for (long e = e0 + (long)tid * 8; e < e1; e += 256 * 8) {
+    if (!BWD) {
+      const short8 vx = *reinterpret_cast<const short8*>(x + e);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const float v = s16_to_f32<T16>(vx[u]);
+        s0[u] += v;
+        s1[u] += v * v;
+      }
+    } else {
+      const short8 vx = *reinterpret_cast<const short8*>(x + e);
+      const short8 vd = *reinterpret_cast<const short8*>(dy + e);
+      short8 vy = {};
+      if (y) vy = *reinterpret_cast<const short8*>(y + e);
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        float d = s16_to_f32<T16>(vd[u]);
+        if (y && s16_to_f32<T16>(vy[u]) <= 0.f) d = 0.f;
+        const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
+        s0[u] += d * xh;
+        s1[u] += d;
+      }
+    }
+  }
+  for (int c = tid; c < 2 * C; c += 256) lsum[c] = 0.f;
+  __syncthreads();
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    atomicAdd(lsum + c0 + u, s0[u]);
+    atomicAdd(lsum + C + c0 + u, s1[u]);
+  }
+  __syncthreads();
+  for (int c = tid; c < 2 * C; c += 256)
+    if (gridDim.x == 1)
+      out[c] = lsum[c];
+    else
+      atomicAdd(out + c, lsum[c]);
+}
+
+inline bool bn_fast_ok(long M, int C) {
+  return C % 8 == 0 && 2048 % C == 0;
+}
+
+inline dim3 bn_fast_grid(long E, long& e_per_block) {
+  long blocks = std::min<long>(cdiv_l(E, 2048), 1024);
+  e_per_block = cdiv_l(cdiv_l(E, blocks), 2048) * 2048;
+  blocks = cdiv_l(E, e_per_block);
+  return dim3((unsigned)blocks);
 }
 
 inline dim3 chan_grid(long M, int C, long& m_per_chunk, int block = 256) {
@@ -128,6 +216,20 @@ at::Tensor bn_stats(at::Tensor x) {
   CHECK_16BIT(x);
   const int C = x.size(-1);
   const long M = x.numel() / C;
+  if (bn_fast_ok(M, C)) {
+    const long E = M * (long)C;
+    long e_per_block;
+    dim3 grid = bn_fast_grid(E, e_per_block);
+    auto out = grid.x == 1 ? at::empty({2, C}, x.options().dtype(at::kFloat))
+                           : at::zeros({2, C}, x.options().dtype(at::kFloat));
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((bn_reduce_fast<T16, false>), grid, dim3(256),
+                         2 * C * sizeof(float), cur_stream(),
+                         (const T16*)x.data_ptr(), nullptr, nullptr, nullptr,
+                         nullptr, out.data_ptr<float>(), E, C, e_per_block);
+    });
+    return out;
+  }
   long m_per_chunk;
   dim3 grid = chan_grid(M, C, m_per_chunk);
   auto out = grid.x == 1 ? at::empty({2, C}, x.options().dtype(at::kFloat))
@@ -161,12 +263,28 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
   return y;
 }
 
-at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor mean,
-                         at::Tensor invstd) {
+at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
+                         at::Tensor mean, at::Tensor invstd) {
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
   const long M = x.numel() / C;
+  if (bn_fast_ok(M, C)) {
+    const long E = M * (long)C;
+    long e_per_block;
+    dim3 grid = bn_fast_grid(E, e_per_block);
+    auto out = grid.x == 1 ? at::empty({2, C}, x.options().dtype(at::kFloat))
+                           : at::zeros({2, C}, x.options().dtype(at::kFloat));
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((bn_reduce_fast<T16, true>), grid, dim3(256),
+                         2 * C * sizeof(float), cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)dy.data_ptr(),
+                         y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         out.data_ptr<float>(), E, C, e_per_block);
+    });
+    return out;
+  }
   long m_per_chunk;
   dim3 grid = chan_grid(M, C, m_per_chunk);
   auto out = grid.x == 1 ? at::empty({2, C}, x.options().dtype(at::kFloat))
@@ -174,28 +292,37 @@ at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor mean,
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(bn_bwd_reduce_kernel<T16>, grid, dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       (const T16*)dy.data_ptr(),
+                       y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), out.data_ptr<float>(), M, C,
                        m_per_chunk);
   });
   return out;
 }
 
-at::Tensor bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor mean,
-                     at::Tensor invstd, at::Tensor gamma, at::Tensor dgamma,
-                     at::Tensor dbeta, double m_total) {
+std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
+                                  at::Tensor mean, at::Tensor invstd,
+                                  at::Tensor gamma, at::Tensor dgamma,
+                                  at::Tensor dbeta, double m_total,
+                                  bool want_dres) {
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
   const long M = x.numel() / C;
   auto dx = at::empty_like(x);
+  auto dres = want_dres ? at::empty_like(x) : at::Tensor();
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(bn_bwd_dx_kernel<T16>, dim3(ew_grid2(M * C)),
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)dy.data_ptr(), mean.data_ptr<float>(),
+                       (const T16*)dy.data_ptr(),
+                       y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                       mean.data_ptr<float>(),
                        invstd.data_ptr<float>(), gamma.data_ptr<float>(),
                        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                       (T16*)dx.data_ptr(), M, C, (float)(1.0 / m_total));
+                       (T16*)dx.data_ptr(),
+                       want_dres ? (T16*)dres.data_ptr() : nullptr, M, C,
+                       (float)(1.0 / m_total));
   });
-  return dx;
+  return {dx, want_dres ? dres : at::Tensor()};
 }

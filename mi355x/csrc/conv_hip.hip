@@ -113,21 +113,31 @@ __global__ void conv_wgrad_direct(const T16* __restrict__ x,
     const int r = (int)((t / ((long)C * S)) % R);
     const int k = (int)(t / ((long)C * S * R));
     float acc = 0.f;
+    // incremental (n,p,q) decode: one div/mod at entry, carries after
+    int q = (int)(m0 % Q);
+    long np0 = m0 / Q;
+    int p = (int)(np0 % P);
+    int n = (int)(np0 / P);
     for (long m = m0; m < m1; ++m) {
-      const int q = (int)(m % Q);
-      long np = m / Q;
-      const int p = (int)(np % P);
-      const int n = (int)(np / P);
       const int ih = p * stride - pad + r;
       const int iw = q * stride - pad + s;
-      if (ih < 0 || ih >= H || iw < 0 || iw >= W) continue;
-      acc += F16<T16>::to_f32(dy[m * K + k]) *
-             F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * C + c]);
+      if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+        acc += F16<T16>::to_f32(dy[m * K + k]) *
+               F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * C + c]);
+      if (++q == Q) {
+        q = 0;
+        if (++p == P) {
+          p = 0;
+          ++n;
+        }
+      }
     }
+    // store in the parameter layout [K,C,R,S]
+    float* out = dw + ((long)k * C + c) * ((long)R * S) + (long)r * S + s;
     if (nchunks == 1)
-      dw[t] = acc;
+      *out = acc;
     else
-      atomicAdd(dw + t, acc);
+      atomicAdd(out, acc);
   }
 }
 
@@ -211,19 +221,19 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
   const long M = (long)N * P * Q;
   const long total_w = (long)K * R * S * C;
   if (conv_mfma_supported(C, K)) {
-    auto dw = at::zeros({K, R, S, (long)C}, x.options().dtype(at::kFloat));
+    auto dw = at::zeros({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad);
     return dw;
   }
   // split the NPQ reduction so small filters still fill the chip
-  int nchunks = (int)std::min<long>(cdiv_l(M, 4096),
-                                    std::max<long>(1, (256L * 2048) / std::max(total_w, 1L)));
+  int nchunks = (int)std::min<long>(cdiv_l(M, 1024),
+                                    std::max<long>(1, (256L * 8192) / std::max(total_w, 1L)));
   nchunks = std::max(nchunks, 1);
   const long m_per_chunk = cdiv_l(M, nchunks);
   auto dw = nchunks == 1
-                ? at::empty({K, R, S, (long)C},
+                ? at::empty({K, (long)C, R, S},
                             x.options().dtype(at::kFloat))
-                : at::zeros({K, R, S, (long)C}, x.options().dtype(at::kFloat));
+                : at::zeros({K, (long)C, R, S}, x.options().dtype(at::kFloat));
   dim3 grid((unsigned)std::min<long>(cdiv_l(total_w, 256), 4096), nchunks);
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(conv_wgrad_direct<T16>, grid, dim3(256), 0,

@@ -272,13 +272,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     }
   }
 
-  // ---- scatter the 32x32 fp32 tile ----
-  const long kg_base = (long)(r_ * S + s_) * CI + c0;
-  const long RSC = (long)R * S * CI;
+  // ---- scatter the 32x32 fp32 tile, directly in the parameter layout
+  // [KO, CI, R, S] (no host-side permute; contiguous across lanes for the
+  // 1x1 convs where R*S == 1) ----
+  const int rs = r_ * S + s_;
+  const long RS = (long)R * S;
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;  // KO row
-    float* p = dw + (long)(k0 + i0 + i) * RSC + kg_base + j0 + li;
+    const int c_abs = c0 + j0 + li;
+    float* p = dw + ((long)(k0 + i0 + i) * CI + c_abs) * RS + rs;
     if (nchunks == 1)
       *p = acc[reg];
     else

@@ -53,7 +53,11 @@ def _w16_conv(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     cache = getattr(weight, "_mi355x_w16", None)
     if cache is not None and cache[0] == key:
         return cache[1]
-    w16 = weight.detach().to(dtype).permute(0, 2, 3, 1).contiguous()
+    if weight.is_cuda:
+        w16 = ext().cast_permute_krsc(
+            weight.detach(), torch.empty(0, dtype=dtype, device=weight.device))
+    else:
+        w16 = weight.detach().to(dtype).permute(0, 2, 3, 1).contiguous()
     weight._mi355x_w16 = (key, w16)
     return w16
 
@@ -67,7 +71,11 @@ def _w16_conv_flip(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     cache = getattr(weight, "_mi355x_wflip", None)
     if cache is not None and cache[0] == key:
         return cache[1]
-    wf = weight.detach().to(dtype).permute(2, 3, 1, 0).contiguous()
+    if weight.is_cuda:
+        wf = ext().cast_permute_rsck(
+            weight.detach(), torch.empty(0, dtype=dtype, device=weight.device))
+    else:
+        wf = weight.detach().to(dtype).permute(2, 3, 1, 0).contiguous()
     weight._mi355x_wflip = (key, wf)
     return wf
 
@@ -119,8 +127,9 @@ class _ConvFn(torch.autograd.Function):
                                     x.shape[1], x.shape[2])
         dw = None
         if ctx.needs_input_grad[1]:
-            dw_krsc = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride, padding)
-            dw = dw_krsc.permute(0, 3, 1, 2).contiguous()  # fp32 [K,C,R,S]
+            # wgrad kernels emit fp32 [K,C,R,S] directly (parameter layout)
+            dw = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride,
+                                    padding)
         db = None
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 1, 2))
@@ -225,17 +234,23 @@ class _BNFn(torch.autograd.Function):
         x, y, mean, invstd, gamma = ctx.saved_tensors
         act, has_res, m_total, pg = ctx.conf
         dy = dy.contiguous()
-        if act == _ACT_RELU:
-            dy = _relu_mask_bwd(dy, y)
-        r = ext().bn_bwd_reduce(x, dy, mean, invstd)  # f32 [2,C]: sum_dy_xhat, sum_dy
+        # ReLU mask fused into both backward kernels (no standalone pass)
+        y_arg = y if act == _ACT_RELU else torch.empty(0, device=x.device,
+                                                       dtype=x.dtype)
+        r = ext().bn_bwd_reduce(x, dy, y_arg, mean, invstd)  # [2,C]
         if pg is not None:
             import torch.distributed as dist
 
             dist.all_reduce(r, group=pg)
         dgamma, dbeta = r[0], r[1]
-        dx = ext().bn_bwd_dx(x, dy, mean, invstd, gamma.detach().float(),
-                             dgamma, dbeta, float(m_total))
-        dres = dy if has_res else None
+        want_dres = has_res and act == _ACT_RELU
+        dx, dres = ext().bn_bwd_dx(x, dy, y_arg, mean, invstd,
+                                   gamma.detach().float(), dgamma, dbeta,
+                                   float(m_total), want_dres)
+        if has_res and not want_dres:
+            dres = dy  # no activation: residual grad is dy itself
+        elif not has_res:
+            dres = None
         return dx, dgamma, dbeta, dres, None, None, None, None, None
 
 

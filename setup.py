@@ -20,6 +20,7 @@ SRC = [
     "mi355x/csrc/conv.hip",
     "mi355x/csrc/conv_mfma.hip",
     "mi355x/csrc/mfma_probe.hip",
+    "mi355x/csrc/wcast.hip",
     "mi355x/csrc/bn.hip",
     "mi355x/csrc/pool.hip",
     "mi355x/csrc/gemm.hip",
