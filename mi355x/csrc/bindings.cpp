@@ -9,10 +9,11 @@ void sgd_step(at::Tensor p, at::Tensor g, at::Tensor m, double lr, double mu,
               double wd, double gscale);
 at::Tensor cast_to_16(at::Tensor src, at::Tensor like);
 at::Tensor cast_permute_krsc(at::Tensor w, at::Tensor like);
+at::Tensor cast_permute_krsc_pad(at::Tensor w, at::Tensor like);
 at::Tensor cast_permute_rsck(at::Tensor w, at::Tensor like);
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
-                      long stride, long pad, long act);
+                      long stride, long pad, long act, long kR, long kS);
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wflip,
                         long stride, long pad, long H, long W);
 at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
@@ -54,6 +55,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step);
   m.def("cast_to_16", &cast_to_16);
   m.def("cast_permute_krsc", &cast_permute_krsc);
+  m.def("cast_permute_krsc_pad", &cast_permute_krsc_pad);
   m.def("cast_permute_rsck", &cast_permute_rsck);
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);

@@ -154,22 +154,31 @@ static inline int out_dim(int in, int k, int stride, int pad) {
 bool conv_mfma_supported(long CI, long KO);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act);
+void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
+                               at::Tensor y, long R, long S, long stride,
+                               long pad, long act);
 void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                             long R, long S, long stride, long pad);
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                             long R, long S, long stride, long pad);
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
-                      long stride, long pad, long act) {
+                      long stride, long pad, long act, long kR, long kS) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
   CHECK_CONTIG(w);
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
-  const int K = w.size(0), R = w.size(1), S = w.size(2);
-  TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
+  const int K = w.size(0);
+  const int R = w.dim() == 4 ? w.size(1) : kR;
+  const int S = w.dim() == 4 ? w.size(2) : kS;
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
   auto y = at::empty({N, P, Q, K}, x.options());
+  if (w.dim() == 2) {  // padded [KO,KGP] weight: generic small-C MFMA path
+    conv_fwd_mfma_genc_launch(x, w, bias, y, R, S, stride, pad, act);
+    return y;
+  }
+  TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
   if (conv_mfma_supported(C, K)) {
     conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act);
     return y;

@@ -51,9 +51,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long b_row_stride, const long b_rs_stride,
     const int act, const int has_bias) {
-  __shared__ T16 lds[BM * LDK + BN * LDK];
-  T16* ldsA = lds;
-  T16* ldsB = lds + BM * LDK;
+  // double-buffered: one barrier per k-step; iter j's staging writes go to
+  // buffer (j+1)&1 while every wave MFMAs from buffer j&1
+  __shared__ T16 lds[2 * (BM * LDK + BN * LDK)];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -129,20 +129,28 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     sb[1] = *reinterpret_cast<const short8*>(wp + 8);
   };
 
-  load_step(0);
-  for (int j = 0; j < ksteps; ++j) {
-    __syncthreads();  // previous MFMA phase done reading LDS
-    {
-      short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
+  auto stage = [&](int buf) {
+    T16* ldsA = lds + buf * (BM * LDK + BN * LDK);
+    T16* ldsB = ldsA + BM * LDK;
+    short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
-      short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
-      *reinterpret_cast<short8*>(pb) = sb[0];
-      *reinterpret_cast<short8*>(pb + 8) = sb[1];
-    }
-    __syncthreads();
-    if (j + 1 < ksteps) load_step(j + 1);  // HBM latency hides under MFMA
+    for (int i = 0; i < 4; ++i)
+      *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
+    short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
+    *reinterpret_cast<short8*>(pb) = sb[0];
+    *reinterpret_cast<short8*>(pb + 8) = sb[1];
+  };
+
+  load_step(0);
+  stage(0);
+  __syncthreads();
+  for (int j = 0; j < ksteps; ++j) {
+    if (j + 1 < ksteps) {
+      load_step(j + 1);   // HBM latency hides under this step's MFMA
+      stage((j + 1) & 1); // other buffer: safe since the barrier below
+    }                     // ordered iter j-1's reads before these writes
+    const T16* ldsA = lds + (j & 1) * (BM * LDK + BN * LDK);
+    const T16* ldsB = ldsA + BM * LDK;
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -154,6 +162,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       acc0 = Mfma32<T16>::run(af, bf0, acc0);
       acc1 = Mfma32<T16>::run(af, bf1, acc1);
     }
+    __syncthreads();
   }
 
   // ---- epilogue: bias + act + store ----
@@ -201,9 +210,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long m_per_chunk, const int nchunks) {
-  __shared__ T16 lds[2 * 64 * LDM];
-  T16* ldsDyT = lds;            // [64 k][LDM m]
-  T16* ldsXT = lds + 64 * LDM;  // [64 kg][LDM m]
+  // double-buffered ([dyT | xT] per buffer): one barrier per m-step
+  __shared__ T16 lds[2 * 2 * 64 * LDM];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -220,7 +228,6 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   const int t = tid & 127;
   const int sm = (t & 15) * 4;           // m offset (4 rows)
   const int sk = (t >> 4) * 8;           // element offset within the 64-row
-  T16* ldsT = st_x ? ldsXT : ldsDyT;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -231,9 +238,32 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
 
   f32x16 acc = {};
 
-  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
-    // ---- gather + transpose-stage 4 m-rows of 8 elems ----
-    short8 v[4];
+  // incremental (n,p,q) decode for the x-gather: one div/mod at entry,
+  // add-with-carry as m advances by WGM per step (divisions in the inner
+  // loop were ~40% of this kernel's time)
+  int dn = 0, dp = 0, dq = 0;
+  if (st_x) {
+    const long m_first = m_begin + sm;
+    dn = (int)(m_first / ((long)Ho * Wo));
+    const int pq = (int)(m_first % ((long)Ho * Wo));
+    dp = pq / Wo;
+    dq = pq % Wo;
+  }
+  auto advance = [&](int by) {
+    dq += by;
+    while (dq >= Wo) {
+      dq -= Wo;
+      if (++dp == Ho) {
+        dp = 0;
+        ++dn;
+      }
+    }
+  };
+
+  // gather + pack 4 m-rows of 8 elems into registers for m-step m0
+  short8 v[4];
+  auto load_m = [&](long m0) {
+    int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const long m = m0 + sm + mi;
@@ -241,10 +271,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
       long off = 0;
       if (ok) {
         if (st_x) {
-          const int n_ = (int)(m / ((long)Ho * Wo));
-          const int pq = (int)(m % ((long)Ho * Wo));
-          const int ih = (pq / Wo) * stride - pad + r_;
-          const int iw = (pq % Wo) * stride - pad + s_;
+          const int ih = p_ * stride - pad + r_;
+          const int iw = q_ * stride - pad + s_;
           ok = (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
           if (ok) off = (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
         } else {
@@ -253,16 +281,38 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
       }
       v[mi] = ok ? *reinterpret_cast<const short8*>((st_x ? x : dy) + off)
                  : short8{};
+      if (st_x && mi < 3 && ++q_ == Wo) {
+        q_ = 0;
+        if (++p_ == Ho) {
+          p_ = 0;
+          ++n_;
+        }
+      }
     }
-    __syncthreads();  // previous MFMA phase done
+    if (st_x) advance(WGM);  // position for the NEXT m-step's gather
+  };
+  // transpose-write the 4x8 register patch into buffer `buf`
+  auto stage_m = [&](int buf) {
+    T16* ldsT = lds + buf * (2 * 64 * LDM) + (st_x ? 64 * LDM : 0);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
       *reinterpret_cast<short4v*>(
           reinterpret_cast<short*>(ldsT + (sk + e) * LDM + sm)) = pk;
     }
-    __syncthreads();
-    // ---- MFMA over the 64-m step ----
+  };
+
+  load_m(m_begin);
+  stage_m(0);
+  __syncthreads();
+  int buf = 0;
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM, buf ^= 1) {
+    if (m0 + WGM < m_end) {
+      load_m(m0 + WGM);
+      stage_m(buf ^ 1);
+    }
+    const T16* ldsDyT = lds + buf * (2 * 64 * LDM);
+    const T16* ldsXT = ldsDyT + 64 * LDM;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -271,6 +321,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
           ldsXT + (j0 + li) * LDM + kk + kh * 8);
       acc = Mfma32<T16>::run(af, bf, acc);
     }
+    __syncthreads();
   }
 
   // ---- scatter the 32x32 fp32 tile, directly in the parameter layout
