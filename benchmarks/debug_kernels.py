@@ -44,6 +44,9 @@ CASES = [
     ((3, 7, 7, 64), 192, 3, 2, 1),
     ((4, 16, 16, 64), 128, 3, 2, 1),
     ((2, 9, 9, 256), 512, 1, 2, 0),
+    ((2, 32, 32, 3), 64, 3, 1, 1),
+    ((2, 32, 32, 3), 64, 7, 2, 3),
+    ((2, 10, 10, 24), 64, 3, 1, 1),
 ]
 
 

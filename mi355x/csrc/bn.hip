@@ -168,12 +168,26 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
       }
     }
   }
+  // wave-level pre-reduction: lanes whose 8-channel group repeats within
+  // the wave (group stride C/8 lanes) fold via shfl_xor before touching
+  // LDS — cuts LDS-atomic collisions by 512/C
+  const int gstride = C / 8;  // lanes between same-group threads
+  for (int off = 32; off >= gstride && off >= 1; off >>= 1) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      s0[u] += __shfl_xor(s0[u], off, kWave);
+      s1[u] += __shfl_xor(s1[u], off, kWave);
+    }
+  }
+  const bool leader = (threadIdx.x & 63) < gstride || gstride >= 64;
   for (int c = tid; c < 2 * C; c += 256) lsum[c] = 0.f;
   __syncthreads();
+  if (leader) {
 #pragma unroll
-  for (int u = 0; u < 8; ++u) {
-    atomicAdd(lsum + c0 + u, s0[u]);
-    atomicAdd(lsum + C + c0 + u, s1[u]);
+    for (int u = 0; u < 8; ++u) {
+      atomicAdd(lsum + c0 + u, s0[u]);
+      atomicAdd(lsum + C + c0 + u, s1[u]);
+    }
   }
   __syncthreads();
   for (int c = tid; c < 2 * C; c += 256)

@@ -57,14 +57,16 @@ __global__ void conv_fwd_direct(const T16* __restrict__ x,
   }
 }
 
+// weights in the transposed [R,S,C,K] layout (wt[r,s,c,k] = w[k,c,r,s]):
+// the k-loop reads are contiguous per lane (vectorized when K % 8 == 0)
 template <typename T16>
 __global__ void conv_dgrad_direct(const T16* __restrict__ dy,
-                                  const T16* __restrict__ w,
+                                  const T16* __restrict__ wt,
                                   T16* __restrict__ dx, int N, int H, int W,
                                   int C, int K, int R, int S, int P, int Q,
                                   int stride, int pad) {
   const long total = (long)N * H * W * C;
-  const long wkstride = (long)R * S * C;
+  const bool kvec = (K % 8 == 0);
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const int c = (int)(t % C);
@@ -85,10 +87,19 @@ __global__ void conv_dgrad_direct(const T16* __restrict__ dy,
         const int q = qw / stride;
         if (q >= Q) continue;
         const T16* dyp = dy + (((long)n * P + p) * Q + q) * K;
-        const T16* wp = w + (long)(r * S + s) * C + c;
-        for (int k = 0; k < K; ++k)
-          acc += F16<T16>::to_f32(dyp[k]) *
-                 F16<T16>::to_f32(wp[(long)k * wkstride]);
+        const T16* wp = wt + ((long)(r * S + s) * C + c) * K;
+        if (kvec) {
+          for (int k = 0; k < K; k += 8) {
+            short8 dv = *reinterpret_cast<const short8*>(dyp + k);
+            short8 wv = *reinterpret_cast<const short8*>(wp + k);
+#pragma unroll
+            for (int u = 0; u < 8; ++u)
+              acc += s16_to_f32<T16>(dv[u]) * s16_to_f32<T16>(wv[u]);
+          }
+        } else {
+          for (int k = 0; k < K; ++k)
+            acc += F16<T16>::to_f32(dyp[k]) * F16<T16>::to_f32(wp[k]);
+        }
       }
     }
     dx[t] = F16<T16>::from_f32(acc);
@@ -155,22 +166,31 @@ static inline int out_dim(int in, int k, int stride, int pad) {
 bool conv_mfma_supported(long CI, long KO);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act);
+void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
+                               at::Tensor y, long R, long S, long stride,
+                               long pad, long act);
 void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                             long R, long S, long stride, long pad);
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                             long R, long S, long stride, long pad);
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
-                      long stride, long pad, long act) {
+                      long stride, long pad, long act, long kR, long kS) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
   CHECK_CONTIG(w);
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
-  const int K = w.size(0), R = w.size(1), S = w.size(2);
-  TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
+  const int K = w.size(0);
+  const int R = w.dim() == 4 ? w.size(1) : kR;
+  const int S = w.dim() == 4 ? w.size(2) : kS;
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
   auto y = at::empty({N, P, Q, K}, x.options());
+  if (w.dim() == 2) {  // padded [KO,KGP] weight: generic small-C MFMA path
+    conv_fwd_mfma_genc_launch(x, w, bias, y, R, S, stride, pad, act);
+    return y;
+  }
+  TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
   if (conv_mfma_supported(C, K)) {
     conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act);
     return y;
@@ -188,15 +208,15 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   return y;
 }
 
-at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wflip,
-                        long stride, long pad, long H, long W) {
+at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
+                        long pad, long H, long W) {
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
   CHECK_16BIT(dy);
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), K = dy.size(3);
-  const int R = w.size(1), S = w.size(2), C = w.size(3);
+  const int R = wflip.size(0), S = wflip.size(1), C = wflip.size(2);
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
-  if (wflip.numel() > 0 && conv_mfma_supported(K, C)) {
+  if (conv_mfma_supported(K, C)) {
     conv_dgrad_mfma_launch(dy, wflip, dx, R, S, stride, pad);
     return dx;
   }
@@ -204,7 +224,7 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wflip,
   DISPATCH_16(dy, T16, {
     hipLaunchKernelGGL(conv_dgrad_direct<T16>, conv_grid(total), dim3(256), 0,
                        cur_stream(), (const T16*)dy.data_ptr(),
-                       (const T16*)w.data_ptr(), (T16*)dx.data_ptr(), N,
+                       (const T16*)wflip.data_ptr(), (T16*)dx.data_ptr(), N,
                        (int)H, (int)W, C, K, R, S, P, Q, (int)stride,
                        (int)pad);
   });

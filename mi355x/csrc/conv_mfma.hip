@@ -53,9 +53,10 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long b_row_stride, const long b_rs_stride,
     const int act, const int has_bias) {
-  // double-buffered: one barrier per k-step; iter j's staging writes go to
-  // buffer (j+1)&1 while every wave MFMAs from buffer j&1
-  __shared__ T16 lds[2 * (BM * LDK + BN * LDK)];
+  // single-buffered: 2x LDS for a double buffer measurably LOSES here —
+  // it halves blocks/CU and the cross-block overlap it sacrifices was
+  // already hiding the staging latency (guide common-mistake #5)
+  __shared__ T16 lds[BM * LDK + BN * LDK];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -170,9 +171,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     sb[1] = *reinterpret_cast<const short8*>(wp + 8);
   };
 
-  auto stage = [&](int buf) {
-    T16* ldsA = lds + buf * (BM * LDK + BN * LDK);
-    T16* ldsB = ldsA + BM * LDK;
+  auto stage = [&]() {
+    T16* ldsA = lds;
+    T16* ldsB = lds + BM * LDK;
     short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -183,15 +184,13 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   };
 
   load_step(0);
-  stage(0);
-  __syncthreads();
   for (int j = 0; j < ksteps; ++j) {
-    if (j + 1 < ksteps) {
-      load_step(j + 1);   // HBM latency hides under this step's MFMA
-      stage((j + 1) & 1); // other buffer: safe since the barrier below
-    }                     // ordered iter j-1's reads before these writes
-    const T16* ldsA = lds + (j & 1) * (BM * LDK + BN * LDK);
-    const T16* ldsB = ldsA + BM * LDK;
+    __syncthreads();  // previous MFMA phase done reading LDS
+    stage();
+    __syncthreads();
+    if (j + 1 < ksteps) load_step(j + 1);  // overlaps the MFMA phase
+    const T16* ldsA = lds;
+    const T16* ldsB = lds + BM * LDK;
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -203,7 +202,6 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       acc0 = Mfma32<T16>::run(af, bf0, acc0);
       acc1 = Mfma32<T16>::run(af, bf1, acc1);
     }
-    __syncthreads();
   }
 
   // ---- epilogue: bias + act + store ----
@@ -251,8 +249,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long m_per_chunk, const int nchunks) {
-  // double-buffered ([dyT | xT] per buffer): one barrier per m-step
-  __shared__ T16 lds[2 * 2 * 64 * LDM];
+  __shared__ T16 lds[2 * 64 * LDM];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -332,9 +329,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     }
     if (st_x) advance(WGM);  // position for the NEXT m-step's gather
   };
-  // transpose-write the 4x8 register patch into buffer `buf`
-  auto stage_m = [&](int buf) {
-    T16* ldsT = lds + buf * (2 * 64 * LDM) + (st_x ? 64 * LDM : 0);
+  // transpose-write the 4x8 register patch
+  auto stage_m = [&]() {
+    T16* ldsT = lds + (st_x ? 64 * LDM : 0);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
@@ -344,16 +341,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   };
 
   load_m(m_begin);
-  stage_m(0);
-  __syncthreads();
-  int buf = 0;
-  for (long m0 = m_begin; m0 < m_end; m0 += WGM, buf ^= 1) {
-    if (m0 + WGM < m_end) {
-      load_m(m0 + WGM);
-      stage_m(buf ^ 1);
-    }
-    const T16* ldsDyT = lds + buf * (2 * 64 * LDM);
-    const T16* ldsXT = ldsDyT + 64 * LDM;
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
+    __syncthreads();
+    stage_m();
+    __syncthreads();
+    if (m0 + WGM < m_end) load_m(m0 + WGM);
+    const T16* ldsDyT = lds;
+    const T16* ldsXT = lds + 64 * LDM;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -362,7 +356,6 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
           ldsXT + (j0 + li) * LDM + kk + kh * 8);
       acc = Mfma32<T16>::run(af, bf, acc);
     }
-    __syncthreads();
   }
 
   // ---- scatter the 32x32 fp32 tile, directly in the parameter layout

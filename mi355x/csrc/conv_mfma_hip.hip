@@ -41,7 +41,10 @@ constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
 
 // TRANS=false (fwd): src row = ho*stride - pad + r, in [0,Hi)
 // TRANS=true (dgrad): src row = (ho + pad - r), valid iff %stride==0, /stride in [0,Hi)
-template <typename T16, bool TRANS>
+// GENC (fwd only): CI % 64 != 0 — the weight is pre-padded to
+// [KO, KGP=ceil(R*S*CI/64)*64] with kg=(r*S+s)*CI+c and zeros beyond, and
+// the A side gathers per-ELEMENT across tap boundaries (stem convs C=3/6).
+template <typename T16, bool TRANS, bool GENC = false>
 __global__ __launch_bounds__(256) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
@@ -51,9 +54,10 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long b_row_stride, const long b_rs_stride,
     const int act, const int has_bias) {
-  // double-buffered: one barrier per k-step; iter j's staging writes go to
-  // buffer (j+1)&1 while every wave MFMAs from buffer j&1
-  __shared__ T16 lds[2 * (BM * LDK + BN * LDK)];
+  // single-buffered: 2x LDS for a double buffer measurably LOSES here —
+  // it halves blocks/CU and the cross-block overlap it sacrifices was
+  // already hiding the staging latency (guide common-mistake #5)
+  __shared__ T16 lds[BM * LDK + BN * LDK];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -90,13 +94,52 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
 
   f32x16 acc0 = {}, acc1 = {};
 
-  const int cchunks = CI / BK;
-  const int ksteps = R * S * cchunks;
+  const int cchunks = GENC ? 1 : CI / BK;
+  const int ksteps = GENC ? (int)(b_row_stride / BK)  // KGP/64
+                          : R * S * cchunks;
 
   short8 sa[4];  // 32 channels = 4 x 16B
   short8 sb[2];  // 16 elements = 2 x 16B
 
+  auto load_step_genc = [&](int j) {
+    // per-element gather: kg walks (tap, c) with carries (fully unrolled,
+    // so the vector-register indices stay compile-time — guide rule 20)
+    const int kg0 = j * BK + sa_c;
+    int tap = kg0 / CI;
+    int c = kg0 - tap * CI;
+    int r_ = tap / S;
+    int s_ = tap - r_ * S;
+#pragma unroll
+    for (int i = 0; i < 32; ++i) {
+      short val = 0;
+      if (m_ok && tap < R * S) {
+        const int ih = ih0 + r_;
+        const int iw = iw0 + s_;
+        if ((unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi) {
+          const T16 v = in[(((long)n_ * Hi + ih) * Wi + iw) * CI + c];
+          __builtin_memcpy(&val, &v, 2);
+        }
+      }
+      sa[i / 8][i % 8] = val;
+      if (++c == CI) {
+        c = 0;
+        ++tap;
+        if (++s_ == S) {
+          s_ = 0;
+          ++r_;
+        }
+      }
+    }
+    const T16* wp = wgt + (long)(k0 + sb_n) * b_row_stride + j * BK + sb_c;
+    sb[0] = *reinterpret_cast<const short8*>(wp);
+    sb[1] = *reinterpret_cast<const short8*>(wp + 8);
+  };
+
   auto load_step = [&](int j) {
+    if constexpr (GENC) {
+      load_step_genc(j);
+      return;
+    }
     const int c0 = (j % cchunks) * BK;
     const int s_ = (j / cchunks) % S;
     const int r_ = j / (cchunks * S);
@@ -129,9 +172,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     sb[1] = *reinterpret_cast<const short8*>(wp + 8);
   };
 
-  auto stage = [&](int buf) {
-    T16* ldsA = lds + buf * (BM * LDK + BN * LDK);
-    T16* ldsB = ldsA + BM * LDK;
+  auto stage = [&]() {
+    T16* ldsA = lds;
+    T16* ldsB = lds + BM * LDK;
     short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -142,15 +185,13 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   };
 
   load_step(0);
-  stage(0);
-  __syncthreads();
   for (int j = 0; j < ksteps; ++j) {
-    if (j + 1 < ksteps) {
-      load_step(j + 1);   // HBM latency hides under this step's MFMA
-      stage((j + 1) & 1); // other buffer: safe since the barrier below
-    }                     // ordered iter j-1's reads before these writes
-    const T16* ldsA = lds + (j & 1) * (BM * LDK + BN * LDK);
-    const T16* ldsB = ldsA + BM * LDK;
+    __syncthreads();  // previous MFMA phase done reading LDS
+    stage();
+    __syncthreads();
+    if (j + 1 < ksteps) load_step(j + 1);  // overlaps the MFMA phase
+    const T16* ldsA = lds;
+    const T16* ldsB = lds + BM * LDK;
 #pragma unroll
     for (int kk = 0; kk < BK; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -162,7 +203,6 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       acc0 = Mfma32<T16>::run(af, bf0, acc0);
       acc1 = Mfma32<T16>::run(af, bf1, acc1);
     }
-    __syncthreads();
   }
 
   // ---- epilogue: bias + act + store ----
@@ -210,8 +250,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long m_per_chunk, const int nchunks) {
-  // double-buffered ([dyT | xT] per buffer): one barrier per m-step
-  __shared__ T16 lds[2 * 2 * 64 * LDM];
+  __shared__ T16 lds[2 * 64 * LDM];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -291,9 +330,9 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     }
     if (st_x) advance(WGM);  // position for the NEXT m-step's gather
   };
-  // transpose-write the 4x8 register patch into buffer `buf`
-  auto stage_m = [&](int buf) {
-    T16* ldsT = lds + buf * (2 * 64 * LDM) + (st_x ? 64 * LDM : 0);
+  // transpose-write the 4x8 register patch
+  auto stage_m = [&]() {
+    T16* ldsT = lds + (st_x ? 64 * LDM : 0);
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
       short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
@@ -303,16 +342,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   };
 
   load_m(m_begin);
-  stage_m(0);
-  __syncthreads();
-  int buf = 0;
-  for (long m0 = m_begin; m0 < m_end; m0 += WGM, buf ^= 1) {
-    if (m0 + WGM < m_end) {
-      load_m(m0 + WGM);
-      stage_m(buf ^ 1);
-    }
-    const T16* ldsDyT = lds + buf * (2 * 64 * LDM);
-    const T16* ldsXT = ldsDyT + 64 * LDM;
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
+    __syncthreads();
+    stage_m();
+    __syncthreads();
+    if (m0 + WGM < m_end) load_m(m0 + WGM);
+    const T16* ldsDyT = lds;
+    const T16* ldsXT = lds + 64 * LDM;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
@@ -321,7 +357,6 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
           ldsXT + (j0 + li) * LDM + kk + kh * 8);
       acc = Mfma32<T16>::run(af, bf, acc);
     }
-    __syncthreads();
   }
 
   // ---- scatter the 32x32 fp32 tile, directly in the parameter layout
@@ -365,6 +400,28 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                        (const T16*)dy.data_ptr(), dw.data_ptr<float>(), N, Hi,
                        Wi, CI, KO, Ho, Wo, (int)R, (int)S, (int)stride,
                        (int)pad, m_per_chunk, nchunks);
+  });
+}
+
+// GENC fwd: wpad = [KO, KGP] zero-padded (cast_permute_krsc_pad)
+void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
+                               at::Tensor y, long R, long S, long stride,
+                               long pad, long act) {
+  const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
+  const int KO = wpad.size(0);
+  const long KGP = wpad.size(1);
+  const int Ho = y.size(1), Wo = y.size(2);
+  const long M = (long)N * Ho * Wo;
+  dim3 grid((unsigned)cdiv_l(M, BM), KO / BN);
+  const int has_bias = bias.numel() > 0;
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL((conv_gather_gemm<T16, false, true>), grid, dim3(256),
+                       0, cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)wpad.data_ptr(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, (int)R,
+                       (int)S, (int)stride, (int)pad, KGP, 0, (int)act,
+                       has_bias);
   });
 }
 

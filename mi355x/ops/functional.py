@@ -54,8 +54,13 @@ def _w16_conv(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     if cache is not None and cache[0] == key:
         return cache[1]
     if weight.is_cuda:
-        w16 = ext().cast_permute_krsc(
-            weight.detach(), torch.empty(0, dtype=dtype, device=weight.device))
+        like = torch.empty(0, dtype=dtype, device=weight.device)
+        K, C = weight.shape[0], weight.shape[1]
+        if C % 64 != 0 and K % 64 == 0:
+            # generic small-C MFMA path wants [KO, KGP] zero-padded
+            w16 = ext().cast_permute_krsc_pad(weight.detach(), like)
+        else:
+            w16 = ext().cast_permute_krsc(weight.detach(), like)
     else:
         w16 = weight.detach().to(dtype).permute(0, 2, 3, 1).contiguous()
     weight._mi355x_w16 = (key, w16)
@@ -107,7 +112,8 @@ class _ConvFn(torch.autograd.Function):
         dtype = x.dtype
         w16 = _w16_conv(weight, dtype)
         b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
-        y = ext().conv2d_fwd(x, w16, b32, stride, padding, act)
+        y = ext().conv2d_fwd(x, w16, b32, stride, padding, act,
+                             weight.shape[2], weight.shape[3])
         ctx.save_for_backward(x, w16, y)
         ctx.weight_ref = weight  # for the cached dgrad weight-flip
         ctx.conf = (stride, padding, act, bias is not None, weight.shape)
@@ -123,7 +129,7 @@ class _ConvFn(torch.autograd.Function):
         dx = None
         if ctx.needs_input_grad[0]:
             wflip = _w16_conv_flip(ctx.weight_ref, dy.dtype)
-            dx = ext().conv2d_dgrad(dy, w16, wflip, stride, padding,
+            dx = ext().conv2d_dgrad(dy, wflip, stride, padding,
                                     x.shape[1], x.shape[2])
         dw = None
         if ctx.needs_input_grad[1]:

@@ -43,6 +43,9 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 8, 8, 64), 64, 3, 1, 1, False, "relu"),  # MFMA path shape
     ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
+    ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
+    ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # GENC 7x7/2 stem
+    ((2, 10, 10, 24), 64, 3, 1, 1, False, None),   # GENC C=24
 ])
 def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
     C = shape[-1]
