@@ -34,7 +34,7 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch", type=int,
-                    default=int(os.environ.get("MI355X_BENCH_BATCH", "256")),
+                    default=int(os.environ.get("MI355X_BENCH_BATCH", "1024")),
                     help="per-GPU batch size")
     ap.add_argument("--model", default=os.environ.get("MI355X_BENCH_MODEL",
                                                       "resnet18"))

@@ -130,12 +130,34 @@ def r50_fwd_steps():
     print("forward complete", flush=True)
 
 
+def wgradperf():
+    """Per-variant wgrad timing (layer1 shape) with hip events."""
+    import mi355x.ops as O
+    torch.manual_seed(0)
+    N, H, W, C, K = 256, 32, 32, 64, 64
+    x = torch.randn(N, H, W, C, device="cuda").to(torch.bfloat16)
+    dy = torch.randn(N, H, W, K, device="cuda").to(torch.bfloat16)
+    for _ in range(3):
+        O.ext().conv2d_wgrad(x, dy, 3, 3, 1, 1)
+    torch.cuda.synchronize()
+    import time
+    t0 = time.perf_counter()
+    for _ in range(20):
+        O.ext().conv2d_wgrad(x, dy, 3, 3, 1, 1)
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / 20 * 1e6
+    gf = 2 * N * H * W * 64 * 64 * 9 / 1e9
+    print(f"wgrad layer1: {dt:.1f} us/call  ({gf / dt * 1e3:.0f} TF)")
+
+
 if __name__ == "__main__":
     what = sys.argv[1] if len(sys.argv) > 1 else "all"
     if what in ("probe", "all"):
         probe()
     if what in ("conv", "all"):
         conv_cases()
+    if what == "wgradperf":
+        wgradperf()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):

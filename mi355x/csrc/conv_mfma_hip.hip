@@ -388,8 +388,13 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   const long M = (long)N * Ho * Wo;
   const long blocks_xy = (KO / 64) * (R * S * (CI / 64));
+  static const long cap = [] {  // ablation knob: MI355X_WGRAD_CHUNKS
+    const char* e = getenv("MI355X_WGRAD_CHUNKS");
+    return e ? atol(e) : 0L;
+  }();
   int nchunks = (int)std::min<long>(std::max<long>(2048 / blocks_xy, 1),
                                     cdiv_l(M, 512));
+  if (cap > 0) nchunks = (int)std::min<long>(nchunks, cap);
   nchunks = std::max(nchunks, 1);
   long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
   nchunks = (int)cdiv_l(M, m_per_chunk);
