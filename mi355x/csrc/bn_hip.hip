@@ -198,6 +198,78 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
       atomicAdd(out + c, lsum[c]);
 }
 
+// vectorized elementwise BN apply / backward-dx: 16 B/lane linear sweep,
+// per-thread FIXED 8-channel group (same condition as bn_reduce_fast)
+template <typename T16>
+__global__ __launch_bounds__(256) void bn_apply_fast(
+    const T16* __restrict__ x, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const T16* __restrict__ res,
+    T16* __restrict__ y, long E, int C, int act) {
+  const int c0 = ((blockIdx.x * blockDim.x + threadIdx.x) * 8) % C;
+  float sc[8], sh[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    sc[u] = gamma[c0 + u] * invstd[c0 + u];
+    sh[u] = beta[c0 + u] - mean[c0 + u] * sc[u];
+  }
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; e < E;
+       e += stride) {
+    const short8 vx = *reinterpret_cast<const short8*>(x + e);
+    short8 vr = {};
+    if (res) vr = *reinterpret_cast<const short8*>(res + e);
+    short8 o;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float v = s16_to_f32<T16>(vx[u]) * sc[u] + sh[u];
+      if (res) v += s16_to_f32<T16>(vr[u]);
+      if (act == 1) v = fmaxf(v, 0.f);
+      o[u] = f32_to_s16<T16>(v);
+    }
+    *reinterpret_cast<short8*>(y + e) = o;
+  }
+}
+
+template <typename T16>
+__global__ __launch_bounds__(256) void bn_bwd_dx_fast(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    const T16* __restrict__ y, const float* __restrict__ mean,
+    const float* __restrict__ invstd, const float* __restrict__ gamma,
+    const float* __restrict__ dgamma, const float* __restrict__ dbeta,
+    T16* __restrict__ dx, T16* __restrict__ dres, long E, int C,
+    float inv_m) {
+  const int c0 = ((blockIdx.x * blockDim.x + threadIdx.x) * 8) % C;
+  float mu[8], is[8], g_[8], a_[8], b_[8];
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    mu[u] = mean[c0 + u];
+    is[u] = invstd[c0 + u];
+    g_[u] = gamma[c0 + u] * is[u];
+    a_[u] = dgamma[c0 + u] * inv_m;
+    b_[u] = dbeta[c0 + u] * inv_m;
+  }
+  const long stride = (long)gridDim.x * blockDim.x * 8;
+  for (long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; e < E;
+       e += stride) {
+    const short8 vx = *reinterpret_cast<const short8*>(x + e);
+    const short8 vd = *reinterpret_cast<const short8*>(dy + e);
+    short8 vy = {};
+    if (y) vy = *reinterpret_cast<const short8*>(y + e);
+    short8 odx, ods;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      float d = s16_to_f32<T16>(vd[u]);
+      if (y && s16_to_f32<T16>(vy[u]) <= 0.f) d = 0.f;
+      if (dres) ods[u] = f32_to_s16<T16>(d);
+      const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
+      odx[u] = f32_to_s16<T16>(g_[u] * (d - b_[u] - xh * a_[u]));
+    }
+    *reinterpret_cast<short8*>(dx + e) = odx;
+    if (dres) *reinterpret_cast<short8*>(dres + e) = ods;
+  }
+}
+
 inline bool bn_fast_ok(long M, int C) {
   return C % 8 == 0 && 2048 % C == 0;
 }
@@ -266,6 +338,19 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
   const long M = x.numel() / C;
   auto y = at::empty_like(x);
   const int has_res = res.numel() > 0;
+  if (bn_fast_ok(M, C)) {
+    const long E = M * (long)C;
+    const int grid = (int)std::min<long>(cdiv_l(E, 256 * 8), 2048);
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL(bn_apply_fast<T16>, dim3(grid), dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                         has_res ? (const T16*)res.data_ptr() : nullptr,
+                         (T16*)y.data_ptr(), E, C, (int)act);
+    });
+    return y;
+  }
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(bn_apply_kernel<T16>, dim3(ew_grid2(M * C)), dim3(256),
                        0, cur_stream(), (const T16*)x.data_ptr(),
@@ -326,6 +411,22 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
   const long M = x.numel() / C;
   auto dx = at::empty_like(x);
   auto dres = want_dres ? at::empty_like(x) : at::Tensor();
+  if (bn_fast_ok(M, C)) {
+    const long E = M * (long)C;
+    const int grid = (int)std::min<long>(cdiv_l(E, 256 * 8), 2048);
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL(bn_bwd_dx_fast<T16>, dim3(grid), dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)dy.data_ptr(),
+                         y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+                         dbeta.data_ptr<float>(), (T16*)dx.data_ptr(),
+                         want_dres ? (T16*)dres.data_ptr() : nullptr, E, C,
+                         (float)(1.0 / m_total));
+    });
+    return {dx, want_dres ? dres : at::Tensor()};
+  }
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(bn_bwd_dx_kernel<T16>, dim3(ew_grid2(M * C)),
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),

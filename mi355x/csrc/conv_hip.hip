@@ -156,6 +156,68 @@ inline dim3 conv_grid(long total, int block = 256, int cap = 4096) {
   return dim3((unsigned)std::min<long>(cdiv_l(total, block), cap));
 }
 
+// ---- small-C wgrad (stems: C=3/6, K<=64) -------------------------------
+// lane = output channel k, all S_*C_ taps of one filter row r in REGISTERS
+// (compile-time C_/S_ keep the accumulator array in VGPRs — guide rule
+// 20); dy reads coalesced across lanes; x reads wave-uniform (broadcast).
+// Each wave writes its own partial slab (no atomics); wgrad_reduce sums.
+template <typename T16, int C_, int S_>
+__global__ __launch_bounds__(256) void conv_wgrad_smallc(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    float* __restrict__ part,  // [gridDim.x*4][K*R*S*C]
+    int N, int H, int W, int K, int Ho, int Wo, int R, int stride, int pad,
+    long m_per_chunk, long Mtot) {
+  const int r = blockIdx.y;
+  const int k = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  float acc[C_ * S_] = {};
+  const long m_begin = (long)blockIdx.x * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+  const long per_wave = (m_end - m_begin + 3) / 4;
+  const long w0 = m_begin + wv * per_wave;
+  const long w1 = min(m_end, w0 + per_wave);
+  if (w0 < w1) {
+    int q = (int)(w0 % Wo);
+    long np = w0 / Wo;
+    int p = (int)(np % Ho);
+    int n = (int)(np / Ho);
+    for (long m = w0; m < w1; ++m) {
+      const float dyv =
+          (k < K) ? F16<T16>::to_f32(dy[m * K + k]) : 0.f;
+      const int ih = p * stride - pad + r;
+      if (ih >= 0 && ih < H && dyv != 0.f) {
+        const T16* row = x + ((long)n * H + ih) * W * C_;
+#pragma unroll
+        for (int s = 0; s < S_; ++s) {
+          const int iw = q * stride - pad + s;
+          if (iw >= 0 && iw < W) {
+            const T16* px = row + (long)iw * C_;
+#pragma unroll
+            for (int c = 0; c < C_; ++c)
+              acc[s * C_ + c] += dyv * F16<T16>::to_f32(px[c]);
+          }
+        }
+      }
+      if (++q == Wo) {
+        q = 0;
+        if (++p == Ho) {
+          p = 0;
+          ++n;
+        }
+      }
+    }
+  }
+  if (k < K) {
+    const long E = (long)K * R * S_ * C_;
+    float* slab = part + ((long)blockIdx.x * 4 + wv) * E;
+#pragma unroll
+    for (int s = 0; s < S_; ++s)
+#pragma unroll
+      for (int c = 0; c < C_; ++c)
+        slab[(((long)k * C_ + c) * R + r) * S_ + s] = acc[s * C_ + c];
+  }
+}
+
 }  // namespace
 
 static inline int out_dim(int in, int k, int stride, int pad) {
@@ -164,6 +226,7 @@ static inline int out_dim(int in, int k, int stride, int pad) {
 
 // conv_mfma.hip — MFMA implicit-GEMM path for C%64==0 && K%64==0
 bool conv_mfma_supported(long CI, long KO);
+void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act);
 void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
@@ -243,6 +306,45 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
   if (conv_mfma_supported(C, K)) {
     auto dw = at::zeros({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad);
+    return dw;
+  }
+  const bool smallc = K <= 64 &&
+      ((C == 3 && (S == 3 || S == 5 || S == 7)) || (C == 6 && S == 5));
+  if (smallc) {
+    const long E = (long)K * R * S * C;
+    long nchunks = std::min<long>(std::max<long>(768 / R, 1), cdiv_l(M, 512));
+    const long m_per_chunk = cdiv_l(M, nchunks);
+    nchunks = cdiv_l(M, m_per_chunk);
+    auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
+    auto part = at::zeros({nchunks * 4, E}, x.options().dtype(at::kFloat));
+    dim3 grid((unsigned)nchunks, R);
+    DISPATCH_16(x, T16, {
+      if (C == 3 && S == 3)
+        hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 3>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, H, W, K, P, Q, R, (int)stride, (int)pad,
+                           m_per_chunk, M);
+      else if (C == 3 && S == 5)
+        hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 5>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, H, W, K, P, Q, R, (int)stride, (int)pad,
+                           m_per_chunk, M);
+      else if (C == 3 && S == 7)
+        hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 7>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, H, W, K, P, Q, R, (int)stride, (int)pad,
+                           m_per_chunk, M);
+      else
+        hipLaunchKernelGGL((conv_wgrad_smallc<T16, 6, 5>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, H, W, K, P, Q, R, (int)stride, (int)pad,
+                           m_per_chunk, M);
+    });
+    wgrad_reduce_launch(part, dw, E, nchunks * 4);
     return dw;
   }
   // split the NPQ reduction so small filters still fill the chip

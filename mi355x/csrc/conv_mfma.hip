@@ -359,19 +359,42 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   }
 
   // ---- scatter the 32x32 fp32 tile, directly in the parameter layout
-  // [KO, CI, R, S] (no host-side permute; contiguous across lanes for the
-  // 1x1 convs where R*S == 1) ----
+  // [KO, CI, R, S]. Split-M chunks write DISJOINT per-chunk slabs (plain
+  // stores; a reduce kernel sums them) — fp32 atomicAdd contention on the
+  // small dw region measured ~21 ns/op amortized (layer1 chunk ablation),
+  // dominating the kernel at high chunk counts ----
   const int rs = r_ * S + s_;
   const long RS = (long)R * S;
+  float* slab = dw + (long)blockIdx.z * ((long)KO * CI * RS);
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;  // KO row
     const int c_abs = c0 + j0 + li;
-    float* p = dw + ((long)(k0 + i0 + i) * CI + c_abs) * RS + rs;
-    if (nchunks == 1)
-      *p = acc[reg];
-    else
-      atomicAdd(p, acc[reg]);
+    slab[((long)(k0 + i0 + i) * CI + c_abs) * RS + rs] = acc[reg];
+  }
+}
+
+// dw[e] = sum_z part[z][e] (fp32, vectorized)
+__global__ void wgrad_reduce_chunks(const float* __restrict__ part,
+                                    float* __restrict__ dw, long E, int nz) {
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  const long stride = (long)gridDim.x * blockDim.x * 4;
+  for (long i = i0; i < E; i += stride) {
+    if (i + 4 <= E) {
+      float4v acc = *reinterpret_cast<const float4v*>(part + i);
+      for (int z = 1; z < nz; ++z) {
+        float4v v = *reinterpret_cast<const float4v*>(part + (long)z * E + i);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) acc[u] += v[u];
+      }
+      *reinterpret_cast<float4v*>(dw + i) = acc;
+    } else {
+      for (long e = i; e < E; ++e) {
+        float a = part[e];
+        for (int z = 1; z < nz; ++z) a += part[(long)z * E + e];
+        dw[e] = a;
+      }
+    }
   }
 }
 
@@ -379,6 +402,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
 
 bool conv_mfma_supported(long CI, long KO) {
   return CI % 64 == 0 && KO % 64 == 0;
+}
+
+void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz) {
+  const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
+  hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(grid_r), dim3(256), 0,
+                     cur_stream(), part.data_ptr<float>(),
+                     dw.data_ptr<float>(), E, (int)nz);
 }
 
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
@@ -391,23 +421,30 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     const char* e = getenv("MI355X_WGRAD_CHUNKS");
     return e ? atol(e) : 0L;
   }();
-  // measured sweet spot (layer1 ablation: 216ch 289us / 64ch 169us /
-  // 16ch 327us): ~2.5 blocks/CU balances grid fill against the split-M
-  // atomic traffic (total_w x nchunks fp32 atomicAdds)
-  int nchunks = (int)std::min<long>(std::max<long>(640 / blocks_xy, 1),
+  int nchunks = (int)std::min<long>(std::max<long>(2048 / blocks_xy, 1),
                                     cdiv_l(M, 512));
   if (cap > 0) nchunks = (int)std::min<long>(nchunks, cap);
   nchunks = std::max(nchunks, 1);
   long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
   nchunks = (int)cdiv_l(M, m_per_chunk);
   dim3 grid(KO / 64, R * S * (CI / 64), nchunks);
+  const long E = (long)KO * R * S * CI;
+  at::Tensor part = nchunks > 1
+                        ? at::empty({nchunks, E}, dw.options())
+                        : dw;
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(conv_wgrad_mfma_kernel<T16>, grid, dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)dy.data_ptr(), dw.data_ptr<float>(), N, Hi,
-                       Wi, CI, KO, Ho, Wo, (int)R, (int)S, (int)stride,
+                       (const T16*)dy.data_ptr(), part.data_ptr<float>(), N,
+                       Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S, (int)stride,
                        (int)pad, m_per_chunk, nchunks);
   });
+  if (nchunks > 1) {
+    const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
+    hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(grid_r), dim3(256), 0,
+                       cur_stream(), part.data_ptr<float>(),
+                       dw.data_ptr<float>(), E, nchunks);
+  }
 }
 
 // GENC fwd: wpad = [KO, KGP] zero-padded (cast_permute_krsc_pad)
