@@ -155,6 +155,113 @@ inline dim3 conv_grid(long total, int block = 256, int cap = 4096) {
   return dim3((unsigned)std::min<long>(cdiv_l(total, block), cap));
 }
 
+// ---- small-C fwd (stems: 3x3, C=3/6, K<=64, stride 1/2) ----------------
+// lane = output channel k; the 3x3xC input window and the lane's weights
+// live in REGISTERS; x loads are wave-uniform (one k-row broadcast) and
+// the window slides along q so each output costs R*stride*C new loads.
+template <typename T16, int C_, int S_>
+__global__ __launch_bounds__(256) void conv_fwd_smallc(
+    const T16* __restrict__ x, const T16* __restrict__ wgt,
+    const float* __restrict__ bias, T16* __restrict__ y, int N, int H, int W,
+    int K, int Ho, int Wo, int stride, int pad, long wrow_stride, int act,
+    int has_bias, long m_per_chunk, long Mtot) {
+  const int k = threadIdx.x & 63;
+  const int wv = threadIdx.x >> 6;
+  // per-lane weights: w[r][s][c] from [K][kg] (kg=(r*S+s)*C+c, maybe padded)
+  float wr[S_ * S_ * C_];
+  {
+    const T16* wp = wgt + (long)min(k, K - 1) * wrow_stride;
+#pragma unroll
+    for (int i = 0; i < S_ * S_ * C_; ++i) wr[i] = F16<T16>::to_f32(wp[i]);
+  }
+  const float bk = has_bias ? bias[min(k, K - 1)] : 0.f;
+
+  const long m_begin = (long)blockIdx.x * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+  const long per_wave = (m_end - m_begin + 3) / 4;
+  long w0 = m_begin + wv * per_wave;
+  const long w1 = min(m_end, w0 + per_wave);
+  if (w0 >= w1) return;
+
+  int q = (int)(w0 % Wo);
+  long np = w0 / Wo;
+  int p = (int)(np % Ho);
+  int n = (int)(np / Ho);
+
+  float xw[S_ * S_ * C_];  // sliding window [r][s][c]
+  bool fresh = true;
+  for (long m = w0; m < w1; ++m) {
+    const int ih0 = p * stride - pad;
+    const int iw0 = q * stride - pad;
+    if (fresh) {
+#pragma unroll
+      for (int r = 0; r < S_; ++r) {
+        const int ih = ih0 + r;
+        const bool rok = (unsigned)ih < (unsigned)H;
+        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+#pragma unroll
+        for (int ss = 0; ss < S_; ++ss) {
+          const int iw = iw0 + ss;
+          const bool ok = rok && (unsigned)iw < (unsigned)W;
+#pragma unroll
+          for (int c = 0; c < C_; ++c)
+            xw[(r * S_ + ss) * C_ + c] =
+                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+        }
+      }
+      fresh = false;
+    } else {
+      // slide: shift left by `stride` columns, load the new ones
+      if (stride == 1) {
+#pragma unroll
+        for (int r = 0; r < S_; ++r)
+#pragma unroll
+          for (int ss = 0; ss < S_ - 1; ++ss)
+#pragma unroll
+            for (int c = 0; c < C_; ++c)
+              xw[(r * S_ + ss) * C_ + c] = xw[(r * S_ + ss + 1) * C_ + c];
+      } else {
+#pragma unroll
+        for (int r = 0; r < S_; ++r)
+#pragma unroll
+          for (int ss = 0; ss < S_ - 2; ++ss)
+#pragma unroll
+            for (int c = 0; c < C_; ++c)
+              xw[(r * S_ + ss) * C_ + c] = xw[(r * S_ + ss + 2) * C_ + c];
+      }
+      const int nnew = stride;  // new columns on the right
+#pragma unroll
+      for (int r = 0; r < S_; ++r) {
+        const int ih = ih0 + r;
+        const bool rok = (unsigned)ih < (unsigned)H;
+        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+        for (int j = 0; j < nnew; ++j) {
+          const int ss = S_ - nnew + j;
+          const int iw = iw0 + ss;
+          const bool ok = rok && (unsigned)iw < (unsigned)W;
+#pragma unroll
+          for (int c = 0; c < C_; ++c)
+            xw[(r * S_ + ss) * C_ + c] =
+                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+        }
+      }
+    }
+    float acc = bk;
+#pragma unroll
+    for (int i = 0; i < S_ * S_ * C_; ++i) acc += xw[i] * wr[i];
+    if (act == 1) acc = fmaxf(acc, 0.f);
+    if (k < K) y[m * K + k] = F16<T16>::from_f32(acc);
+    if (++q == Wo) {
+      q = 0;
+      fresh = true;
+      if (++p == Ho) {
+        p = 0;
+        ++n;
+      }
+    }
+  }
+}
+
 // ---- small-C wgrad (stems: C=3/6, K<=64) -------------------------------
 // lane = output channel k, all S_*C_ taps of one filter row r in REGISTERS
 // (compile-time C_/S_ keep the accumulator array in VGPRs — guide rule
@@ -248,6 +355,35 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int S = w.dim() == 4 ? w.size(2) : kS;
   const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
   auto y = at::empty({N, P, Q, K}, x.options());
+  const bool smallc_fwd = K <= 64 && R == 3 && S == 3 && (C == 3 || C == 6) &&
+                          (stride == 1 || stride == 2);
+  if (smallc_fwd) {
+    const long M = (long)N * P * Q;
+    const long wrow = w.dim() == 2 ? w.size(1) : (long)R * S * C;
+    long nchunks = std::min<long>(768, cdiv_l(M, 512));
+    const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
+    nchunks = cdiv_l(M, m_per_chunk);
+    const int has_bias = bias.numel() > 0;
+    DISPATCH_16(x, T16, {
+      if (C == 3)
+        hipLaunchKernelGGL((conv_fwd_smallc<T16, 3, 3>), dim3(nchunks),
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)stride,
+                           (int)pad, wrow, (int)act, has_bias, m_per_chunk,
+                           M);
+      else
+        hipLaunchKernelGGL((conv_fwd_smallc<T16, 6, 3>), dim3(nchunks),
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)stride,
+                           (int)pad, wrow, (int)act, has_bias, m_per_chunk,
+                           M);
+    });
+    return y;
+  }
   if (w.dim() == 2) {  // padded [KO,KGP] weight: generic small-C MFMA path
     conv_fwd_mfma_genc_launch(x, w, bias, y, R, S, stride, pad, act);
     return y;
