@@ -83,7 +83,11 @@ def main():
     else:
         flat = FlatState(net)
         gscale = 1.0
-    optimizer = optim.SGD(flat, lr=0.1, momentum=0.9, grad_scale=gscale)
+    # fp16 (BASELINE config 5): static loss scaling, graph-safe; the
+    # 1/scale is folded into the fused SGD grad_scale
+    loss_scale = 256.0 if args.fp16 else 1.0
+    optimizer = optim.SGD(flat, lr=0.1, momentum=0.9,
+                          grad_scale=gscale / loss_scale)
 
     # synthetic on-device data pool (new batch each step, cycling)
     g = torch.Generator(device="cpu").manual_seed(4321 + rank)
@@ -97,7 +101,7 @@ def main():
         optimizer.zero_grad()
         out = net(x)
         loss = cross_entropy(out, y)
-        loss.backward()
+        (loss * loss_scale).backward() if loss_scale != 1.0 else loss.backward()
         if distributed:
             net.finish_grad_sync()
         optimizer.step()

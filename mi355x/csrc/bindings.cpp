@@ -20,6 +20,9 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                         long stride, long pad);
 
 at::Tensor bn_stats(at::Tensor x);
+at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
+                       at::Tensor running_var, double m_total,
+                       double momentum, double eps);
 at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                     at::Tensor gamma, at::Tensor beta, at::Tensor res,
                     long act);
@@ -61,6 +64,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("bn_stats", &bn_stats);
+  m.def("bn_finalize", &bn_finalize);
   m.def("bn_apply", &bn_apply);
   m.def("bn_bwd_reduce", &bn_bwd_reduce);
   m.def("bn_bwd_dx", &bn_bwd_dx);

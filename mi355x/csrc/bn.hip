@@ -269,6 +269,26 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
   }
 }
 
+// one tiny kernel replacing the Python stats glue (mean/var/invstd/
+// running-stat updates were ~8 torch launches per BN layer per step)
+__global__ void bn_finalize_kernel(const float* __restrict__ stats,  // [2,C]
+                                   float* __restrict__ out,  // [2,C] mean,invstd
+                                   float* __restrict__ rmean,
+                                   float* __restrict__ rvar, int C,
+                                   float inv_m, float unbias, float momentum,
+                                   float eps) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const float mean = stats[c] * inv_m;
+  const float var = stats[C + c] * inv_m - mean * mean;
+  out[c] = mean;
+  out[C + c] = rsqrtf(var + eps);
+  if (rmean) {
+    rmean[c] = rmean[c] * (1.f - momentum) + mean * momentum;
+    rvar[c] = rvar[c] * (1.f - momentum) + var * unbias * momentum;
+  }
+}
+
 inline bool bn_fast_ok(long M, int C) {
   return C % 8 == 0 && 2048 % C == 0;
 }
@@ -294,6 +314,23 @@ inline int ew_grid2(long n, int block = 256) {
 }
 
 }  // namespace
+
+at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
+                       at::Tensor running_var, double m_total,
+                       double momentum, double eps) {
+  const int C = stats.size(1);
+  auto out = at::empty_like(stats);
+  const bool has_run = running_mean.numel() > 0;
+  const double unbias = m_total / std::max(m_total - 1.0, 1.0);
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3(cdiv_i(C, 256)), dim3(256), 0,
+                     cur_stream(), stats.data_ptr<float>(),
+                     out.data_ptr<float>(),
+                     has_run ? running_mean.data_ptr<float>() : nullptr,
+                     has_run ? running_var.data_ptr<float>() : nullptr, C,
+                     (float)(1.0 / m_total), (float)unbias, (float)momentum,
+                     (float)eps);
+  return out;
+}
 
 at::Tensor bn_stats(at::Tensor x) {
   CHECK_GPU(x);

@@ -287,25 +287,54 @@ __global__ __launch_bounds__(256) void conv_wgrad_smallc(
     long np = w0 / Wo;
     int p = (int)(np % Ho);
     int n = (int)(np / Ho);
+    // sliding x window for the fixed filter row r: S_*C_ values, shifted
+    // by `stride` columns per output step (wave-uniform loads)
+    float xq[S_ * C_];
+    bool fresh = true;
     for (long m = w0; m < w1; ++m) {
-      const float dyv =
-          (k < K) ? F16<T16>::to_f32(dy[m * K + k]) : 0.f;
       const int ih = p * stride - pad + r;
-      if (ih >= 0 && ih < H && dyv != 0.f) {
-        const T16* row = x + ((long)n * H + ih) * W * C_;
+      const bool rok = (unsigned)ih < (unsigned)H;
+      const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+      const int iw0 = q * stride - pad;
+      if (fresh || !rok) {
 #pragma unroll
         for (int s = 0; s < S_; ++s) {
-          const int iw = q * stride - pad + s;
-          if (iw >= 0 && iw < W) {
-            const T16* px = row + (long)iw * C_;
+          const int iw = iw0 + s;
+          const bool ok = rok && (unsigned)iw < (unsigned)W;
 #pragma unroll
-            for (int c = 0; c < C_; ++c)
-              acc[s * C_ + c] += dyv * F16<T16>::to_f32(px[c]);
-          }
+          for (int c = 0; c < C_; ++c)
+            xq[s * C_ + c] =
+                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
         }
+        fresh = false;
+      } else {
+        if (stride == 1) {
+#pragma unroll
+          for (int i = 0; i < (S_ - 1) * C_; ++i) xq[i] = xq[i + C_];
+        } else {
+#pragma unroll
+          for (int i = 0; i < (S_ - 2) * C_; ++i) xq[i] = xq[i + 2 * C_];
+        }
+        const int nnew = stride;
+        for (int j = 0; j < nnew; ++j) {
+          const int s = S_ - nnew + j;
+          const int iw = iw0 + s;
+          const bool ok = (unsigned)iw < (unsigned)W;
+#pragma unroll
+          for (int c = 0; c < C_; ++c)
+            xq[s * C_ + c] =
+                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+        }
+      }
+      const float dyv =
+          (k < K) ? F16<T16>::to_f32(dy[m * K + k]) : 0.f;
+      if (rok && dyv != 0.f) {
+#pragma unroll
+        for (int i = 0; i < S_ * C_; ++i) acc[i] += dyv * xq[i];
       }
       if (++q == Wo) {
         q = 0;
+        fresh = true;
         if (++p == Ho) {
           p = 0;
           ++n;

@@ -37,13 +37,15 @@ struct Mfma32<__half> {
 
 constexpr int BM = 128, BN = 64, BK = 64;
 constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
+// NT = 32-wide N(output-channel) tiles per block: 2 (BN=64) or 4 (BN=128);
+// wider tiles double the MFMA work per barrier pair for the K>=128 layers
 
 // TRANS=false (fwd): src row = ho*stride - pad + r, in [0,Hi)
 // TRANS=true (dgrad): src row = (ho + pad - r), valid iff %stride==0, /stride in [0,Hi)
 // GENC (fwd only): CI % 64 != 0 — the weight is pre-padded to
 // [KO, KGP=ceil(R*S*CI/64)*64] with kg=(r*S+s)*CI+c and zeros beyond, and
 // the A side gathers per-ELEMENT across tap boundaries (stem convs C=3/6).
-template <typename T16, bool TRANS, bool GENC = false>
+template <typename T16, bool TRANS, bool GENC, int NT>
 __global__ __launch_bounds__(256) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
@@ -56,12 +58,13 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   // single-buffered: 2x LDS for a double buffer measurably LOSES here —
   // it halves blocks/CU and the cross-block overlap it sacrifices was
   // already hiding the staging latency (guide common-mistake #5)
-  __shared__ T16 lds[BM * LDK + BN * LDK];
+  __shared__ T16 lds[BM * LDK + NT * 32 * LDK];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
   const long bm0 = (long)blockIdx.x * BM;
-  const int k0 = blockIdx.y * BN;
+  constexpr int BNT = NT * 32;
+  const int k0 = blockIdx.y * BNT;
 
   // ---- A staging coords (2 threads per m-row, 32 channels each) ----
   const int sa_m = tid >> 1;
@@ -79,9 +82,11 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   const int ih0 = TRANS ? p_ : p_ * stride - pad;
   const int iw0 = TRANS ? q_ : q_ * stride - pad;
 
-  // ---- B staging coords (4 threads per n-row, 16 elements each) ----
-  const int sb_n = tid >> 2;
-  const int sb_c = (tid & 3) * (BK / 4);
+  // ---- B staging coords (256/BNT threads per n-row) ----
+  constexpr int TPR = 256 / BNT;          // threads per B row
+  constexpr int EPT = BK / TPR;           // elements per thread
+  const int sb_n = tid / TPR;
+  const int sb_c = (tid % TPR) * EPT;
   const T16* wrow = wgt + (long)(k0 + sb_n) * b_row_stride + sb_c;
 
   // ---- wave/lane fragment coords ----
@@ -91,14 +96,14 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   const int kh = lane >> 5;  // k-half: lane holds k = kk + kh*8 + e
   const int wm = wave * 32;  // wave's m-offset inside the block tile
 
-  f32x16 acc0 = {}, acc1 = {};
+  f32x16 acc[NT] = {};
 
   const int cchunks = GENC ? 1 : CI / BK;
   const int ksteps = GENC ? (int)(b_row_stride / BK)  // KGP/64
                           : R * S * cchunks;
 
-  short8 sa[4];  // 32 channels = 4 x 16B
-  short8 sb[2];  // 16 elements = 2 x 16B
+  short8 sa[4];            // 32 channels = 4 x 16B
+  short8 sb[EPT / 8];      // B elements per thread
 
   auto load_step_genc = [&](int j) {
     // per-element gather: kg walks (tap, c) with carries (fully unrolled,
@@ -130,8 +135,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       }
     }
     const T16* wp = wgt + (long)(k0 + sb_n) * b_row_stride + j * BK + sb_c;
-    sb[0] = *reinterpret_cast<const short8*>(wp);
-    sb[1] = *reinterpret_cast<const short8*>(wp + 8);
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
   };
 
   auto load_step = [&](int j) {
@@ -167,8 +173,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       for (int i = 0; i < 4; ++i) sa[i] = short8{};
     }
     const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
-    sb[0] = *reinterpret_cast<const short8*>(wp);
-    sb[1] = *reinterpret_cast<const short8*>(wp + 8);
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
   };
 
   auto stage = [&]() {
@@ -179,8 +186,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     for (int i = 0; i < 4; ++i)
       *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
     short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
-    *reinterpret_cast<short8*>(pb) = sb[0];
-    *reinterpret_cast<short8*>(pb + 8) = sb[1];
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      *reinterpret_cast<short8*>(pb + 8 * i) = sb[i];
   };
 
   load_step(0);
@@ -195,31 +203,31 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     for (int kk = 0; kk < BK; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
           ldsA + (wm + li) * LDK + kk + kh * 8);
-      const short8 bf0 = *reinterpret_cast<const short8*>(
-          ldsB + li * LDK + kk + kh * 8);
-      const short8 bf1 = *reinterpret_cast<const short8*>(
-          ldsB + (32 + li) * LDK + kk + kh * 8);
-      acc0 = Mfma32<T16>::run(af, bf0, acc0);
-      acc1 = Mfma32<T16>::run(af, bf1, acc1);
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        const short8 bf = *reinterpret_cast<const short8*>(
+            ldsB + (tnt * 32 + li) * LDK + kk + kh * 8);
+        acc[tnt] = Mfma32<T16>::run(af, bf, acc[tnt]);
+      }
     }
   }
 
   // ---- epilogue: bias + act + store ----
-  const float b0 = has_bias ? bias[k0 + li] : 0.f;
-  const float b1 = has_bias ? bias[k0 + 32 + li] : 0.f;
+  float bv[NT];
+#pragma unroll
+  for (int tnt = 0; tnt < NT; ++tnt)
+    bv[tnt] = has_bias ? bias[k0 + tnt * 32 + li] : 0.f;
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
     const long m_out = bm0 + wm + row;
     if (m_out < Mtot) {
-      float v0 = acc0[reg] + b0;
-      float v1 = acc1[reg] + b1;
-      if (act == 1) {
-        v0 = fmaxf(v0, 0.f);
-        v1 = fmaxf(v1, 0.f);
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        float v = acc[tnt][reg] + bv[tnt];
+        if (act == 1) v = fmaxf(v, 0.f);
+        out[m_out * KO + k0 + tnt * 32 + li] = F16<T16>::from_f32(v);
       }
-      out[m_out * KO + k0 + li] = F16<T16>::from_f32(v0);
-      out[m_out * KO + k0 + 32 + li] = F16<T16>::from_f32(v1);
     }
   }
 }
@@ -460,8 +468,8 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
   dim3 grid((unsigned)cdiv_l(M, BM), KO / BN);
   const int has_bias = bias.numel() > 0;
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL((conv_gather_gemm<T16, false, true>), grid, dim3(256),
-                       0, cur_stream(), (const T16*)x.data_ptr(),
+    hipLaunchKernelGGL((conv_gather_gemm<T16, false, true, 2>), grid,
+                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
                        (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, (int)R,
@@ -477,16 +485,26 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int KO = w.size(0), R = w.size(1), S = w.size(2);
   const int Ho = y.size(1), Wo = y.size(2);
   const long M = (long)N * Ho * Wo;
-  dim3 grid((unsigned)cdiv_l(M, BM), KO / BN);
   const int has_bias = bias.numel() > 0;
+  const bool wide = KO % 128 == 0;  // BN=128 halves barriers per MFMA
+  dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL((conv_gather_gemm<T16, false>), grid, dim3(256), 0,
-                       cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)w.data_ptr(),
-                       has_bias ? bias.data_ptr<float>() : nullptr,
-                       (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
-                       (int)stride, (int)pad, (long)R * S * CI, (long)CI,
-                       (int)act, has_bias);
+    if (wide)
+      hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4>), grid,
+                         dim3(256), 0, cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
+                         (int)stride, (int)pad, (long)R * S * CI, (long)CI,
+                         (int)act, has_bias);
+    else
+      hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 2>), grid,
+                         dim3(256), 0, cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
+                         (int)stride, (int)pad, (long)R * S * CI, (long)CI,
+                         (int)act, has_bias);
   });
 }
 
@@ -497,14 +515,24 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
-  dim3 grid((unsigned)cdiv_l(M, BM), CI / BN);
-  auto empty_bias = at::Tensor();
+  const bool wide = CI % 128 == 0;
+  dim3 grid((unsigned)cdiv_l(M, BM), CI / (wide ? 128 : 64));
   DISPATCH_16(dy, T16, {
-    hipLaunchKernelGGL((conv_gather_gemm<T16, true>), grid, dim3(256), 0,
-                       cur_stream(), (const T16*)dy.data_ptr(),
-                       (const T16*)wflip.data_ptr(), nullptr,
-                       (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
-                       (int)S, (int)stride, (int)pad, (long)KO,
-                       (long)CI * KO, 0, 0);
+    if (wide)
+      hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 4>), grid,
+                         dim3(256), 0, cur_stream(),
+                         (const T16*)dy.data_ptr(),
+                         (const T16*)wflip.data_ptr(), nullptr,
+                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
+                         (int)S, (int)stride, (int)pad, (long)KO,
+                         (long)CI * KO, 0, 0);
+    else
+      hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 2>), grid,
+                         dim3(256), 0, cur_stream(),
+                         (const T16*)dy.data_ptr(),
+                         (const T16*)wflip.data_ptr(), nullptr,
+                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
+                         (int)S, (int)stride, (int)pad, (long)KO,
+                         (long)CI * KO, 0, 0);
   });
 }

@@ -220,14 +220,12 @@ class _BNFn(torch.autograd.Function):
 
             dist.all_reduce(s, group=process_group)
             m_total = m_local * dist.get_world_size(process_group)
-        mean = s[0] / m_total
-        var = s[1] / m_total - mean * mean
-        invstd = torch.rsqrt(var + BN_EPS)
-        if running_mean is not None:
-            with torch.no_grad():
-                unbiased = var * (m_total / max(m_total - 1, 1))
-                running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-                running_var.mul_(1 - momentum).add_(unbiased, alpha=momentum)
+        empty = torch.empty(0, device=x.device)
+        mi = ext().bn_finalize(
+            s, running_mean if running_mean is not None else empty,
+            running_var if running_var is not None else empty,
+            float(m_total), momentum, BN_EPS)  # [2,C]: mean, invstd
+        mean, invstd = mi[0], mi[1]
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
         y = ext().bn_apply(x, mean, invstd, gamma.detach().float(),
                            beta.detach().float(), res, act)

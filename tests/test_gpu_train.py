@@ -77,3 +77,12 @@ def test_checkpoint_gpu_cpu_interchange(tmp_path):
         {k: v.cpu() for k, v in torch.load(p, weights_only=True).items()})
     x = torch.randn(2, 3, 32, 32)
     torch.testing.assert_close(net2(x), net.cpu()(x), rtol=1e-4, atol=1e-5)
+
+
+def test_resnet18_fp16_trains_on_gpu():
+    from mi355x import amp
+    torch.manual_seed(0)
+    with amp.autocast(torch.float16):
+        losses = _train(build_model("resnet18"), steps=6)
+    assert all(l == l for l in losses), losses
+    assert losses[-1] < losses[0] * 1.05, losses
