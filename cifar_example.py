@@ -50,6 +50,11 @@ def main():
     flat = FlatState(net)
     optimizer = optim.SGD(flat, lr=lr, momentum=0.9)
 
+    meter = None
+    if os.environ.get("MI355X_LOG_SPEED", "0") == "1":
+        from mi355x.utils import SpeedMeter
+        meter = SpeedMeter(print_every=100)
+
     t0 = time.time()
     steps = 0
     for epoch in range(epochs):
@@ -62,6 +67,8 @@ def main():
             loss.backward()
             optimizer.step()
 
+            if meter is not None:
+                meter.step(inputs.shape[0])
             running_loss += loss.item()
             if i % 2000 == 1999:
                 print("[%d, %5d] loss: %.3f" %

@@ -78,6 +78,11 @@ def main(args):
     optimizer = optim.SGD(net.flat, lr=lr, momentum=0.9,
                           grad_scale=net.grad_scale)
 
+    meter = None
+    if os.environ.get("MI355X_LOG_SPEED", "0") == "1":
+        from mi355x.utils import SpeedMeter
+        meter = SpeedMeter(print_every=100)
+
     t0 = time.time()
     steps = 0
     for epoch in range(epochs):
@@ -92,6 +97,8 @@ def main(args):
             net.finish_grad_sync()
             optimizer.step()
 
+            if meter is not None:
+                meter.step(inputs.shape[0])
             running_loss += loss.item()
             if i % 2000 == 1999 and args.rank == 0:
                 print("[%d, %5d] loss: %.3f" %
