@@ -229,20 +229,32 @@ __global__ __launch_bounds__(256) void conv_fwd_smallc(
             for (int c = 0; c < C_; ++c)
               xw[(r * S_ + ss) * C_ + c] = xw[(r * S_ + ss + 2) * C_ + c];
       }
-      const int nnew = stride;  // new columns on the right
+      // new right-hand columns: compile-time ss indices (a runtime index
+      // into xw[] would force the whole window to scratch — rule 20)
 #pragma unroll
       for (int r = 0; r < S_; ++r) {
         const int ih = ih0 + r;
         const bool rok = (unsigned)ih < (unsigned)H;
         const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
-        for (int j = 0; j < nnew; ++j) {
-          const int ss = S_ - nnew + j;
+        if (stride == 1) {
+          constexpr int ss = S_ - 1;
           const int iw = iw0 + ss;
           const bool ok = rok && (unsigned)iw < (unsigned)W;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
             xw[(r * S_ + ss) * C_ + c] =
                 ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+        } else {
+#pragma unroll
+          for (int jj = 0; jj < 2; ++jj) {
+            const int ss = S_ - 2 + jj;
+            const int iw = iw0 + ss;
+            const bool ok = rok && (unsigned)iw < (unsigned)W;
+#pragma unroll
+            for (int c = 0; c < C_; ++c)
+              xw[(r * S_ + ss) * C_ + c] =
+                  ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+          }
         }
       }
     }
@@ -315,15 +327,25 @@ __global__ __launch_bounds__(256) void conv_wgrad_smallc(
 #pragma unroll
           for (int i = 0; i < (S_ - 2) * C_; ++i) xq[i] = xq[i + 2 * C_];
         }
-        const int nnew = stride;
-        for (int j = 0; j < nnew; ++j) {
-          const int s = S_ - nnew + j;
-          const int iw = iw0 + s;
+        if (stride == 1) {
+          constexpr int ss = S_ - 1;
+          const int iw = iw0 + ss;
           const bool ok = (unsigned)iw < (unsigned)W;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
-            xq[s * C_ + c] =
+            xq[ss * C_ + c] =
                 ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+        } else {
+#pragma unroll
+          for (int jj = 0; jj < 2; ++jj) {
+            const int ss = S_ - 2 + jj;
+            const int iw = iw0 + ss;
+            const bool ok = (unsigned)iw < (unsigned)W;
+#pragma unroll
+            for (int c = 0; c < C_; ++c)
+              xq[ss * C_ + c] =
+                  ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+          }
         }
       }
       const float dyv =

@@ -58,22 +58,28 @@ __global__ void linear_dgrad_kernel(const T16* __restrict__ dy,
   }
 }
 
-// dw[N,K] (f32) = dy[M,N]^T @ x[M,K]
+// dw[N,K] (f32) = dy[M,N]^T @ x[M,K]; split-M over blockIdx.y with
+// atomics (classifier dw is small, the M loop is the long axis)
 template <typename T16>
 __global__ void linear_wgrad_kernel(const T16* __restrict__ x,
                                     const T16* __restrict__ dy,
                                     float* __restrict__ dw, int M, int N,
-                                    int K) {
+                                    int K, int m_per_chunk) {
   const long total = (long)N * K;
+  const int m0 = blockIdx.y * m_per_chunk;
+  const int m1 = min(M, m0 + m_per_chunk);
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const int k = (int)(t % K);
     const int nn = (int)(t / K);
     float acc = 0.f;
-    for (int m = 0; m < M; ++m)
+    for (int m = m0; m < m1; ++m)
       acc += F16<T16>::to_f32(dy[(long)m * N + nn]) *
              F16<T16>::to_f32(x[(long)m * K + k]);
-    dw[t] = acc;
+    if (gridDim.y == 1)
+      dw[t] = acc;
+    else
+      atomicAdd(dw + t, acc);
   }
 }
 
@@ -117,12 +123,20 @@ at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   const int M = x.size(0), K = x.size(1), N = dy.size(1);
-  auto dw = at::empty({N, K}, x.options().dtype(at::kFloat));
+  const long total = (long)N * K;
+  const long xy_blocks = cdiv_l(total, 256);
+  int nchunks = (int)std::min<long>(std::max<long>(512 / xy_blocks, 1),
+                                    cdiv_l(M, 64));
+  const int m_per_chunk = (int)cdiv_l(M, nchunks);
+  nchunks = (int)cdiv_l(M, m_per_chunk);
+  auto dw = nchunks == 1 ? at::empty({N, K}, x.options().dtype(at::kFloat))
+                         : at::zeros({N, K}, x.options().dtype(at::kFloat));
+  dim3 grid((unsigned)std::min<long>(xy_blocks, 4096), nchunks);
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL(linear_wgrad_kernel<T16>, dim3(ggrid((long)N * K)),
-                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
+    hipLaunchKernelGGL(linear_wgrad_kernel<T16>, grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)dy.data_ptr(), dw.data_ptr<float>(), M, N,
-                       K);
+                       K, m_per_chunk);
   });
   return dw;
 }
