@@ -107,6 +107,11 @@ def main():
         optimizer.step()
         return loss
 
+    # world>1 graph capture (RCCL collectives inside hipGraph) is untested
+    # on this pool — default it off for the multi-GPU scaling run unless
+    # explicitly requested (MI355X_GRAPH_DIST=1)
+    if distributed and os.environ.get("MI355X_GRAPH_DIST", "0") != "1":
+        args.graph = False
     use_graph = args.graph and use_cuda
     if use_graph:
       try:

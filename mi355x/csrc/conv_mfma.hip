@@ -409,6 +409,142 @@ __global__ void wgrad_reduce_chunks(const float* __restrict__ part,
 
 }  // namespace
 
+// ---------------------------------------------------------------------------
+// halo wgrad for the dominant 3x3/stride-1/pad-1 convs: ONE block computes
+// all NINE taps of a 64(KO) x 64(CI) dw tile over its m-chunk, so dy and x
+// are staged ONCE per 64-pixel m-step instead of once per tap (the tap-
+// per-block kernel re-reads each ~9x and measured HBM/L3-traffic-bound).
+// x is staged as a zero-bordered 2-D spatial halo image [(64/W)+2 rows]
+// [W+2 cols][64 ch]; tap (r,s) of pixel (p,q) reads image[p-prow0+r][q+s]
+// so padding needs no per-element masking. The b-fragment m-gather is 8
+// scalar u16 LDS reads with power-of-two (W) row decode.
+// ---------------------------------------------------------------------------
+
+namespace {
+
+constexpr int HALO_IMG_MAX = 224;  // pixels: covers W=8..64 (+2 halo rows)
+
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_wgrad_halo3(
+    const T16* __restrict__ x,   // [N,H,W,CI]
+    const T16* __restrict__ dy,  // [M,KO]
+    float* __restrict__ part,    // [nchunks][KO*CI*9] (KCRS slabs)
+    const int N, const int H, const int W, const int CI, const int KO,
+    const int wshift, const long m_per_chunk) {
+  __shared__ T16 lds[64 * LDM + HALO_IMG_MAX * 64];
+  T16* dyT = lds;                 // [64 k][LDM m]
+  T16* ximg = lds + 64 * LDM;     // [pix][64 c]
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * H * W;
+  const int k0 = blockIdx.x * 64;
+  const int c0 = blockIdx.y * 64;
+  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+
+  const int RW = 64 >> wshift;    // pixel rows per m-step
+  const int IMG_C = W + 2;
+  const int npix = (RW + 2) * IMG_C;
+
+  // dyT staging (threads 0..127): 4 m-rows x 8 k each, transposed write
+  const int t7 = tid & 127;
+  const int sm = (t7 & 15) * 4;
+  const int sk = (t7 >> 4) * 8;
+  // ximg staging (all 256): 4 threads per pixel, 16 channels each
+  const int px_slot = tid >> 2;
+  const int px_c = (tid & 3) * 16;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int i0 = (wave & 1) * 32;   // KO half
+  const int j0 = (wave >> 1) * 32;  // CI half
+
+  f32x16 acc[9] = {};
+
+  // uniform per-step position (advanced incrementally below)
+  int n_ = (int)(m_begin / ((long)H * W));
+  int prow0 = (int)((m_begin % ((long)H * W)) >> wshift);
+
+  for (long m0 = m_begin; m0 < m_end; m0 += 64) {
+    __syncthreads();  // previous MFMA phase done with LDS
+    // ---- stage dyT ----
+    if (tid < 128) {
+      short8 v[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        v[mi] = *reinterpret_cast<const short8*>(dy + (m0 + sm + mi) * KO +
+                                                 k0 + sk);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(dyT + (sk + e) * LDM + sm)) = pk;
+      }
+    }
+    // ---- stage the zero-bordered x halo image ----
+    for (int pix = px_slot; pix < npix; pix += 64) {
+      const int ir = pix / IMG_C;
+      const int ic = pix - ir * IMG_C;
+      const int p_src = prow0 - 1 + ir;
+      const int q_src = ic - 1;
+      const bool ok = (unsigned)p_src < (unsigned)H &&
+                      (unsigned)q_src < (unsigned)W;
+      short8 a = {}, b = {};
+      if (ok) {
+        const T16* src =
+            x + (((long)n_ * H + p_src) * W + q_src) * CI + c0 + px_c;
+        a = *reinterpret_cast<const short8*>(src);
+        b = *reinterpret_cast<const short8*>(src + 8);
+      }
+      short* dst = reinterpret_cast<short*>(ximg + (long)pix * 64 + px_c);
+      *reinterpret_cast<short8*>(dst) = a;
+      *reinterpret_cast<short8*>(dst + 8) = b;
+    }
+    __syncthreads();
+    // ---- 4 k-steps x 9 taps of MFMA ----
+#pragma unroll
+    for (int kk = 0; kk < 64; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          dyT + (i0 + li) * LDM + kk + kh * 8);
+#pragma unroll
+      for (int tap = 0; tap < 9; ++tap) {
+        const int r = tap / 3, ss = tap % 3;
+        short8 bf;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int mrel = kk + kh * 8 + e;
+          const int prow = mrel >> wshift;
+          const int pcol = mrel & (W - 1);
+          bf[e] = *reinterpret_cast<const short*>(
+              ximg + ((long)(prow + r) * IMG_C + pcol + ss) * 64 + j0 + li);
+        }
+        acc[tap] = Mfma32<T16>::run(af, bf, acc[tap]);
+      }
+    }
+    prow0 += RW;
+    if (prow0 == H) {
+      prow0 = 0;
+      ++n_;
+    }
+  }
+
+  // ---- per-chunk slab store, parameter layout [KO,CI,3,3] ----
+  float* slab = part + (long)blockIdx.z * ((long)KO * CI * 9);
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+      slab[((long)(k0 + i0 + i) * CI + c0 + j0 + li) * 9 + tap] =
+          acc[tap][reg];
+    }
+  }
+}
+
+}  // namespace
+
 bool conv_mfma_supported(long CI, long KO) {
   return CI % 64 == 0 && KO % 64 == 0;
 }
@@ -425,6 +561,31 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   const long M = (long)N * Ho * Wo;
+  // 3x3/s1/p1 with power-of-two W (8..64): all-taps halo kernel
+  const bool halo = R == 3 && S == 3 && stride == 1 && pad == 1 &&
+                    Hi == Ho && Wi == Wo && Wo >= 8 && Wo <= 64 &&
+                    (Wo & (Wo - 1)) == 0 && ((long)Ho * Wo) % 64 == 0;
+  if (halo) {
+    int wshift = 0;
+    while ((1 << wshift) < Wo) ++wshift;
+    const long blocks_xy = (KO / 64) * (CI / 64);
+    long nchunks = std::min<long>(std::max<long>(512 / blocks_xy, 1),
+                                  cdiv_l(M, 64));
+    long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), 64) * 64;
+    nchunks = cdiv_l(M, m_per_chunk);
+    const long E = (long)KO * CI * 9;
+    at::Tensor part =
+        nchunks > 1 ? at::empty({nchunks, E}, dw.options()) : dw;
+    dim3 grid(KO / 64, CI / 64, (unsigned)nchunks);
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL(conv_wgrad_halo3<T16>, grid, dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                         N, Hi, Wi, CI, KO, wshift, m_per_chunk);
+    });
+    if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
+    return;
+  }
   const long blocks_xy = (KO / 64) * (R * S * (CI / 64));
   static const long cap = [] {  // ablation knob: MI355X_WGRAD_CHUNKS
     const char* e = getenv("MI355X_WGRAD_CHUNKS");
