@@ -74,3 +74,17 @@ def test_grad_scale_folds_average():
     before = flat.flat_param.clone()
     opt.step()
     torch.testing.assert_close(flat.flat_param, before - 1.0)
+
+
+def test_checkpoint_after_flatten(tmp_path):
+    """state_dict of a flat-param model round-trips (params are views of
+    one shared flat storage; torch.save dedups the storage)."""
+    torch.manual_seed(5)
+    net = Net()
+    FlatState(net)
+    p = tmp_path / "flat_ck.pth"
+    torch.save(net.state_dict(), p)
+    net2 = Net()  # plain (unflattened) model loads the same checkpoint
+    net2.load_state_dict(torch.load(p, weights_only=True))
+    x = torch.randn(2, 3, 32, 32)
+    torch.testing.assert_close(net(x), net2(x))
