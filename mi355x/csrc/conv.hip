@@ -411,7 +411,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   if (smallc_fwd) {
     const long M = (long)N * P * Q;
     const long wrow = w.dim() == 2 ? w.size(1) : (long)R * S * C;
-    long nchunks = std::min<long>(768, cdiv_l(M, 512));
+    long nchunks = std::min<long>(2048, cdiv_l(M, 256));
     const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
     nchunks = cdiv_l(M, m_per_chunk);
     const int has_bias = bias.numel() > 0;
@@ -498,7 +498,8 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
       ((C == 3 && (S == 3 || S == 5 || S == 7)) || (C == 6 && S == 5));
   if (smallc) {
     const long E = (long)K * R * S * C;
-    long nchunks = std::min<long>(std::max<long>(768 / R, 1), cdiv_l(M, 512));
+    long nchunks = std::min<long>(std::max<long>(2048 / R, 1),
+                                  cdiv_l(M, 256));
     const long m_per_chunk = cdiv_l(M, nchunks);
     nchunks = cdiv_l(M, m_per_chunk);
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));

@@ -562,8 +562,15 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   const long M = (long)N * Ho * Wo;
-  // 3x3/s1/p1 with power-of-two W (8..64): all-taps halo kernel
-  const bool halo = R == 3 && S == 3 && stride == 1 && pad == 1 &&
+  // 3x3/s1/p1 with power-of-two W (8..64): all-taps halo kernel.
+  // Measured SLOWER than the tap-per-block kernel (413 vs ~280 us avg:
+  // v236 VGPRs -> 2 waves/SIMD, and the 8-scalar-u16 b-fragment gather is
+  // LDS/VALU-bound) — kept behind MI355X_WGRAD_HALO=1 for further work.
+  static const bool halo_on = [] {
+    const char* e = getenv("MI355X_WGRAD_HALO");
+    return e && e[0] == '1';
+  }();
+  const bool halo = halo_on && R == 3 && S == 3 && stride == 1 && pad == 1 &&
                     Hi == Ho && Wi == Wo && Wo >= 8 && Wo <= 64 &&
                     (Wo & (Wo - 1)) == 0 && ((long)Ho * Wo) % 64 == 0;
   if (halo) {
