@@ -598,7 +598,9 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     const char* e = getenv("MI355X_WGRAD_CHUNKS");
     return e ? atol(e) : 0L;
   }();
-  int nchunks = (int)std::min<long>(std::max<long>(2048 / blocks_xy, 1),
+  // A/B on layer1 (partial slabs): 227ch 189us / 128ch 143 / 64ch 137 —
+  // ~576 total blocks balances fill vs slab+reduce traffic
+  int nchunks = (int)std::min<long>(std::max<long>(576 / blocks_xy, 1),
                                     cdiv_l(M, 512));
   if (cap > 0) nchunks = (int)std::min<long>(nchunks, cap);
   nchunks = std::max(nchunks, 1);
