@@ -598,10 +598,11 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     const char* e = getenv("MI355X_WGRAD_CHUNKS");
     return e ? atol(e) : 0L;
   }();
-  // A/B on layer1 (partial slabs): 227ch 189us / 128ch 143 / 64ch 137 —
-  // ~576 total blocks balances fill vs slab+reduce traffic
-  int nchunks = (int)std::min<long>(std::max<long>(576 / blocks_xy, 1),
-                                    cdiv_l(M, 512));
+  // A/B on layer1 b256 (partial slabs): m/chunk 1216 -> 189us, 4096 ->
+  // 137us, 16384 -> 327us: the lever is PER-CHUNK depth (~4096 m), which
+  // also holds as M grows with batch (576-total-blocks regressed b1024)
+  int nchunks = (int)std::max<long>(
+      std::min<long>(cdiv_l(M, 4096), 4096), 1);
   if (cap > 0) nchunks = (int)std::min<long>(nchunks, cap);
   nchunks = std::max(nchunks, 1);
   long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
