@@ -658,7 +658,9 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int Ho = y.size(1), Wo = y.size(2);
   const long M = (long)N * Ho * Wo;
   const int has_bias = bias.numel() > 0;
-  const bool wide = KO % 128 == 0;  // BN=128 halves barriers per MFMA
+  // BN=128 halves barriers per MFMA but also halves the grid — only use
+  // it when M is large enough to keep the chip full at BM=128 tiles
+  const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
   DISPATCH_16(x, T16, {
     if (wide)
@@ -687,7 +689,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
-  const bool wide = CI % 128 == 0;
+  const bool wide = CI % 128 == 0 && cdiv_l(M, BM) * (CI / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), CI / (wide ? 128 : 64));
   DISPATCH_16(dy, T16, {
     if (wide)
