@@ -130,6 +130,31 @@ def r50_fwd_steps():
     print("forward complete", flush=True)
 
 
+def bnperf():
+    """bn_stats / bn_bwd_reduce timing on the layer1 shape."""
+    import time
+    import mi355x.ops as O
+    x = torch.randn(1024, 32, 32, 64, device="cuda").to(torch.bfloat16)
+    dy = torch.randn_like(x)
+    y = torch.relu(x)
+    s = O.ext().bn_stats(x)
+    mean = s[0] / x.numel() * 64
+    invstd = torch.rsqrt(torch.ones(64, device="cuda"))
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(50):
+        O.ext().bn_stats(x)
+    torch.cuda.synchronize()
+    t1 = time.perf_counter()
+    for _ in range(50):
+        O.ext().bn_bwd_reduce(x, dy, y, mean, invstd)
+    torch.cuda.synchronize()
+    t2 = time.perf_counter()
+    mb = x.numel() * 2 / 1e6
+    print(f"bn_stats {((t1-t0)/50)*1e6:.1f} us ({mb/((t1-t0)/50*1e6)*1e3:.2f} TB/s); "
+          f"bn_bwd_reduce {((t2-t1)/50)*1e6:.1f} us ({3*mb/((t2-t1)/50*1e6)*1e3:.2f} TB/s)")
+
+
 def wgradperf():
     """Per-variant wgrad timing (layer1 shape) with hip events."""
     import mi355x.ops as O
@@ -158,6 +183,8 @@ if __name__ == "__main__":
         conv_cases()
     if what == "wgradperf":
         wgradperf()
+    if what == "bnperf":
+        bnperf()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):

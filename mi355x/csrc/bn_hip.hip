@@ -295,7 +295,11 @@ inline bool bn_fast_ok(long M, int C) {
 }
 
 inline dim3 bn_fast_grid(long E, long& e_per_block) {
-  long blocks = std::min<long>(cdiv_l(E, 2048), 1024);
+  static const long cap = [] {  // ablation knob
+    const char* e = getenv("MI355X_BN_BLOCKS");
+    return e ? atol(e) : 1024L;
+  }();
+  long blocks = std::min<long>(cdiv_l(E, 2048), cap);
   e_per_block = cdiv_l(cdiv_l(E, blocks), 2048) * 2048;
   blocks = cdiv_l(E, e_per_block);
   return dim3((unsigned)blocks);
