@@ -172,7 +172,10 @@ def main():
     images_per_sec = global_batch * args.steps / elapsed
     if rank == 0:
         print(json.dumps({
-            "metric": "images/sec (whole node), ResNet-18 CIFAR-10",
+            "metric": ("images/sec (whole node), ResNet-18 CIFAR-10"
+                       if args.model == "resnet18" and args.size == 32
+                       else f"images/sec (whole node), {args.model} "
+                            f"{args.size}x{args.size}"),
             "value": round(images_per_sec, 1),
             "unit": "images/sec",
             "n_gpus": n_gpus if use_cuda else world,
