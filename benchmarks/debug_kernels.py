@@ -130,6 +130,32 @@ def r50_fwd_steps():
     print("forward complete", flush=True)
 
 
+def convperf():
+    """fwd conv timing per layer shape (glds A/B via MI355X_CONV_GLDS)."""
+    import time
+    import mi355x.ops as O
+    shapes = [  # (N,H,W,C,K,ksz,stride,pad) — r18-CIFAR b1024 layers
+        (1024, 32, 32, 64, 64, 3, 1, 1),
+        (1024, 16, 16, 128, 128, 3, 1, 1),
+        (1024, 8, 8, 256, 256, 3, 1, 1),
+        (1024, 4, 4, 512, 512, 3, 1, 1),
+    ]
+    e = torch.empty(0, device="cuda")
+    for (N, H, W, C, K, ksz, st, pad) in shapes:
+        x = torch.randn(N, H, W, C, device="cuda").to(torch.bfloat16)
+        w = (torch.randn(K, ksz, ksz, C, device="cuda") * 0.1).to(torch.bfloat16)
+        for _ in range(3):
+            O.ext().conv2d_fwd(x, w, e, st, pad, 0, ksz, ksz)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(30):
+            O.ext().conv2d_fwd(x, w, e, st, pad, 0, ksz, ksz)
+        torch.cuda.synchronize()
+        us = (time.perf_counter() - t0) / 30 * 1e6
+        tf = 2.0 * N * H * W * K * ksz * ksz * C / st / st / 1e12 / (us / 1e6)
+        print(f"fwd C={C} K={K} {H}x{W}: {us:.1f} us  ({tf:.0f} TF)")
+
+
 def bnperf():
     """bn_stats / bn_bwd_reduce timing on the layer1 shape."""
     import time
@@ -185,6 +211,8 @@ if __name__ == "__main__":
         wgradperf()
     if what == "bnperf":
         bnperf()
+    if what == "convperf":
+        convperf()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):

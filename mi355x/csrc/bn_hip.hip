@@ -297,7 +297,7 @@ inline bool bn_fast_ok(long M, int C) {
 inline dim3 bn_fast_grid(long E, long& e_per_block) {
   static const long cap = [] {  // ablation knob
     const char* e = getenv("MI355X_BN_BLOCKS");
-    return e ? atol(e) : 1024L;
+    return e ? atol(e) : 512L;  // A/B: 512 blocks 3.4/5.0 TB/s vs 1024 2.9/4.5
   }();
   long blocks = std::min<long>(cdiv_l(E, 2048), cap);
   e_per_block = cdiv_l(cdiv_l(E, blocks), 2048) * 2048;

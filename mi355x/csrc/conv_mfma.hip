@@ -410,6 +410,163 @@ __global__ void wgrad_reduce_chunks(const float* __restrict__ part,
 }  // namespace
 
 // ---------------------------------------------------------------------------
+// glds-pipelined fwd variant: the A (gathered input) tile is staged by
+// `global_load_lds` (direct HBM->LDS DMA, no VGPR round-trip, no ds_write
+// pass — guide §5 ladder step 3) into a double-buffered LINEAR [BM][BK]
+// image with an XOR-swizzled SOURCE address (rule 21: linear dest +
+// inverse-swizzled source + swizzled read), zero-page redirect for
+// padding rows. ONE __syncthreads per k-step: its implicit vmcnt(0)
+// drains the in-flight DMA for the NEXT step's buffer while this step's
+// MFMA runs above it.
+// ---------------------------------------------------------------------------
+namespace {
+
+template <typename T16, int NT>
+__global__ __launch_bounds__(256) void conv_fwd_glds(
+    const T16* __restrict__ in, const T16* __restrict__ wgt,
+    const float* __restrict__ bias, const T16* __restrict__ zpage,
+    T16* __restrict__ out, const int N, const int Hi, const int Wi,
+    const int CI, const int KO, const int Ho, const int Wo, const int R,
+    const int S, const int stride, const int pad, const int act,
+    const int has_bias) {
+  constexpr int BNT = NT * 32;
+  __shared__ T16 lds[2 * BM * BK + 2 * BNT * LDK];
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const long bm0 = (long)blockIdx.x * BM;
+  const int k0 = blockIdx.y * BNT;
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int wm = wave * 32;
+
+  // ---- A rows this lane feeds (4 glds per wave, 8 rows each) ----
+  const int lr = lane >> 3;          // row-within-8 of each glds
+  const int ce = (lane & 7) * 8;     // element col of the 16B piece
+  const int swz = lr * 8;            // XOR swizzle, elements ((r&7)<<4 B)
+  long nbase[4];
+  int ih0v[4], iw0v[4];
+  bool mokv[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int r = 32 * wave + 8 * i + lr;
+    const long m = bm0 + r;
+    mokv[i] = m < Mtot;
+    int n_ = 0, p_ = 0, q_ = 0;
+    if (mokv[i]) {
+      n_ = (int)(m / ((long)Ho * Wo));
+      const int pq = (int)(m % ((long)Ho * Wo));
+      p_ = pq / Wo;
+      q_ = pq % Wo;
+    }
+    nbase[i] = (long)n_ * Hi * Wi * CI;
+    ih0v[i] = p_ * stride - pad;
+    iw0v[i] = q_ * stride - pad;
+  }
+
+  // ---- B staging (register -> LDS, padded LDK rows) ----
+  constexpr int TPR = 256 / BNT;
+  constexpr int EPT = BK / TPR;
+  const int sb_n = tid / TPR;
+  const int sb_c = (tid % TPR) * EPT;
+  const T16* wrow = wgt + (long)(k0 + sb_n) * R * S * CI + sb_c;
+  short8 sb[EPT / 8];
+
+  const int cchunks = CI / BK;
+  const int ksteps = R * S * cchunks;
+
+  auto issue_gldsA = [&](int j, int buf) {
+    const int c0 = (j % cchunks) * BK;
+    const int s_ = (j / cchunks) % S;
+    const int r_ = j / (cchunks * S);
+    T16* dstA = lds + buf * (BM * BK);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int ih = ih0v[i] + r_;
+      const int iw = iw0v[i] + s_;
+      const bool ok = mokv[i] && (unsigned)ih < (unsigned)Hi &&
+                      (unsigned)iw < (unsigned)Wi;
+      const T16* src =
+          ok ? in + nbase[i] + ((long)ih * Wi + iw) * CI + c0 + (ce ^ swz)
+             : zpage + (ce ^ swz);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(dstA +
+                                                    (32 * wave + 8 * i) * BK),
+          16, 0, 0);
+    }
+  };
+  auto load_B = [&](int j) {
+    const int c0 = (j % cchunks) * BK;
+    const int s_ = (j / cchunks) % S;
+    const int r_ = j / (cchunks * S);
+    const T16* wp = wrow + (long)(r_ * S + s_) * CI + c0;
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
+  };
+  auto stage_B = [&](int buf) {
+    short* pb = reinterpret_cast<short*>(lds + 2 * BM * BK + buf * BNT * LDK +
+                                         sb_n * LDK + sb_c);
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      *reinterpret_cast<short8*>(pb + 8 * i) = sb[i];
+  };
+
+  f32x16 acc[NT] = {};
+
+  issue_gldsA(0, 0);
+  load_B(0);
+  stage_B(0);
+  __syncthreads();  // drains glds(0) (implicit vmcnt 0) + B writes
+  for (int j = 0; j < ksteps; ++j) {
+    const int nxt = (j + 1) & 1;
+    if (j + 1 < ksteps) {
+      issue_gldsA(j + 1, nxt);  // DMA runs under this step's MFMA
+      load_B(j + 1);
+    }
+    const T16* ldsA = lds + (j & 1) * (BM * BK);
+    const T16* ldsB = lds + 2 * BM * BK + (j & 1) * BNT * LDK;
+#pragma unroll
+    for (int kk = 0; kk < BK; kk += 16) {
+      const int m = wm + li;
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsA + m * BK + ((kk + kh * 8) ^ ((m & 7) * 8)));
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        const short8 bf = *reinterpret_cast<const short8*>(
+            ldsB + (tnt * 32 + li) * LDK + kk + kh * 8);
+        acc[tnt] = Mfma32<T16>::run(af, bf, acc[tnt]);
+      }
+    }
+    if (j + 1 < ksteps) stage_B(nxt);  // B reg latency hid under the MFMAs
+    __syncthreads();
+  }
+
+  float bv[NT];
+#pragma unroll
+  for (int tnt = 0; tnt < NT; ++tnt)
+    bv[tnt] = has_bias ? bias[k0 + tnt * 32 + li] : 0.f;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mtot) {
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        float v = acc[tnt][reg] + bv[tnt];
+        if (act == 1) v = fmaxf(v, 0.f);
+        out[m_out * KO + k0 + tnt * 32 + li] = F16<T16>::from_f32(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
 // halo wgrad for the dominant 3x3/stride-1/pad-1 convs: ONE block computes
 // all NINE taps of a 64(KO) x 64(CI) dw tile over its m-chunk, so dy and x
 // are staged ONCE per 64-pixel m-step instead of once per tap (the tap-
@@ -657,6 +814,37 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int Ho = y.size(1), Wo = y.size(2);
   const long M = (long)N * Ho * Wo;
   const int has_bias = bias.numel() > 0;
+  static const int glds_mode = [] {  // A/B knob, default decided by measure
+    const char* e = getenv("MI355X_CONV_GLDS");
+    return e ? atoi(e) : 1;
+  }();
+  if (glds_mode) {
+    // 64-element zero page for padding-row DMA redirect
+    static at::Tensor zp16, zph;
+    at::Tensor& zp = x.scalar_type() == at::kBFloat16 ? zp16 : zph;
+    if (!zp.defined()) zp = at::zeros({128}, x.options());
+    const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
+    dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
+    DISPATCH_16(x, T16, {
+      if (wide)
+        hipLaunchKernelGGL((conv_fwd_glds<T16, 4>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N,
+                           Hi, Wi, CI, KO, Ho, Wo, R, S, (int)stride,
+                           (int)pad, (int)act, has_bias);
+      else
+        hipLaunchKernelGGL((conv_fwd_glds<T16, 2>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N,
+                           Hi, Wi, CI, KO, Ho, Wo, R, S, (int)stride,
+                           (int)pad, (int)act, has_bias);
+    });
+    return;
+  }
   // BN=128 halves barriers per MFMA but also halves the grid — only use
   // it when M is large enough to keep the chip full at BM=128 tiles
   const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
