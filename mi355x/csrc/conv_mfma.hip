@@ -814,9 +814,13 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int Ho = y.size(1), Wo = y.size(2);
   const long M = (long)N * Ho * Wo;
   const int has_bias = bias.numel() > 0;
-  static const int glds_mode = [] {  // A/B knob, default decided by measure
+  // A/B (b1024 layer shapes): glds 597/685/561 TF vs register staging
+  // 632/692/665 — glds LOSES at this kernel's 5-blocks/CU occupancy, as
+  // the guide's regime gate predicts (pays only at ~1 block/CU, vgpr>200).
+  // Kept behind MI355X_CONV_GLDS=1 as a documented negative result.
+  static const int glds_mode = [] {
     const char* e = getenv("MI355X_CONV_GLDS");
-    return e ? atoi(e) : 1;
+    return e ? atoi(e) : 0;
   }();
   if (glds_mode) {
     // 64-element zero page for padding-row DMA redirect
