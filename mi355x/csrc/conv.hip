@@ -162,7 +162,8 @@ inline dim3 conv_grid(long total, int block = 256, int cap = 4096) {
 template <typename T16, int C_, int S_>
 __global__ __launch_bounds__(256) void conv_fwd_smallc(
     const T16* __restrict__ x, const T16* __restrict__ wgt,
-    const float* __restrict__ bias, T16* __restrict__ y, int N, int H, int W,
+    const float* __restrict__ bias, const T16* __restrict__ zpage,
+    T16* __restrict__ y, int N, int H, int W,
     int K, int Ho, int Wo, int stride, int pad, long wrow_stride, int act,
     int has_bias, long m_per_chunk, long Mtot) {
   const int k = threadIdx.x & 63;
@@ -198,15 +199,17 @@ __global__ __launch_bounds__(256) void conv_fwd_smallc(
       for (int r = 0; r < S_; ++r) {
         const int ih = ih0 + r;
         const bool rok = (unsigned)ih < (unsigned)H;
-        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : zpage;
 #pragma unroll
         for (int ss = 0; ss < S_; ++ss) {
           const int iw = iw0 + ss;
           const bool ok = rok && (unsigned)iw < (unsigned)W;
+          // address-select (zero page), never a branch around the load:
+          // per-element branchy loads serialize one vmcnt wait each
+          const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
-            xw[(r * S_ + ss) * C_ + c] =
-                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+            xw[(r * S_ + ss) * C_ + c] = F16<T16>::to_f32(src[c]);
         }
       }
       fresh = false;
@@ -235,25 +238,25 @@ __global__ __launch_bounds__(256) void conv_fwd_smallc(
       for (int r = 0; r < S_; ++r) {
         const int ih = ih0 + r;
         const bool rok = (unsigned)ih < (unsigned)H;
-        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+        const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : zpage;
         if (stride == 1) {
           constexpr int ss = S_ - 1;
           const int iw = iw0 + ss;
           const bool ok = rok && (unsigned)iw < (unsigned)W;
+          const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
-            xw[(r * S_ + ss) * C_ + c] =
-                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+            xw[(r * S_ + ss) * C_ + c] = F16<T16>::to_f32(src[c]);
         } else {
 #pragma unroll
           for (int jj = 0; jj < 2; ++jj) {
             const int ss = S_ - 2 + jj;
             const int iw = iw0 + ss;
             const bool ok = rok && (unsigned)iw < (unsigned)W;
+            const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
             for (int c = 0; c < C_; ++c)
-              xw[(r * S_ + ss) * C_ + c] =
-                  ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+              xw[(r * S_ + ss) * C_ + c] = F16<T16>::to_f32(src[c]);
           }
         }
       }
@@ -282,6 +285,7 @@ __global__ __launch_bounds__(256) void conv_fwd_smallc(
 template <typename T16, int C_, int S_>
 __global__ __launch_bounds__(256) void conv_wgrad_smallc(
     const T16* __restrict__ x, const T16* __restrict__ dy,
+    const T16* __restrict__ zpage,
     float* __restrict__ part,  // [gridDim.x*4][K*R*S*C]
     int N, int H, int W, int K, int Ho, int Wo, int R, int stride, int pad,
     long m_per_chunk, long Mtot) {
@@ -306,17 +310,17 @@ __global__ __launch_bounds__(256) void conv_wgrad_smallc(
     for (long m = w0; m < w1; ++m) {
       const int ih = p * stride - pad + r;
       const bool rok = (unsigned)ih < (unsigned)H;
-      const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : nullptr;
+      const T16* row = rok ? x + ((long)n * H + ih) * W * C_ : zpage;
       const int iw0 = q * stride - pad;
       if (fresh || !rok) {
 #pragma unroll
         for (int s = 0; s < S_; ++s) {
           const int iw = iw0 + s;
           const bool ok = rok && (unsigned)iw < (unsigned)W;
+          const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
-            xq[s * C_ + c] =
-                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+            xq[s * C_ + c] = F16<T16>::to_f32(src[c]);
         }
         fresh = false;
       } else {
@@ -331,20 +335,20 @@ __global__ __launch_bounds__(256) void conv_wgrad_smallc(
           constexpr int ss = S_ - 1;
           const int iw = iw0 + ss;
           const bool ok = (unsigned)iw < (unsigned)W;
+          const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
           for (int c = 0; c < C_; ++c)
-            xq[ss * C_ + c] =
-                ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+            xq[ss * C_ + c] = F16<T16>::to_f32(src[c]);
         } else {
 #pragma unroll
           for (int jj = 0; jj < 2; ++jj) {
             const int ss = S_ - 2 + jj;
             const int iw = iw0 + ss;
             const bool ok = (unsigned)iw < (unsigned)W;
+            const T16* src = ok ? row + (long)iw * C_ : zpage;
 #pragma unroll
             for (int c = 0; c < C_; ++c)
-              xq[ss * C_ + c] =
-                  ok ? F16<T16>::to_f32(row[(long)iw * C_ + c]) : 0.f;
+              xq[ss * C_ + c] = F16<T16>::to_f32(src[c]);
           }
         }
       }
@@ -383,6 +387,7 @@ static inline int out_dim(int in, int k, int stride, int pad) {
 
 // conv_mfma.hip — MFMA implicit-GEMM path for C%64==0 && K%64==0
 bool conv_mfma_supported(long CI, long KO);
+at::Tensor conv_zero_page(const at::Tensor& like);
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act);
@@ -415,12 +420,14 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
     nchunks = cdiv_l(M, m_per_chunk);
     const int has_bias = bias.numel() > 0;
+    at::Tensor zp = conv_zero_page(x);
     DISPATCH_16(x, T16, {
       if (C == 3)
         hipLaunchKernelGGL((conv_fwd_smallc<T16, 3, 3>), dim3(nchunks),
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                            has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp.data_ptr(),
                            (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)stride,
                            (int)pad, wrow, (int)act, has_bias, m_per_chunk,
                            M);
@@ -429,6 +436,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                            has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp.data_ptr(),
                            (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)stride,
                            (int)pad, wrow, (int)act, has_bias, m_per_chunk,
                            M);
@@ -504,30 +512,35 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     nchunks = cdiv_l(M, m_per_chunk);
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     auto part = at::zeros({nchunks * 4, E}, x.options().dtype(at::kFloat));
+    at::Tensor zpw = conv_zero_page(x);
     dim3 grid((unsigned)nchunks, R);
     DISPATCH_16(x, T16, {
       if (C == 3 && S == 3)
         hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 3>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)zpw.data_ptr(), part.data_ptr<float>(),
                            N, H, W, K, P, Q, R, (int)stride, (int)pad,
                            m_per_chunk, M);
       else if (C == 3 && S == 5)
         hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 5>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)zpw.data_ptr(), part.data_ptr<float>(),
                            N, H, W, K, P, Q, R, (int)stride, (int)pad,
                            m_per_chunk, M);
       else if (C == 3 && S == 7)
         hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 7>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)zpw.data_ptr(), part.data_ptr<float>(),
                            N, H, W, K, P, Q, R, (int)stride, (int)pad,
                            m_per_chunk, M);
       else
         hipLaunchKernelGGL((conv_wgrad_smallc<T16, 6, 5>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)zpw.data_ptr(), part.data_ptr<float>(),
                            N, H, W, K, P, Q, R, (int)stride, (int)pad,
                            m_per_chunk, M);
     });

@@ -51,6 +51,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
     const float* __restrict__ bias,  // [KO] or null
+    const T16* __restrict__ zpage,   // >=256 zero elements (OOB gather target)
     T16* __restrict__ out,         // [N*Ho*Wo, KO]
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
@@ -116,15 +117,17 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     int s_ = tap - r_ * S;
 #pragma unroll
     for (int i = 0; i < 32; ++i) {
-      short val = 0;
-      if (m_ok && tap < R * S) {
-        const int ih = ih0 + r_;
-        const int iw = iw0 + s_;
-        if ((unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi) {
-          const T16 v = in[(((long)n_ * Hi + ih) * Wi + iw) * CI + c];
-          __builtin_memcpy(&val, &v, 2);
-        }
-      }
+      const int ih = ih0 + r_;
+      const int iw = iw0 + s_;
+      const bool ok = m_ok && tap < R * S && (unsigned)ih < (unsigned)Hi &&
+                      (unsigned)iw < (unsigned)Wi;
+      // address select, not a branch: keeps all 32 loads unconditional so
+      // they batch under one wait instead of 32 serialized round trips
+      const T16* src =
+          ok ? in + (((long)n_ * Hi + ih) * Wi + iw) * CI + c : zpage + i;
+      short val;
+      const T16 v = *src;
+      __builtin_memcpy(&val, &v, 2);
       sa[i / 8][i % 8] = val;
       if (++c == CI) {
         c = 0;
@@ -164,15 +167,10 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       va = va && (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
       if (va) ioff = (((long)n_ * Hi + ih) * Wi + iw) * CI;
     }
-    if (va) {
-      const T16* xp = in + ioff + c0 + sa_c;
+    const T16* xp = va ? in + ioff + c0 + sa_c : zpage;
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
-        sa[i] = *reinterpret_cast<const short8*>(xp + 8 * i);
-    } else {
-#pragma unroll
-      for (int i = 0; i < 4; ++i) sa[i] = short8{};
-    }
+    for (int i = 0; i < 4; ++i)
+      sa[i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
     const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
@@ -707,6 +705,19 @@ bool conv_mfma_supported(long CI, long KO) {
   return CI % 64 == 0 && KO % 64 == 0;
 }
 
+// 256-element zero page: out-of-bounds gathers SELECT this address instead
+// of branching around the load — a per-element `ok ? load : 0` makes hipcc
+// branch around each load and wait vmcnt(0) per element (guide §5 trap 4c:
+// 32 dependent L2 round trips), which measured as ~1.5 us per OUTPUT in
+// the stem kernels.
+at::Tensor conv_zero_page(const at::Tensor& like) {
+  static at::Tensor zp16, zph;
+  at::Tensor& zp = like.scalar_type() == at::kBFloat16 ? zp16 : zph;
+  if (!zp.defined() || zp.device() != like.device())
+    zp = at::zeros({256}, like.options());
+  return zp;
+}
+
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz) {
   const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
   hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(grid_r), dim3(256), 0,
@@ -796,11 +807,13 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)cdiv_l(M, BM), KO / BN);
   const int has_bias = bias.numel() > 0;
+  at::Tensor zp = conv_zero_page(x);
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL((conv_gather_gemm<T16, false, true, 2>), grid,
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
+                       (const T16*)zp.data_ptr(),
                        (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, (int)R,
                        (int)S, (int)stride, (int)pad, KGP, 0, (int)act,
                        has_bias);
@@ -824,10 +837,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     return e ? atoi(e) : 0;
   }();
   if (glds_mode) {
-    // 64-element zero page for padding-row DMA redirect
-    static at::Tensor zp16, zph;
-    at::Tensor& zp = x.scalar_type() == at::kBFloat16 ? zp16 : zph;
-    if (!zp.defined()) zp = at::zeros({128}, x.options());
+    at::Tensor zp = conv_zero_page(x);
     const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
     dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
     DISPATCH_16(x, T16, {
@@ -854,12 +864,14 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   // it when M is large enough to keep the chip full at BM=128 tiles
   const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
+  at::Tensor zp2 = conv_zero_page(x);
   DISPATCH_16(x, T16, {
     if (wide)
       hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4>), grid,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
+                         (const T16*)zp2.data_ptr(),
                          (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
                          (int)act, has_bias);
@@ -868,6 +880,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
+                         (const T16*)zp2.data_ptr(),
                          (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
                          (int)act, has_bias);
@@ -883,12 +896,14 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   const long M = (long)N * H * W;
   const bool wide = CI % 128 == 0 && cdiv_l(M, BM) * (CI / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), CI / (wide ? 128 : 64));
+  at::Tensor zp = conv_zero_page(dy);
   DISPATCH_16(dy, T16, {
     if (wide)
       hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 4>), grid,
                          dim3(256), 0, cur_stream(),
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
+                         (const T16*)zp.data_ptr(),
                          (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
                          (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);
@@ -897,6 +912,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                          dim3(256), 0, cur_stream(),
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
+                         (const T16*)zp.data_ptr(),
                          (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
                          (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);
