@@ -18,6 +18,8 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
                         long pad, long H, long W);
 at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                         long stride, long pad);
+std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
+                                         long stride, long pad);
 
 at::Tensor bn_stats(at::Tensor x);
 at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
@@ -63,6 +65,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_fwd", &conv2d_fwd);
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
+  m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
   m.def("bn_stats", &bn_stats);
   m.def("bn_finalize", &bn_finalize);
   m.def("bn_apply", &bn_apply);

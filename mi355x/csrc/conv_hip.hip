@@ -391,7 +391,8 @@ bool conv_mfma_supported(long CI, long KO);
 at::Tensor conv_zero_page(const at::Tensor& like);
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
-                          at::Tensor y, long stride, long pad, long act);
+                          at::Tensor y, long stride, long pad, long act,
+                          at::Tensor stats);
 void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                                at::Tensor y, long R, long S, long stride,
                                long pad, long act);
@@ -450,7 +451,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   }
   TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
   if (conv_mfma_supported(C, K)) {
-    conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act);
+    conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act, at::Tensor());
     return y;
   }
   const long total = (long)N * P * Q * K;
@@ -464,6 +465,33 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                        (int)stride, (int)pad, (int)act, has_bias);
   });
   return y;
+}
+
+// conv->BN fusion entry: returns {y, stats[2,C]} with stats computed in
+// the conv epilogue (empty stats if this shape has no fusable path — the
+// caller then runs the separate bn_stats pass)
+std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
+                                         long stride, long pad) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_16BIT(x);
+  CHECK_CONTIG(w);
+  const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int K = w.size(0);
+  if (w.dim() != 4 || !conv_mfma_supported(C, K)) {
+    auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
+    auto y = conv2d_fwd(x, w, empty_bias, stride, pad, 0,
+                        w.dim() == 4 ? w.size(1) : 0,
+                        w.dim() == 4 ? w.size(2) : 0);
+    return {y, at::Tensor()};
+  }
+  const int R = w.size(1), S = w.size(2);
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W, S, stride, pad);
+  auto y = at::empty({N, P, Q, K}, x.options());
+  auto stats = at::zeros({2, K}, x.options().dtype(at::kFloat));
+  auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
+  conv_fwd_mfma_launch(x, w, empty_bias, y, stride, pad, 0, stats);
+  return {y, stats};
 }
 
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,

@@ -53,7 +53,8 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const float* __restrict__ bias,  // [KO] or null
     const T16* __restrict__ zpage,   // >=256 zero elements (OOB gather target)
     T16* __restrict__ out,         // [N*Ho*Wo, KO]
-    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    float* __restrict__ stats_slab,  // null, or [gy][gx][2][BNT] partial
+    const int N, const int Hi, const int Wi, const int CI, const int KO,  //  (sum,sumsq) of this block's output tile (conv->BN fusion)
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long b_row_stride, const long b_rs_stride,
     const int act, const int has_bias) {
@@ -211,8 +212,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     }
   }
 
-  // ---- epilogue: bias + act + store ----
+  // ---- epilogue: bias + act + store (+ optional BN-stats partials) ----
   float bv[NT];
+  float ssum[NT] = {}, ssq[NT] = {};
 #pragma unroll
   for (int tnt = 0; tnt < NT; ++tnt)
     bv[tnt] = has_bias ? bias[k0 + tnt * 32 + li] : 0.f;
@@ -225,10 +227,55 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       for (int tnt = 0; tnt < NT; ++tnt) {
         float v = acc[tnt][reg] + bv[tnt];
         if (act == 1) v = fmaxf(v, 0.f);
+        if (stats_slab) {
+          ssum[tnt] += v;
+          ssq[tnt] += v * v;
+        }
         out[m_out * KO + k0 + tnt * 32 + li] = F16<T16>::from_f32(v);
       }
     }
   }
+  if (stats_slab) {
+    // conv->BN fusion: fold the block's per-lane channel partials in LDS
+    // (8 lanes per channel: 4 waves x 2 kh halves) and store one plain
+    // slab entry per channel — replaces BN's separate bn_stats pass
+    constexpr int BNTC = NT * 32;
+    float* lsum = reinterpret_cast<float*>(lds);  // reuse after barrier
+    __syncthreads();
+    for (int t = tid; t < 2 * BNTC; t += 256) lsum[t] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int tnt = 0; tnt < NT; ++tnt) {
+      atomicAdd(lsum + tnt * 32 + li, ssum[tnt]);
+      atomicAdd(lsum + BNTC + tnt * 32 + li, ssq[tnt]);
+    }
+    __syncthreads();
+    float* slab =
+        stats_slab + ((long)blockIdx.y * gridDim.x + blockIdx.x) * 2 * BNTC;
+    for (int t = tid; t < 2 * BNTC; t += 256) slab[t] = lsum[t];
+  }
+}
+
+// stats[2][C] = sum over gx of slab[gy][gx][2][BNT]; one block per
+// (by, gx-slice), coalesced loads, few global atomics
+__global__ void conv_stats_reduce(const float* __restrict__ slab,
+                                  float* __restrict__ stats, int gx, int bnt2,
+                                  int C, int bx_per_block) {
+  const int by = blockIdx.y;
+  const int t = threadIdx.x;
+  if (t >= bnt2) return;
+  float acc = 0.f;
+  const int b0 = blockIdx.x * bx_per_block;
+  const int b1 = min(gx, b0 + bx_per_block);
+  for (int bx = b0; bx < b1; ++bx)
+    acc += slab[((long)by * gx + bx) * bnt2 + t];
+  const int bnt = bnt2 / 2;
+  const int h = t / bnt;
+  const int c = by * bnt + (t % bnt);
+  if (gridDim.x == 1)
+    stats[h * C + c] = acc;
+  else
+    atomicAdd(stats + h * C + c, acc);
 }
 
 }  // namespace
@@ -814,15 +861,19 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
                        (const T16*)zp.data_ptr(),
-                       (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, (int)R,
-                       (int)S, (int)stride, (int)pad, KGP, 0, (int)act,
-                       has_bias);
+                       (T16*)y.data_ptr(), nullptr, N, Hi, Wi, CI, KO, Ho,
+                       Wo, (int)R, (int)S, (int)stride, (int)pad, KGP, 0,
+                       (int)act, has_bias);
   });
 }
 
-// fwd: in = x[N,Hi,Wi,CI], wgt = w16 [KO, R,S,CI] row-major
+// fwd: in = x[N,Hi,Wi,CI], wgt = w16 [KO, R,S,CI] row-major.
+// stats (optional, zeroed [2,C]): conv->BN fusion — the epilogue emits
+// per-block (sum,sumsq) partials to a slab and conv_stats_reduce folds
+// them, replacing BN's separate full-tensor bn_stats pass.
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
-                          at::Tensor y, long stride, long pad, long act) {
+                          at::Tensor y, long stride, long pad, long act,
+                          at::Tensor stats) {
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int KO = w.size(0), R = w.size(1), S = w.size(2);
   const int Ho = y.size(1), Wo = y.size(2);
@@ -836,7 +887,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     const char* e = getenv("MI355X_CONV_GLDS");
     return e ? atoi(e) : 0;
   }();
-  if (glds_mode) {
+  if (glds_mode && !(stats.defined() && stats.numel() > 0)) {
     at::Tensor zp = conv_zero_page(x);
     const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
     dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
@@ -865,6 +916,14 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
   at::Tensor zp2 = conv_zero_page(x);
+  const int bnt2 = 2 * (wide ? 128 : 64);
+  at::Tensor slab;
+  float* stats_slab_ptr = nullptr;
+  if (stats.defined() && stats.numel() > 0) {
+    slab = at::empty({(long)grid.y * grid.x * bnt2},
+                     x.options().dtype(at::kFloat));
+    stats_slab_ptr = slab.data_ptr<float>();
+  }
   DISPATCH_16(x, T16, {
     if (wide)
       hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4>), grid,
@@ -872,7 +931,8 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
                          (const T16*)zp2.data_ptr(),
-                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
+                         (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                         KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
                          (int)act, has_bias);
     else
@@ -881,10 +941,18 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
                          (const T16*)zp2.data_ptr(),
-                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO, Ho, Wo, R, S,
+                         (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                         KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
                          (int)act, has_bias);
   });
+  if (stats_slab_ptr) {
+    const int bx_per_block = (int)cdiv_l(grid.x, 64);
+    dim3 rgrid((unsigned)cdiv_l(grid.x, bx_per_block), grid.y);
+    hipLaunchKernelGGL(conv_stats_reduce, rgrid, dim3(256), 0, cur_stream(),
+                       stats_slab_ptr, stats.data_ptr<float>(), grid.x, bnt2,
+                       KO, bx_per_block);
+  }
 }
 
 // dgrad: in = dy[N,P,Q,KO], wflip = [R,S,CI,KO] (w[k,R-1-r,S-1-s,c]),
@@ -904,8 +972,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
                          (const T16*)zp.data_ptr(),
-                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
-                         (int)S, (int)stride, (int)pad, (long)KO,
+                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
+                         (int)R, (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);
     else
       hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 2>), grid,
@@ -913,8 +981,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
                          (const T16*)zp.data_ptr(),
-                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W, (int)R,
-                         (int)S, (int)stride, (int)pad, (long)KO,
+                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
+                         (int)R, (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);
   });
 }

@@ -89,6 +89,26 @@ class BatchNorm2d(nn.Module):
                               residual, self.act, self.process_group if self.training else None)
 
 
+def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
+    """Fused conv -> BN(+residual+ReLU): on GPU (training, bias-free,
+    act-free conv with an MFMA path) the BN batch statistics come from the
+    conv epilogue, skipping the separate full-tensor bn_stats pass."""
+    if (x.is_cuda and bn.training and conv.bias is None
+            and conv.act is None):
+        from mi355x.ops import functional as F_
+
+        bn.num_batches_tracked += 1
+
+        y, stats = F_.conv2d_with_stats(x, conv.weight, conv.stride,
+                                        conv.padding)
+        if stats.numel() == 0:
+            stats = None
+        return ops.batch_norm(y, bn.weight, bn.bias, bn.running_mean,
+                              bn.running_var, bn.training, bn.momentum,
+                              residual, bn.act, bn.process_group, stats)
+    return bn(conv(x), residual=residual)
+
+
 class MaxPool2d(nn.Module):
     def __init__(self, kernel_size, stride=None, padding=0):
         super().__init__()

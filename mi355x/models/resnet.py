@@ -12,7 +12,8 @@ from __future__ import annotations
 import torch
 from torch import nn
 
-from .layers import BatchNorm2d, Conv2d, Linear, MaxPool2d, to_model_layout
+from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, conv_bn,
+                     to_model_layout)
 from mi355x import ops
 
 
@@ -29,8 +30,8 @@ class BasicBlock(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out), residual=identity)
+        out = conv_bn(self.conv1, self.bn1, x)
+        out = conv_bn(self.conv2, self.bn2, out, residual=identity)
         return out
 
 
@@ -49,9 +50,9 @@ class Bottleneck(nn.Module):
 
     def forward(self, x):
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out), residual=identity)
+        out = conv_bn(self.conv1, self.bn1, x)
+        out = conv_bn(self.conv2, self.bn2, out)
+        out = conv_bn(self.conv3, self.bn3, out, residual=identity)
         return out
 
 
@@ -62,7 +63,7 @@ class Downsample(nn.Module):
         self.bn = BatchNorm2d(out_planes)
 
     def forward(self, x):
-        return self.bn(self.conv(x))
+        return conv_bn(self.conv, self.bn, x)
 
 
 class ResNet(nn.Module):
@@ -98,7 +99,7 @@ class ResNet(nn.Module):
 
     def forward(self, x):
         x = to_model_layout(x)  # NCHW -> NHWC, 16-bit on GPU
-        x = self.bn1(self.conv1(x))
+        x = conv_bn(self.conv1, self.bn1, x)
         if self.maxpool is not None:
             x = self.maxpool(x)
         x = self.layer1(x)

@@ -107,20 +107,35 @@ def _relu_mask_bwd(dy: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
 
 
 class _ConvFn(torch.autograd.Function):
+    """want_stats: the conv epilogue additionally emits BN (sum,sumsq)
+    per-channel statistics of its output (conv->BN fusion — skips the
+    separate bn_stats pass). The stats output is marked non-differentiable:
+    the BN backward formula already carries the full d(stats)/d(x) terms.
+    """
+
     @staticmethod
-    def forward(ctx, x, weight, bias, stride, padding, act):
+    def forward(ctx, x, weight, bias, stride, padding, act, want_stats=False):
         dtype = x.dtype
         w16 = _w16_conv(weight, dtype)
-        b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
-        y = ext().conv2d_fwd(x, w16, b32, stride, padding, act,
-                             weight.shape[2], weight.shape[3])
+        stats = None
+        if want_stats:
+            y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding)
+        else:
+            b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
+            y = ext().conv2d_fwd(x, w16, b32, stride, padding, act,
+                                 weight.shape[2], weight.shape[3])
         ctx.save_for_backward(x, w16, y)
         ctx.weight_ref = weight  # for the cached dgrad weight-flip
         ctx.conf = (stride, padding, act, bias is not None, weight.shape)
+        if want_stats:
+            if stats is None:
+                stats = torch.empty(0, device=x.device, dtype=torch.float32)
+            ctx.mark_non_differentiable(stats)
+            return y, stats
         return y
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, *unused_stats_grad):
         x, w16, y = ctx.saved_tensors
         stride, padding, act, has_bias, wshape = ctx.conf
         dy = dy.contiguous()
@@ -139,7 +154,15 @@ class _ConvFn(torch.autograd.Function):
         db = None
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 1, 2))
-        return dx, dw, db, None, None, None
+        return dx, dw, db, None, None, None, None
+
+
+def conv2d_with_stats(x, weight, stride=1, padding=0):
+    """GPU conv returning (y, bn_stats[2,C] or None) — the fused
+    conv->BN entry used by the ResNet blocks. stats is None when the
+    shape has no stats-emitting kernel path (caller falls back to
+    bn_stats)."""
+    return _ConvFn.apply(x, weight, None, stride, padding, _ACT_NONE, True)
 
 
 def conv2d(x, weight, bias=None, stride=1, padding=0, act=None):
@@ -210,10 +233,11 @@ class _BNFn(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, x, gamma, beta, residual, running_mean, running_var,
-                momentum, act, process_group):
+                momentum, act, process_group, stats=None):
         N, H, W, C = x.shape
         m_local = N * H * W
-        s = ext().bn_stats(x)  # f32 [2,C]: sum, sumsq
+        # stats precomputed by the producing conv's epilogue when fused
+        s = stats if stats is not None else ext().bn_stats(x)
         m_total = m_local
         if process_group is not None:
             import torch.distributed as dist
@@ -255,21 +279,24 @@ class _BNFn(torch.autograd.Function):
             dres = dy  # no activation: residual grad is dy itself
         elif not has_res:
             dres = None
-        return dx, dgamma, dbeta, dres, None, None, None, None, None
+        return dx, dgamma, dbeta, dres, None, None, None, None, None, None
 
 
 def batch_norm(x, gamma, beta, running_mean, running_var, training,
-               momentum=0.1, residual=None, act=None, process_group=None):
+               momentum=0.1, residual=None, act=None, process_group=None,
+               stats=None):
     """BatchNorm over NHWC with optional fused residual-add + ReLU.
 
     Matches torch BN semantics (momentum convention, unbiased running var).
     process_group != None => SyncBatchNorm (stats all-reduced over ranks).
+    stats: precomputed (sum,sumsq) from the producing conv's epilogue.
     """
     a = _ACT_RELU if act == "relu" else _ACT_NONE
     if x.is_cuda:
         if training:
             return _BNFn.apply(x, gamma, beta, residual, running_mean,
-                               running_var, momentum, a, process_group)
+                               running_var, momentum, a, process_group,
+                               stats)
         mean = running_mean.float()
         invstd = torch.rsqrt(running_var.float() + BN_EPS)
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
