@@ -88,3 +88,16 @@ def test_checkpoint_after_flatten(tmp_path):
     net2.load_state_dict(torch.load(p, weights_only=True))
     x = torch.randn(2, 3, 32, 32)
     torch.testing.assert_close(net(x), net2(x))
+
+
+def test_grad_scaler_dynamic():
+    from mi355x.amp import GradScaler
+    sc = GradScaler(init_scale=16.0, growth_interval=2)
+    g = torch.ones(10)
+    loss = torch.tensor(2.0)
+    assert float(sc.scale(loss)) == 32.0
+    assert sc.step_ok(g) and sc.step_ok(g)
+    assert sc.scale_value == 32.0  # doubled after growth_interval clean steps
+    g[3] = float("inf")
+    assert not sc.step_ok(g)
+    assert sc.scale_value == 16.0  # backed off on overflow
