@@ -295,7 +295,9 @@ namespace {
 constexpr int WGM = 64;   // m per step
 constexpr int LDM = 72;   // m-minor row length (+16B: alignment + banks)
 
-template <typename T16>
+// KT = KO-tile width (64 or 128): wider k-tiles amortize the x staging
+// and double the MFMA work per m-step for the K>=128 layers.
+template <typename T16, int KT>
 __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
@@ -303,11 +305,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
     const int pad, const long m_per_chunk, const int nchunks) {
-  __shared__ T16 lds[2 * 64 * LDM];
+  __shared__ T16 lds[(KT + 64) * LDM];
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
-  const int k0 = blockIdx.x * 64;
+  const int k0 = blockIdx.x * KT;
   const int cchunks = CI / BK;
   const int c0 = (blockIdx.y % cchunks) * BK;
   const int s_ = (blockIdx.y / cchunks) % S;
@@ -315,27 +317,34 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   const long m_begin = (long)blockIdx.z * m_per_chunk;
   const long m_end = min(Mtot, m_begin + m_per_chunk);
 
-  // staging assignment: 128 threads per tile, 4 m-rows x 8 elems each
-  const int st_x = tid >= 128;           // which tile this thread stages
-  const int t = tid & 127;
+  // staging: dyT uses 2*KT threads (4 m-rows x 8 k each); xT uses 128
+  // threads — the upper half when KT==64 (disjoint roles), the lower half
+  // when KT==128 (sequential double duty)
+  const bool do_dy = tid < 2 * KT;
+  const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
+  const int t = do_dy ? tid : 0;
   const int sm = (t & 15) * 4;           // m offset (4 rows)
-  const int sk = (t >> 4) * 8;           // element offset within the 64-row
+  const int sk = (t >> 4) * 8;           // k-element offset (dyT rows)
+  const int tx = tid & 127;
+  const int smx = (tx & 15) * 4;
+  const int skx = (tx >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int li = lane & 31;
   const int kh = lane >> 5;
-  const int i0 = (wave & 1) * 32;   // KO half
-  const int j0 = (wave >> 1) * 32;  // kg half
+  constexpr int NJ = KT / 64;       // kg sub-tiles per wave
+  const int i0 = KT == 64 ? (wave & 1) * 32 : wave * 32;  // KO sub-tile
+  const int j0base = KT == 64 ? (wave >> 1) * 32 : 0;
 
-  f32x16 acc = {};
+  f32x16 acc[NJ] = {};
 
   // incremental (n,p,q) decode for the x-gather: one div/mod at entry,
   // add-with-carry as m advances by WGM per step (divisions in the inner
   // loop were ~40% of this kernel's time)
   int dn = 0, dp = 0, dq = 0;
-  if (st_x) {
-    const long m_first = m_begin + sm;
+  {
+    const long m_first = m_begin + smx;
     dn = (int)(m_first / ((long)Ho * Wo));
     const int pq = (int)(m_first % ((long)Ho * Wo));
     dp = pq / Wo;
@@ -353,44 +362,59 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   };
 
   // gather + pack 4 m-rows of 8 elems into registers for m-step m0
-  short8 v[4];
+  short8 vdy[4], vx[4];
   auto load_m = [&](long m0) {
-    int n_ = dn, p_ = dp, q_ = dq;
+    if (do_dy) {
 #pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
-      const long m = m0 + sm + mi;
-      bool ok = m < m_end;
-      long off = 0;
-      if (ok) {
-        if (st_x) {
-          const int ih = p_ * stride - pad + r_;
-          const int iw = q_ * stride - pad + s_;
-          ok = (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
-          if (ok) off = (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
-        } else {
-          off = m * KO + k0 + sk;
-        }
-      }
-      v[mi] = ok ? *reinterpret_cast<const short8*>((st_x ? x : dy) + off)
-                 : short8{};
-      if (st_x && mi < 3 && ++q_ == Wo) {
-        q_ = 0;
-        if (++p_ == Ho) {
-          p_ = 0;
-          ++n_;
-        }
+      for (int mi = 0; mi < 4; ++mi) {
+        const long m = m0 + sm + mi;
+        vdy[mi] = m < m_end ? *reinterpret_cast<const short8*>(
+                                  dy + m * KO + k0 + sk)
+                            : short8{};
       }
     }
-    if (st_x) advance(WGM);  // position for the NEXT m-step's gather
-  };
-  // transpose-write the 4x8 register patch
-  auto stage_m = [&]() {
-    T16* ldsT = lds + (st_x ? 64 * LDM : 0);
+    if (do_x) {
+      int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
-    for (int e = 0; e < 8; ++e) {
-      short4v pk = {v[0][e], v[1][e], v[2][e], v[3][e]};
-      *reinterpret_cast<short4v*>(
-          reinterpret_cast<short*>(ldsT + (sk + e) * LDM + sm)) = pk;
+      for (int mi = 0; mi < 4; ++mi) {
+        const long m = m0 + smx + mi;
+        const int ih = p_ * stride - pad + r_;
+        const int iw = q_ * stride - pad + s_;
+        const bool ok = m < m_end && (unsigned)ih < (unsigned)Hi &&
+                        (unsigned)iw < (unsigned)Wi;
+        vx[mi] = ok ? *reinterpret_cast<const short8*>(
+                          x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 +
+                          skx)
+                    : short8{};
+        if (mi < 3 && ++q_ == Wo) {
+          q_ = 0;
+          if (++p_ == Ho) {
+            p_ = 0;
+            ++n_;
+          }
+        }
+      }
+      advance(WGM);  // position for the NEXT m-step's gather
+    }
+  };
+  // transpose-write the 4x8 register patches
+  auto stage_m = [&]() {
+    if (do_dy) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+      }
+    }
+    if (do_x) {
+      T16* ldsT = lds + KT * LDM;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {vx[0][e], vx[1][e], vx[2][e], vx[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(ldsT + (skx + e) * LDM + smx)) = pk;
+      }
     }
   };
 
@@ -401,14 +425,17 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     __syncthreads();
     if (m0 + WGM < m_end) load_m(m0 + WGM);
     const T16* ldsDyT = lds;
-    const T16* ldsXT = lds + 64 * LDM;
+    const T16* ldsXT = lds + KT * LDM;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
           ldsDyT + (i0 + li) * LDM + kk + kh * 8);
-      const short8 bf = *reinterpret_cast<const short8*>(
-          ldsXT + (j0 + li) * LDM + kk + kh * 8);
-      acc = Mfma32<T16>::run(af, bf, acc);
+#pragma unroll
+      for (int jj = 0; jj < NJ; ++jj) {
+        const short8 bf = *reinterpret_cast<const short8*>(
+            ldsXT + (j0base + jj * 32 + li) * LDM + kk + kh * 8);
+        acc[jj] = Mfma32<T16>::run(af, bf, acc[jj]);
+      }
     }
   }
 
@@ -421,10 +448,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   const long RS = (long)R * S;
   float* slab = dw + (long)blockIdx.z * ((long)KO * CI * RS);
 #pragma unroll
-  for (int reg = 0; reg < 16; ++reg) {
-    const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;  // KO row
-    const int c_abs = c0 + j0 + li;
-    slab[((long)(k0 + i0 + i) * CI + c_abs) * RS + rs] = acc[reg];
+  for (int jj = 0; jj < NJ; ++jj) {
+#pragma unroll
+    for (int reg = 0; reg < 16; ++reg) {
+      const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;  // KO row
+      const int c_abs = c0 + j0base + jj * 32 + li;
+      slab[((long)(k0 + i0 + i) * CI + c_abs) * RS + rs] = acc[jj][reg];
+    }
   }
 }
 
@@ -822,17 +852,25 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   nchunks = std::max(nchunks, 1);
   long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
   nchunks = (int)cdiv_l(M, m_per_chunk);
-  dim3 grid(KO / 64, R * S * (CI / 64), nchunks);
+  const int KT = KO % 128 == 0 ? 128 : 64;
+  dim3 grid(KO / KT, R * S * (CI / 64), nchunks);
   const long E = (long)KO * R * S * CI;
   at::Tensor part = nchunks > 1
                         ? at::empty({nchunks, E}, dw.options())
                         : dw;
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL(conv_wgrad_mfma_kernel<T16>, grid, dim3(256), 0,
-                       cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)dy.data_ptr(), part.data_ptr<float>(), N,
-                       Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S, (int)stride,
-                       (int)pad, m_per_chunk, nchunks);
+    if (KT == 128)
+      hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128>), grid, dim3(256),
+                         0, cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                         (int)stride, (int)pad, m_per_chunk, nchunks);
+    else
+      hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64>), grid, dim3(256),
+                         0, cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                         (int)stride, (int)pad, m_per_chunk, nchunks);
   });
   if (nchunks > 1) {
     const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
