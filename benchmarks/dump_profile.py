@@ -34,7 +34,11 @@ def main():
         f.write("\n\n| kernel | calls | total ms | avg us | % |\n")
         f.write("|---|---|---|---|---|\n")
         for name, calls, ms, avg in rows[:30]:
-            short = name.split("(")[0][:80] if "(" not in name[:5] else name[:80]
+            # demangled form is "void (anonymous namespace)::name<T>(args)";
+            # drop the namespace wrapper before cutting at the arg list so
+            # the kernel's own name survives
+            n = name.replace("(anonymous namespace)::", "")
+            short = n.split("(")[0][:100] if "(" not in n[:6] else n[:100]
             f.write(f"| `{short}` | {calls} | {ms:.2f} | {avg:.1f} | "
                     f"{100*ms/tot:.1f} |\n")
     print(f"wrote {out}: {tot:.1f} ms total")

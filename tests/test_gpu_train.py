@@ -86,3 +86,20 @@ def test_resnet18_fp16_trains_on_gpu():
         losses = _train(build_model("resnet18"), steps=6)
     assert all(l == l for l in losses), losses
     assert losses[-1] < losses[0] * 1.05, losses
+
+
+def test_device_dataloader_prefetch_roundtrip():
+    """The prefetching H2D loader (mi355x/data.py device= path) delivers
+    exactly the dataset's tensors, in sampler order, on the GPU."""
+    from mi355x.data import DataLoader, SyntheticImageDataset
+
+    ds = SyntheticImageDataset(67, seed=9)
+    dl = DataLoader(ds, batch_size=16, shuffle=False, device="cuda")
+    seen_x, seen_y = [], []
+    for x, y in dl:
+        assert x.is_cuda and y.is_cuda
+        seen_x.append(x.cpu())
+        seen_y.append(y.cpu())
+    torch.cuda.synchronize()
+    torch.testing.assert_close(torch.cat(seen_x), ds.data)
+    torch.testing.assert_close(torch.cat(seen_y), ds.targets)
