@@ -156,6 +156,32 @@ def convperf():
         print(f"fwd C={C} K={K} {H}x{W}: {us:.1f} us  ({tf:.0f} TF)")
 
 
+def stemperf():
+    """ImageNet 7x7/s2 stem fwd + wgrad timing (stem7 dot2 kernel vs floor)."""
+    import time
+    import mi355x.ops as O
+    N, H, W, C, K = 256, 224, 224, 3, 64
+    P = Q = 112
+    x = torch.randn(N, H, W, C, device="cuda").to(torch.bfloat16)
+    w = (torch.randn(K, 7, 7, C, device="cuda") * 0.1).to(torch.bfloat16)
+    dy = torch.randn(N, P, Q, K, device="cuda").to(torch.bfloat16)
+    e = torch.empty(0, device="cuda")
+    for tag, fn_ in [
+        ("fwd", lambda: O.ext().conv2d_fwd(x, w, e, 2, 3, 0, 7, 7)),
+        ("wgrad", lambda: O.ext().conv2d_wgrad(x, dy, 7, 7, 2, 3)),
+    ]:
+        for _ in range(3):
+            fn_()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(20):
+            fn_()
+        torch.cuda.synchronize()
+        us = (time.perf_counter() - t0) / 20 * 1e6
+        tf = 2.0 * N * P * Q * K * 49 * C / 1e12 / (us / 1e6)
+        print(f"stem {tag}: {us:.1f} us  ({tf:.1f} TF)")
+
+
 def bnperf():
     """bn_stats / bn_bwd_reduce timing on the layer1 shape."""
     import time
@@ -213,6 +239,8 @@ if __name__ == "__main__":
         bnperf()
     if what == "convperf":
         convperf()
+    if what == "stemperf":
+        stemperf()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):

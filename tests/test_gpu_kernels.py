@@ -44,7 +44,8 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
     ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
-    ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # GENC 7x7/2 stem
+    ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # stem7 7x7/2 (dot2 kernel)
+    ((2, 29, 29, 3), 32, 7, 2, 3, False, None),    # stem7 odd W, K<64
     ((2, 10, 10, 24), 64, 3, 1, 1, False, None),   # GENC C=24
 ])
 def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
