@@ -68,11 +68,9 @@ def main():
         from mi355x import amp
         amp.set_compute_dtype(torch.float16)
 
-    # conv->BN stats fusion measured positive on 224-px configs only
-    # (mi355x/models/layers.py conv_bn docstring has the A/B numbers)
-    if args.size >= 224:
-        os.environ.setdefault("MI355X_FUSE_BN", "1")
-
+    # conv->BN stats fusion (MI355X_FUSE_BN) measured NEGATIVE on both
+    # configs once the stem kernels were fixed (r50-224: 5356 -> 4740
+    # img/s with it on) — leave it opt-in via env only
     num_classes = args.classes or (1000 if args.size >= 224 else 10)
     model_name = args.model if args.size < 224 or args.model != "resnet18" \
         else "resnet18_imagenet"

@@ -468,6 +468,82 @@ __global__ __launch_bounds__(256) void conv_fwd_stem7(
   }
 }
 
+// ---- 7x7/stride-2 stem fwd v3: cooperative LDS staging -----------------
+// v2 (register sliding window) measured latency-bound: each wave had ~1
+// cache line in flight, issued right before use. v3 stages the whole
+// 7-row input strip for one output row COOPERATIVELY (256 threads, fully
+// coalesced, many lines in flight), then the inner loop reads taps from
+// LDS only: per output 77 ds_read_b32 (wave-uniform broadcast, 4B-aligned
+// since the pair stream starts at byte 12*q) + 77 v_dot2c. Weights stay in
+// VGPRs (77 pairs); no circular buffer, no register shifting.
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_fwd_stem7_lds(
+    const T16* __restrict__ x, const T16* __restrict__ wgt,
+    const float* __restrict__ bias, T16* __restrict__ y, int N, int H,
+    int W, int K, int Ho, int Wo, int pad, long wrow_stride, int act,
+    int has_bias) {
+  using V2 = typename Dot2<T16>::v2;
+  constexpr int ROWS = 7, WPAIR = 11;
+  extern __shared__ char smem[];
+  T16* xs = reinterpret_cast<T16*>(smem);
+  const int selems = (2 * Wo + 5) * 3;          // strip elements per row
+  // +4: 8B-aligned row stride with >=1 zeroed pad element (pair 10's hi
+  // half reads element 6q+21, one past the last tap)
+  const int sstride = (selems + 4) & ~3;
+
+  const int p = blockIdx.x % Ho;
+  const int n = blockIdx.x / Ho;
+  const int ih0 = 2 * p - pad;
+
+  // ---- cooperative strip fill: strip col cs = iw + pad ----
+  const T16 z{};
+  for (int i = threadIdx.x; i < ROWS * sstride; i += 256) {
+    const int rr = i / sstride;
+    const int e = i - rr * sstride;
+    const int cs = e / 3, c = e - cs * 3;
+    const int ih = ih0 + rr, iw = cs - pad;
+    const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
+                    (unsigned)iw < (unsigned)W;
+    xs[i] = ok ? x[(((long)n * H + ih) * W + iw) * 3 + c] : z;
+  }
+
+  const int k = threadIdx.x & 63;
+  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+
+  // per-lane packed weights (77 VGPR pairs)
+  V2 wp[ROWS][WPAIR];
+  {
+    const T16* wk = wgt + (long)min(k, K - 1) * wrow_stride;
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      const T16* wr = wk + r * 21;
+#pragma unroll
+      for (int u = 0; u < WPAIR - 1; ++u)
+        wp[r][u] = pack2(wr[2 * u], wr[2 * u + 1]);
+      wp[r][WPAIR - 1] = pack2(wr[20], z);
+    }
+  }
+  const float bk = has_bias ? bias[min(k, K - 1)] : 0.f;
+  __syncthreads();
+
+  const int qper = (Wo + 3) / 4;
+  const int q0 = wv * qper, q1 = min(Wo, q0 + qper);
+  const long mrow = ((long)n * Ho + p) * Wo;
+  for (int q = q0; q < q1; ++q) {
+    float acc = bk;
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r) {
+      const V2* row = reinterpret_cast<const V2*>(
+          reinterpret_cast<const char*>(xs + r * sstride) + 12 * q);
+#pragma unroll
+      for (int u = 0; u < WPAIR; ++u)
+        acc = Dot2<T16>::fma(row[u], wp[r][u], acc);
+    }
+    if (act == 1) acc = fmaxf(acc, 0.f);
+    if (k < K) y[(mrow + q) * K + k] = F16<T16>::from_f32(acc);
+  }
+}
+
 // ---- 7x7/stride-2 stem wgrad (C=3, K<=64) ------------------------------
 // Lane = output channel k; block (chunk, r) accumulates filter row r.
 // dw[k, r, s, c] += dy[m, k] * x[m, tap(r,s,c)] over the block's m-range.
@@ -478,6 +554,91 @@ __global__ __launch_bounds__(256) void conv_fwd_stem7(
 // conv_wgrad_smallc for this shape (measured 5.8 ms/call on the ResNet-50
 // stem: per-output serialized loads + scalar FMAs).
 typedef float float2v __attribute__((ext_vector_type(2)));
+
+// v3: cooperative LDS staging (same diagnosis as the fwd — v2 was
+// latency-bound on the per-output dy line). A block walks a CHUNK of
+// output rows; per row it stages the full 7-row-tall input strip is not
+// needed — all 7 taps' source rows are staged as ONE f32 strip each? No:
+// a single (n,p) output row reads input rows ih0..ih0+6; we stage all 7
+// as f32 pairs (cvt paid once at staging) plus the dy row (bf16,
+// coalesced), then each wave accumulates its q-subrange for ALL 7 filter
+// rows: per output 1 dy ds_read + 7*11 ds_read_b64 (f32 pairs, 8B-aligned
+// at byte 24q) + 7*11 v_pk_fma_f32. dy is read from HBM ONCE (the r-split
+// variant re-read it 7x, latency-bound).
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    float* __restrict__ part,  // [gridDim.x*4][K*7*7*3]
+    int N, int H, int W, int K, int Ho, int Wo, int pad, int rows_per_chunk) {
+  constexpr int ROWS = 7, APAIR = 11;
+  extern __shared__ char smem[];
+  const int selems = (2 * Wo + 5) * 3;
+  const int sstride = (selems + 4) & ~3;        // f32 elems, 8B-aligned +pad
+  float* xs = reinterpret_cast<float*>(smem);
+  T16* dys = reinterpret_cast<T16*>(smem + ROWS * sstride * 4);
+
+  const int k = threadIdx.x & 63;
+  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+  const long nrows = (long)N * Ho;
+  const long row0 = (long)blockIdx.x * rows_per_chunk;
+  const long row1 = min(nrows, row0 + rows_per_chunk);
+
+  float2v acc[ROWS][APAIR] = {};
+
+  const int qper = (Wo + 3) / 4;
+  const int q0 = wv * qper, q1 = min(Wo, q0 + qper);
+
+  for (long row = row0; row < row1; ++row) {
+    const int p = (int)(row % Ho);
+    const int n = (int)(row / Ho);
+    const int ih0 = 2 * p - pad;
+    __syncthreads();  // previous row's reads done before overwrite
+    for (int i = threadIdx.x; i < ROWS * sstride; i += 256) {
+      const int rr = i / sstride;
+      const int e = i - rr * sstride;
+      const int cs = e / 3, c = e - cs * 3;
+      const int ih = ih0 + rr, iw = cs - pad;
+      const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
+                      (unsigned)iw < (unsigned)W;
+      xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
+                 : 0.f;
+    }
+    const T16* dyrow = dy + row * Wo * K;
+    for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    __syncthreads();
+
+    for (int q = q0; q < q1; ++q) {
+      const float dyv = (k < K) ? F16<T16>::to_f32(dys[q * K + k]) : 0.f;
+      float2v d2;
+      d2.x = dyv;
+      d2.y = dyv;
+#pragma unroll
+      for (int r = 0; r < ROWS; ++r) {
+        const float2v* xrow = reinterpret_cast<const float2v*>(
+            reinterpret_cast<const char*>(xs + r * sstride) + 24 * q);
+#pragma unroll
+        for (int u = 0; u < APAIR; ++u) acc[r][u] += d2 * xrow[u];
+      }
+    }
+  }
+
+  if (k < K) {
+    const long E = (long)K * 7 * 7 * 3;
+    float* slab = part + ((long)blockIdx.x * 4 + wv) * E;
+#pragma unroll
+    for (int r = 0; r < ROWS; ++r)
+#pragma unroll
+      for (int u = 0; u < APAIR; ++u) {
+        const int e0 = 2 * u, e1 = 2 * u + 1;
+        const int s0 = e0 / 3, c0 = e0 % 3;
+        slab[(((long)k * 3 + c0) * 7 + r) * 7 + s0] = acc[r][u].x;
+        if (e1 < 21) {
+          const int s1 = e1 / 3, c1 = e1 % 3;
+          slab[(((long)k * 3 + c1) * 7 + r) * 7 + s1] = acc[r][u].y;
+        }
+      }
+  }
+}
 
 template <typename T16>
 __global__ __launch_bounds__(256) void conv_wgrad_stem7(
@@ -742,6 +903,24 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     nchunks = cdiv_l(M, m_per_chunk);
     const int has_bias = bias.numel() > 0;
     at::Tensor zp = conv_zero_page(x);
+    static const bool v2 = [] {  // fallback knob: register-window variant
+      const char* e = getenv("MI355X_STEM7_V2");
+      return e && e[0] == '1';
+    }();
+    if (!v2) {
+      const int selems = (2 * Q + 5) * 3;
+      const int sstride = (selems + 4) & ~3;
+      const size_t smem = (size_t)7 * sstride * x.element_size();
+      DISPATCH_16(x, T16, {
+        hipLaunchKernelGGL((conv_fwd_stem7_lds<T16>), dim3((unsigned)(N * P)),
+                           dim3(256), smem, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)pad,
+                           wrow, (int)act, has_bias);
+      });
+      return y;
+    }
     static const bool ldsw = [] {  // A/B knob: weights in LDS vs VGPR
       const char* e = getenv("MI355X_STEM7_LDSW");
       return !e || e[0] == '1';  // default on (occupancy 2 -> 3)
@@ -895,12 +1074,34 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     const long m_per_chunk = cdiv_l(M, nchunks);
     nchunks = cdiv_l(M, m_per_chunk);
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
-    // empty, not zeros: every slab element is written (acc starts at 0 and
-    // the store runs even for waves with an empty m-range)
-    auto part = at::empty({nchunks * 4, E}, x.options().dtype(at::kFloat));
-    at::Tensor zpw = conv_zero_page(x);
-    dim3 grid((unsigned)nchunks, R);
     if (C == 3 && S == 7 && stride == 2 && pad <= 4) {
+      static const bool v2 = [] {
+        const char* e = getenv("MI355X_STEM7_V2");
+        return e && e[0] == '1';
+      }();
+      if (!v2) {
+        const long nrows = (long)N * P;
+        long nc = std::min<long>(2048, nrows);
+        const int rows_per_chunk = (int)cdiv_l(nrows, nc);
+        nc = cdiv_l(nrows, rows_per_chunk);
+        auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));
+        const int selems = (2 * Q + 5) * 3;
+        const int sstride = (selems + 4) & ~3;
+        const size_t smem = (size_t)7 * sstride * 4 + (size_t)Q * K * 2;
+        DISPATCH_16(x, T16, {
+          hipLaunchKernelGGL((conv_wgrad_stem7_lds<T16>), dim3((unsigned)nc),
+                             dim3(256), smem, cur_stream(),
+                             (const T16*)x.data_ptr(),
+                             (const T16*)dy.data_ptr(),
+                             partl.data_ptr<float>(), N, H, W, K, P, Q,
+                             (int)pad, rows_per_chunk);
+        });
+        wgrad_reduce_launch(partl, dw, E, nc * 4);
+        return dw;
+      }
+      auto part = at::empty({nchunks * 4, E}, x.options().dtype(at::kFloat));
+      at::Tensor zpw = conv_zero_page(x);
+      dim3 grid((unsigned)nchunks, R);
       DISPATCH_16(x, T16, {
         hipLaunchKernelGGL((conv_wgrad_stem7<T16>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
@@ -911,6 +1112,11 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
       wgrad_reduce_launch(part, dw, E, nchunks * 4);
       return dw;
     }
+    // empty, not zeros: every slab element is written (acc starts at 0 and
+    // the store runs even for waves with an empty m-range)
+    auto part = at::empty({nchunks * 4, E}, x.options().dtype(at::kFloat));
+    at::Tensor zpw = conv_zero_page(x);
+    dim3 grid((unsigned)nchunks, R);
     DISPATCH_16(x, T16, {
       if (C == 3 && S == 3)
         hipLaunchKernelGGL((conv_wgrad_smallc<T16, 3, 3>), grid, dim3(256), 0,
