@@ -458,6 +458,35 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
   }
 }
 
+// out[zc][e] = sum over this block-row's z-range of part[z][e] — the
+// z-parallel first level of the two-level reduce (small E can't fill the
+// chip with element-parallelism alone)
+__global__ void wgrad_reduce_zchunk(const float* __restrict__ part,
+                                    float* __restrict__ out, long E, int nz,
+                                    int z_per) {
+  const int z0 = blockIdx.y * z_per;
+  const int z1 = min(nz, z0 + z_per);
+  long i0 = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  const long stride = (long)gridDim.x * blockDim.x * 4;
+  for (long i = i0; i < E; i += stride) {
+    if (i + 4 <= E) {
+      float4v acc = *reinterpret_cast<const float4v*>(part + (long)z0 * E + i);
+      for (int z = z0 + 1; z < z1; ++z) {
+        float4v v = *reinterpret_cast<const float4v*>(part + (long)z * E + i);
+#pragma unroll
+        for (int u = 0; u < 4; ++u) acc[u] += v[u];
+      }
+      *reinterpret_cast<float4v*>(out + (long)blockIdx.y * E + i) = acc;
+    } else {
+      for (long e = i; e < E; ++e) {
+        float a = part[(long)z0 * E + e];
+        for (int z = z0 + 1; z < z1; ++z) a += part[(long)z * E + e];
+        out[(long)blockIdx.y * E + e] = a;
+      }
+    }
+  }
+}
+
 // dw[e] = sum_z part[z][e] (fp32, vectorized)
 __global__ void wgrad_reduce_chunks(const float* __restrict__ part,
                                     float* __restrict__ dw, long E, int nz) {
@@ -795,8 +824,27 @@ at::Tensor conv_zero_page(const at::Tensor& like) {
 }
 
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz) {
-  const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
-  hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(grid_r), dim3(256), 0,
+  // Small E (stem filters: E ~ 9.4K) caps the element-parallel grid at a
+  // handful of blocks — with thousands of slabs that left >95% of the chip
+  // idle on a multi-hundred-MB reduction (measured: the stem wgrad was
+  // ~5.6 ms END-TO-END of which the main kernel was a fraction — this
+  // reduce was the rest). Two-level: z-chunked partial pass into [zc, E],
+  // then the final single-pass sum.
+  const int eblocks = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
+  long zc = std::min<long>(nz / 8, 2048 / std::max(eblocks, 1));
+  if (zc > 1) {
+    auto tmp = at::empty({zc, E}, part.options());
+    const long per = cdiv_l(nz, zc);
+    zc = cdiv_l(nz, per);
+    hipLaunchKernelGGL(wgrad_reduce_zchunk, dim3(eblocks, (unsigned)zc),
+                       dim3(256), 0, cur_stream(), part.data_ptr<float>(),
+                       tmp.data_ptr<float>(), E, (int)nz, (int)per);
+    hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(eblocks), dim3(256), 0,
+                       cur_stream(), tmp.data_ptr<float>(),
+                       dw.data_ptr<float>(), E, (int)zc);
+    return;
+  }
+  hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(eblocks), dim3(256), 0,
                      cur_stream(), part.data_ptr<float>(),
                      dw.data_ptr<float>(), E, (int)nz);
 }
@@ -872,12 +920,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                          N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                          (int)stride, (int)pad, m_per_chunk, nchunks);
   });
-  if (nchunks > 1) {
-    const int grid_r = (int)std::min<long>(cdiv_l(E, 256 * 4), 2048);
-    hipLaunchKernelGGL(wgrad_reduce_chunks, dim3(grid_r), dim3(256), 0,
-                       cur_stream(), part.data_ptr<float>(),
-                       dw.data_ptr<float>(), E, nchunks);
-  }
+  if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
 }
 
 // GENC fwd: wpad = [KO, KGP] zero-padded (cast_permute_krsc_pad)
