@@ -1081,7 +1081,9 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
       }();
       if (!v2) {
         const long nrows = (long)N * P;
-        long nc = std::min<long>(2048, nrows);
+        // 512 blocks x 4 waves at occupancy 2 fills the chip exactly once;
+        // fewer chunks = 4x less partial-slab traffic for the reduce
+        long nc = std::min<long>(512, nrows);
         const int rows_per_chunk = (int)cdiv_l(nrows, nc);
         nc = cdiv_l(nrows, rows_per_chunk);
         auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));

@@ -1,6 +1,13 @@
 // NHWC pooling kernels: maxpool fwd (+argmax indices),
-// maxpool bwd (index scatter; fp32 accumulation for overlapping windows),
-// and global average pool.
+// maxpool bwd, and global average pool.
+//
+// Argmax indices are stored as the RELATIVE window position r*kw+s in ONE
+// byte (torch stores absolute int32 indices — 4x the traffic), and the
+// backward is a GATHER: each dx element scans the <=ceil(k/stride)^2
+// windows that could have picked it and sums matching dy. No fp32 scratch
+// buffer, no zero-init pass, no atomics, no final cast (the old scatter
+// variant measured 1.18 ms on the ResNet-50 stem pool; the gather floor
+// is ~80 us of pure streaming).
 #include "common.h"
 
 namespace {
@@ -8,7 +15,7 @@ namespace {
 template <typename T16>
 __global__ void maxpool_fwd_kernel(const T16* __restrict__ x,
                                    T16* __restrict__ y,
-                                   int* __restrict__ idx, int N, int H, int W,
+                                   unsigned char* __restrict__ idx, int N, int H, int W,
                                    int C, int P, int Q, int kh, int kw,
                                    int stride, int pad) {
   const long total = (long)N * P * Q * C;
@@ -21,7 +28,7 @@ __global__ void maxpool_fwd_kernel(const T16* __restrict__ x,
     const int p = (int)(np % P);
     const int n = (int)(np / P);
     float best = -3.4e38f;
-    int best_hw = 0;
+    int best_rs = 0;
     for (int r = 0; r < kh; ++r) {
       const int ih = p * stride - pad + r;
       if (ih < 0 || ih >= H) continue;
@@ -32,31 +39,47 @@ __global__ void maxpool_fwd_kernel(const T16* __restrict__ x,
             F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * C + c]);
         if (v > best) {
           best = v;
-          best_hw = ih * W + iw;
+          best_rs = r * kw + s;
         }
       }
     }
     y[t] = F16<T16>::from_f32(best);
-    idx[t] = best_hw;
+    idx[t] = (unsigned char)best_rs;
   }
 }
 
+// gather backward: dx[n,ih,iw,c] = sum over candidate windows (p,q) of
+// dy[n,p,q,c] where the stored relative argmax points back at (ih,iw)
 template <typename T16>
 __global__ void maxpool_bwd_kernel(const T16* __restrict__ dy,
-                                   const int* __restrict__ idx,
-                                   float* __restrict__ dxf, int N, int H,
-                                   int W, int C, int P, int Q, int overlap) {
-  const long total = (long)N * P * Q * C;
+                                   const unsigned char* __restrict__ idx,
+                                   T16* __restrict__ dx, int N, int H, int W,
+                                   int C, int P, int Q, int kh, int kw,
+                                   int stride, int pad) {
+  const long total = (long)N * H * W * C;
   for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
        t += (long)gridDim.x * blockDim.x) {
     const int c = (int)(t % C);
-    const int n = (int)(t / ((long)P * Q * C));
-    const long dst = ((long)n * H * W + idx[t]) * C + c;
-    const float v = F16<T16>::to_f32(dy[t]);
-    if (overlap)
-      atomicAdd(dxf + dst, v);
-    else
-      dxf[dst] = v;
+    long nhw = t / C;
+    const int iw = (int)(nhw % W);
+    long nh = nhw / W;
+    const int ih = (int)(nh % H);
+    const int n = (int)(nh / H);
+    // windows (p,q) with p*stride-pad <= ih <= p*stride-pad+kh-1
+    const int p1 = min(P - 1, (ih + pad) / stride);
+    const int p0 = max(0, (ih + pad - kh + stride) / stride);
+    const int q1 = min(Q - 1, (iw + pad) / stride);
+    const int q0 = max(0, (iw + pad - kw + stride) / stride);
+    float acc = 0.f;
+    for (int p = p0; p <= p1; ++p) {
+      const int r = ih - (p * stride - pad);
+      for (int q = q0; q <= q1; ++q) {
+        const int s = iw - (q * stride - pad);
+        const long o = (((long)n * P + p) * Q + q) * C + c;
+        if (idx[o] == r * kw + s) acc += F16<T16>::to_f32(dy[o]);
+      }
+    }
+    dx[t] = F16<T16>::from_f32(acc);
   }
 }
 
@@ -90,13 +113,14 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
   const int P = (H + 2 * pad - kernel) / stride + 1;
   const int Q = (W + 2 * pad - kernel) / stride + 1;
   auto y = at::empty({N, P, Q, C}, x.options());
-  auto idx = at::empty({N, P, Q, C}, x.options().dtype(at::kInt));
+  auto idx = at::empty({N, P, Q, C}, x.options().dtype(at::kByte));
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(maxpool_fwd_kernel<T16>,
                        dim3(pgrid((long)N * P * Q * C)), dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
-                       (T16*)y.data_ptr(), idx.data_ptr<int>(), N, H, W, C, P,
-                       Q, (int)kernel, (int)kernel, (int)stride, (int)pad);
+                       (T16*)y.data_ptr(), idx.data_ptr<unsigned char>(), N,
+                       H, W, C, P, Q, (int)kernel, (int)kernel, (int)stride,
+                       (int)pad);
   });
   return {y, idx};
 }
@@ -106,16 +130,16 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
-  const int overlap = stride < kernel;
-  auto dxf = at::zeros({N, H, W, (long)C}, dy.options().dtype(at::kFloat));
+  auto dx = at::empty({N, H, W, (long)C}, dy.options());
   DISPATCH_16(dy, T16, {
     hipLaunchKernelGGL(maxpool_bwd_kernel<T16>,
-                       dim3(pgrid((long)N * P * Q * C)), dim3(256), 0,
+                       dim3(pgrid((long)N * H * W * C)), dim3(256), 0,
                        cur_stream(), (const T16*)dy.data_ptr(),
-                       idx.data_ptr<int>(), dxf.data_ptr<float>(), N, (int)H,
-                       (int)W, C, P, Q, overlap);
+                       idx.data_ptr<unsigned char>(), (T16*)dx.data_ptr(), N,
+                       (int)H, (int)W, C, P, Q, (int)kernel, (int)kernel,
+                       (int)stride, (int)pad);
   });
-  return cast_to_16(dxf, dy);
+  return dx;
 }
 
 at::Tensor global_avg_pool(at::Tensor x) {
