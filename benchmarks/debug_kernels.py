@@ -156,6 +156,31 @@ def convperf():
         print(f"fwd C={C} K={K} {H}x{W}: {us:.1f} us  ({tf:.0f} TF)")
 
 
+def membw():
+    """HBM bandwidth calibration: what do simple streams actually reach?
+    Gives the real 'floor' to price memory-bound kernels against."""
+    import time
+    n = 256 * 1024 * 1024  # 512 MB bf16
+    a = torch.randn(n, device="cuda").to(torch.bfloat16)
+    b = torch.empty_like(a)
+    idx = torch.zeros(n // 8, dtype=torch.uint8, device="cuda")
+    for tag, fn_, bytes_ in [
+        ("copy (torch)", lambda: b.copy_(a), 2 * n * 2),
+        ("add (torch)", lambda: torch.add(a, a, out=b), 3 * n * 2),
+        ("u8 read + bf16 write", lambda: b[: n // 8].copy_(idx.to(torch.bfloat16)),
+         idx.numel() * 3),
+    ]:
+        for _ in range(3):
+            fn_()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            fn_()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        print(f"membw {tag}: {bytes_ / dt / 1e12:.2f} TB/s")
+
+
 def stemperf():
     """ImageNet 7x7/s2 stem fwd + wgrad timing (stem7 dot2 kernel vs floor)."""
     import time
@@ -180,6 +205,38 @@ def stemperf():
         us = (time.perf_counter() - t0) / 20 * 1e6
         tf = 2.0 * N * P * Q * K * 49 * C / 1e12 / (us / 1e6)
         print(f"stem {tag}: {us:.1f} us  ({tf:.1f} TF)")
+
+
+def pmcprobe():
+    """Run each hot kernel ~10x in one process for a rocprofv3 --pmc pass
+    (counters aggregate per kernel symbol)."""
+    import mi355x.ops as O
+    e = torch.empty(0, device="cuda")
+    # stem shapes (r50-224 b256)
+    x7 = torch.randn(256, 224, 224, 3, device="cuda").to(torch.bfloat16)
+    w7 = (torch.randn(64, 7, 7, 3, device="cuda") * 0.1).to(torch.bfloat16)
+    dy7 = torch.randn(256, 112, 112, 64, device="cuda").to(torch.bfloat16)
+    # maxpool (r50 stem pool)
+    xp = torch.randn(256, 112, 112, 64, device="cuda").to(torch.bfloat16)
+    # bn apply (r50 layer1-ish)
+    xb = torch.randn(256, 56, 56, 256, device="cuda").to(torch.bfloat16)
+    g1 = torch.randn(256, device="cuda")
+    # layer1 wgrad MFMA (r18 shape)
+    xw = torch.randn(256, 32, 32, 64, device="cuda").to(torch.bfloat16)
+    dyw = torch.randn(256, 32, 32, 64, device="cuda").to(torch.bfloat16)
+    yp, ip = O.ext().maxpool_fwd(xp, 3, 2, 1)
+    st = O.ext().bn_stats(xb)
+    mean = st[0] / xb.numel() * 256
+    inv = torch.rsqrt((st[1] / (xb.numel() / 256) - mean * mean).clamp_min(1e-5))
+    for _ in range(10):
+        O.ext().conv2d_fwd(x7, w7, e, 2, 3, 0, 7, 7)
+        O.ext().conv2d_wgrad(x7, dy7, 7, 7, 2, 3)
+        O.ext().maxpool_fwd(xp, 3, 2, 1)
+        O.ext().maxpool_bwd(yp.clone(), ip, 112, 112, 3, 2, 1)
+        O.ext().bn_apply(xb, mean, inv, g1, torch.zeros_like(g1), e, 1)
+        O.ext().conv2d_wgrad(xw, dyw, 3, 3, 1, 1)
+    torch.cuda.synchronize()
+    print("pmcprobe done")
 
 
 def bnperf():
@@ -241,6 +298,10 @@ if __name__ == "__main__":
         convperf()
     if what == "stemperf":
         stemperf()
+    if what == "membw":
+        membw()
+    if what == "pmcprobe":
+        pmcprobe()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):
