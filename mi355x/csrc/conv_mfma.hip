@@ -104,19 +104,13 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
   const int ksteps = GENC ? (int)(b_row_stride / BK)  // KGP/64
                           : R * S * cchunks;
 
-  // TWO staging register sets: loads issue two k-steps ahead of their
-  // ds_write so the vmcnt wait at the staging barrier is covered by ~2
-  // steps of MFMA work (same diagnosis as the wgrad kernel: PMC showed
-  // the 1-deep version parked on WAIT_ANY for most of its cycles)
-  short8 sa[2][4];         // 32 channels = 4 x 16B
-  short8 sb[2][EPT / 8];   // B elements per thread
+  short8 sa[4];            // 32 channels = 4 x 16B
+  short8 sb[EPT / 8];      // B elements per thread
 
-  auto load_step_genc = [&](int j, int buf) {
+  auto load_step_genc = [&](int j) {
     // per-element gather: kg walks (tap, c) with carries (fully unrolled,
     // so the vector-register indices stay compile-time — guide rule 20)
     const int kg0 = j * BK + sa_c;
-    short8* sav = sa[buf];
-    short8* sbv = sb[buf];
     int tap = kg0 / CI;
     int c = kg0 - tap * CI;
     int r_ = tap / S;
@@ -134,7 +128,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
       short val;
       const T16 v = *src;
       __builtin_memcpy(&val, &v, 2);
-      sav[i / 8][i % 8] = val;
+      sa[i / 8][i % 8] = val;
       if (++c == CI) {
         c = 0;
         ++tap;
@@ -147,12 +141,12 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const T16* wp = wgt + (long)(k0 + sb_n) * b_row_stride + j * BK + sb_c;
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
-      sbv[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
   };
 
-  auto load_step = [&](int j, int buf) {
+  auto load_step = [&](int j) {
     if constexpr (GENC) {
-      load_step_genc(j, buf);
+      load_step_genc(j);
       return;
     }
     const int c0 = (j % cchunks) * BK;
@@ -176,34 +170,32 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(
     const T16* xp = va ? in + ioff + c0 + sa_c : zpage;
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      sa[buf][i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
+      sa[i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
     const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
-      sb[buf][i] = *reinterpret_cast<const short8*>(wp + 8 * i);
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
   };
 
-  auto stage = [&](int buf) {
+  auto stage = [&]() {
     T16* ldsA = lds;
     T16* ldsB = lds + BM * LDK;
     short* pa = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      *reinterpret_cast<short8*>(pa + 8 * i) = sa[buf][i];
+      *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
     short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
-      *reinterpret_cast<short8*>(pb + 8 * i) = sb[buf][i];
+      *reinterpret_cast<short8*>(pb + 8 * i) = sb[i];
   };
 
-  load_step(0, 0);
-  if (1 < ksteps) load_step(1, 1);
+  load_step(0);
   for (int j = 0; j < ksteps; ++j) {
-    const int pb_ = j & 1;
     __syncthreads();  // previous MFMA phase done reading LDS
-    stage(pb_);
+    stage();
     __syncthreads();
-    if (j + 2 < ksteps) load_step(j + 2, pb_);  // overlaps the MFMA phase
+    if (j + 1 < ksteps) load_step(j + 1);  // overlaps the MFMA phase
     const T16* ldsA = lds;
     const T16* ldsB = lds + BM * LDK;
 #pragma unroll
@@ -369,21 +361,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     }
   };
 
-  // gather + pack 4 m-rows of 8 elems into registers for m-step m0.
-  // TWO register buffers: loads are issued two m-steps ahead of their
-  // ds_write, so the vmcnt wait before staging is covered by ~2 full
-  // steps of MFMA + staging instead of one kk-loop (PMC: 66% WAIT_ANY on
-  // the 1-deep version — the gather's HBM latency was almost entirely
-  // exposed at the staging barrier).
-  short8 vdy[2][4], vx[2][4];
-  auto load_m = [&](long m0, int buf) {
+  // gather + pack 4 m-rows of 8 elems into registers for m-step m0
+  short8 vdy[4], vx[4];
+  auto load_m = [&](long m0) {
     if (do_dy) {
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const long m = m0 + sm + mi;
-        vdy[buf][mi] = m < m_end ? *reinterpret_cast<const short8*>(
-                                       dy + m * KO + k0 + sk)
-                                 : short8{};
+        vdy[mi] = m < m_end ? *reinterpret_cast<const short8*>(
+                                  dy + m * KO + k0 + sk)
+                            : short8{};
       }
     }
     if (do_x) {
@@ -395,10 +382,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
         const int iw = q_ * stride - pad + s_;
         const bool ok = m < m_end && (unsigned)ih < (unsigned)Hi &&
                         (unsigned)iw < (unsigned)Wi;
-        vx[buf][mi] = ok ? *reinterpret_cast<const short8*>(
-                               x + (((long)n_ * Hi + ih) * Wi + iw) * CI +
-                               c0 + skx)
-                         : short8{};
+        vx[mi] = ok ? *reinterpret_cast<const short8*>(
+                          x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 +
+                          skx)
+                    : short8{};
         if (mi < 3 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -411,12 +398,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
     }
   };
   // transpose-write the 4x8 register patches
-  auto stage_m = [&](int buf) {
+  auto stage_m = [&]() {
     if (do_dy) {
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        short4v pk = {vdy[buf][0][e], vdy[buf][1][e], vdy[buf][2][e],
-                      vdy[buf][3][e]};
+        short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
         *reinterpret_cast<short4v*>(
             reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
       }
@@ -425,22 +411,19 @@ __global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
       T16* ldsT = lds + KT * LDM;
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
-        short4v pk = {vx[buf][0][e], vx[buf][1][e], vx[buf][2][e],
-                      vx[buf][3][e]};
+        short4v pk = {vx[0][e], vx[1][e], vx[2][e], vx[3][e]};
         *reinterpret_cast<short4v*>(
             reinterpret_cast<short*>(ldsT + (skx + e) * LDM + smx)) = pk;
       }
     }
   };
 
-  load_m(m_begin, 0);
-  if (m_begin + WGM < m_end) load_m(m_begin + WGM, 1);
-  int pb = 0;
-  for (long m0 = m_begin; m0 < m_end; m0 += WGM, pb ^= 1) {
+  load_m(m_begin);
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
     __syncthreads();
-    stage_m(pb);
+    stage_m();
     __syncthreads();
-    if (m0 + 2 * WGM < m_end) load_m(m0 + 2 * WGM, pb);
+    if (m0 + WGM < m_end) load_m(m0 + WGM);
     const T16* ldsDyT = lds;
     const T16* ldsXT = lds + KT * LDM;
 #pragma unroll
