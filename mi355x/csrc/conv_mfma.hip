@@ -46,7 +46,8 @@ constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
 // [KO, KGP=ceil(R*S*CI/64)*64] with kg=(r*S+s)*CI+c and zeros beyond, and
 // the A side gathers per-ELEMENT across tap boundaries (stem convs C=3/6).
 template <typename T16, bool TRANS, bool GENC, int NT>
-__global__ __launch_bounds__(256) void conv_gather_gemm(
+// min-blocks hint: same effect as on the wgrad kernel (see comment there)
+__global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
     const float* __restrict__ bias,  // [KO] or null
@@ -298,7 +299,11 @@ constexpr int LDM = 72;   // m-minor row length (+16B: alignment + banks)
 // KT = KO-tile width (64 or 128): wider k-tiles amortize the x staging
 // and double the MFMA work per m-step for the K>=128 layers.
 template <typename T16, int KT>
-__global__ __launch_bounds__(256) void conv_wgrad_mfma_kernel(
+// the second launch-bounds arg (min 2 blocks/CU) is load-bearing: without
+// it the compiler allocates 188-256 VGPRs (occupancy 1-2) for no speedup;
+// with it 104-124 VGPRs, zero spills, occupancy 4 - the extra waves are
+// what covers the gather's HBM latency (PMC: 66% WAIT_ANY at occ 2)
+__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
     float* __restrict__ dw,       // [KO, R*S*CI]

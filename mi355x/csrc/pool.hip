@@ -12,6 +12,66 @@
 
 namespace {
 
+// C%8 fast path: one thread owns 8 channels of one output; window taps
+// are short8 loads (16B), argmax kept per sub-lane, indices stored as one
+// 8-byte pack. 8x fewer memory instructions than the scalar variant
+// (PMC: scalar version sat at 77% WAIT_ANY on 2B accesses).
+template <typename T16>
+__global__ void maxpool_fwd_vec8(const T16* __restrict__ x,
+                                 T16* __restrict__ y,
+                                 unsigned char* __restrict__ idx, int N,
+                                 int H, int W, int C, int P, int Q, int kh,
+                                 int kw, int stride, int pad) {
+  const long total = (long)N * P * Q * (C / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(t % (C / 8));
+    long npq = t / (C / 8);
+    const int q = (int)(npq % Q);
+    long np = npq / Q;
+    const int p = (int)(np % P);
+    const int n = (int)(np / P);
+    float best[8];
+    int rs[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      best[u] = -3.4e38f;
+      rs[u] = 0;
+    }
+    for (int r = 0; r < kh; ++r) {
+      const int ih = p * stride - pad + r;
+      if (ih < 0 || ih >= H) continue;
+      for (int sx = 0; sx < kw; ++sx) {
+        const int iw = q * stride - pad + sx;
+        if (iw < 0 || iw >= W) continue;
+        const short8 v = *reinterpret_cast<const short8*>(
+            x + (((long)n * H + ih) * W + iw) * C + c8 * 8);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const float f = s16_to_f32<T16>(v[u]);
+          if (f > best[u]) {
+            best[u] = f;
+            rs[u] = r * kw + sx;
+          }
+        }
+      }
+    }
+    short8 yv;
+    unsigned char iv[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      T16 h = F16<T16>::from_f32(best[u]);
+      short hs;
+      __builtin_memcpy(&hs, &h, 2);
+      yv[u] = hs;
+      iv[u] = (unsigned char)rs[u];
+    }
+    const long o = ((((long)n * P + p) * Q + q) * C + c8 * 8);
+    *reinterpret_cast<short8*>(y + o) = yv;
+    __builtin_memcpy(idx + o, iv, 8);
+  }
+}
+
 template <typename T16>
 __global__ void maxpool_fwd_kernel(const T16* __restrict__ x,
                                    T16* __restrict__ y,
@@ -45,6 +105,54 @@ __global__ void maxpool_fwd_kernel(const T16* __restrict__ x,
     }
     y[t] = F16<T16>::from_f32(best);
     idx[t] = (unsigned char)best_rs;
+  }
+}
+
+// C%8 fast path of the gather backward
+template <typename T16>
+__global__ void maxpool_bwd_vec8(const T16* __restrict__ dy,
+                                 const unsigned char* __restrict__ idx,
+                                 T16* __restrict__ dx, int N, int H, int W,
+                                 int C, int P, int Q, int kh, int kw,
+                                 int stride, int pad) {
+  const long total = (long)N * H * W * (C / 8);
+  for (long t = (long)blockIdx.x * blockDim.x + threadIdx.x; t < total;
+       t += (long)gridDim.x * blockDim.x) {
+    const int c8 = (int)(t % (C / 8));
+    long nhw = t / (C / 8);
+    const int iw = (int)(nhw % W);
+    long nh = nhw / W;
+    const int ih = (int)(nh % H);
+    const int n = (int)(nh / H);
+    const int p1 = min(P - 1, (ih + pad) / stride);
+    const int p0 = max(0, (ih + pad - kh + stride) / stride);
+    const int q1 = min(Q - 1, (iw + pad) / stride);
+    const int q0 = max(0, (iw + pad - kw + stride) / stride);
+    float acc[8] = {};
+    for (int p = p0; p <= p1; ++p) {
+      const int r = ih - (p * stride - pad);
+      for (int q = q0; q <= q1; ++q) {
+        const int sx = iw - (q * stride - pad);
+        const long o = (((long)n * P + p) * Q + q) * C + c8 * 8;
+        unsigned char iv[8];
+        __builtin_memcpy(iv, idx + o, 8);
+        const short8 dv = *reinterpret_cast<const short8*>(dy + o);
+        const unsigned char want = (unsigned char)(r * kw + sx);
+#pragma unroll
+        for (int u = 0; u < 8; ++u)
+          if (iv[u] == want) acc[u] += s16_to_f32<T16>(dv[u]);
+      }
+    }
+    short8 ov;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      T16 h = F16<T16>::from_f32(acc[u]);
+      short hs;
+      __builtin_memcpy(&hs, &h, 2);
+      ov[u] = hs;
+    }
+    *reinterpret_cast<short8*>(
+        dx + ((((long)n * H + ih) * W + iw) * C + c8 * 8)) = ov;
   }
 }
 
@@ -115,12 +223,20 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
   auto y = at::empty({N, P, Q, C}, x.options());
   auto idx = at::empty({N, P, Q, C}, x.options().dtype(at::kByte));
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL(maxpool_fwd_kernel<T16>,
-                       dim3(pgrid((long)N * P * Q * C)), dim3(256), 0,
-                       cur_stream(), (const T16*)x.data_ptr(),
-                       (T16*)y.data_ptr(), idx.data_ptr<unsigned char>(), N,
-                       H, W, C, P, Q, (int)kernel, (int)kernel, (int)stride,
-                       (int)pad);
+    if (C % 8 == 0)
+      hipLaunchKernelGGL(maxpool_fwd_vec8<T16>,
+                         dim3(pgrid((long)N * P * Q * (C / 8))), dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         (T16*)y.data_ptr(), idx.data_ptr<unsigned char>(), N,
+                         H, W, C, P, Q, (int)kernel, (int)kernel, (int)stride,
+                         (int)pad);
+    else
+      hipLaunchKernelGGL(maxpool_fwd_kernel<T16>,
+                         dim3(pgrid((long)N * P * Q * C)), dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         (T16*)y.data_ptr(), idx.data_ptr<unsigned char>(), N,
+                         H, W, C, P, Q, (int)kernel, (int)kernel, (int)stride,
+                         (int)pad);
   });
   return {y, idx};
 }
@@ -132,12 +248,20 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
   DISPATCH_16(dy, T16, {
-    hipLaunchKernelGGL(maxpool_bwd_kernel<T16>,
-                       dim3(pgrid((long)N * H * W * C)), dim3(256), 0,
-                       cur_stream(), (const T16*)dy.data_ptr(),
-                       idx.data_ptr<unsigned char>(), (T16*)dx.data_ptr(), N,
-                       (int)H, (int)W, C, P, Q, (int)kernel, (int)kernel,
-                       (int)stride, (int)pad);
+    if (C % 8 == 0)
+      hipLaunchKernelGGL(maxpool_bwd_vec8<T16>,
+                         dim3(pgrid((long)N * H * W * (C / 8))), dim3(256), 0,
+                         cur_stream(), (const T16*)dy.data_ptr(),
+                         idx.data_ptr<unsigned char>(), (T16*)dx.data_ptr(),
+                         N, (int)H, (int)W, C, P, Q, (int)kernel, (int)kernel,
+                         (int)stride, (int)pad);
+    else
+      hipLaunchKernelGGL(maxpool_bwd_kernel<T16>,
+                         dim3(pgrid((long)N * H * W * C)), dim3(256), 0,
+                         cur_stream(), (const T16*)dy.data_ptr(),
+                         idx.data_ptr<unsigned char>(), (T16*)dx.data_ptr(),
+                         N, (int)H, (int)W, C, P, Q, (int)kernel, (int)kernel,
+                         (int)stride, (int)pad);
   });
   return dx;
 }
