@@ -463,6 +463,182 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
   }
 }
 
+// s-grouped wgrad for the dominant 3x3/stride-1/pad-1 convs: ONE block
+// computes all 3 s-taps of one filter row r from ONE dy+x stage. The
+// three x operand images are m-SHIFTED copies of the same center gather
+// (x_im2col[m, s] = x_center[m + s - 1] within a q-row): each staging
+// thread gathers 6 m-positions (4 + one halo on each side) and packs the
+// three aligned 4-m transposed writes from its own registers, zeroing
+// slots that cross a q-row boundary (q==0 for s=0, q==Wo-1 for s=2).
+// Triples the MFMA work per staged dy/x byte — the plain kernel was
+// staging-bound at ~190 TF while the fwd gather-GEMM reaches 470-670.
+template <typename T16, int KT>
+__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
+    const T16* __restrict__ x,    // [N, Hi, Wi, CI]
+    const T16* __restrict__ dy,   // [M, KO]
+    float* __restrict__ dw,       // chunk slabs of [KO, 3*3*CI]
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    const int Ho, const int Wo, const long m_per_chunk, const int nchunks) {
+  __shared__ T16 lds[(KT + 3 * 64) * LDM];
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const int k0 = blockIdx.x * KT;
+  const int cchunks = CI / BK;
+  const int c0 = (blockIdx.y % cchunks) * BK;
+  const int r_ = blockIdx.y / cchunks;
+  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+
+  const bool do_dy = tid < 2 * KT;
+  const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
+  const int t = do_dy ? tid : 0;
+  const int sm = (t & 15) * 4;
+  const int sk = (t >> 4) * 8;
+  const int tx = tid & 127;
+  const int smx = (tx & 15) * 4;
+  const int skx = (tx >> 4) * 8;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  constexpr int NJ = KT / 64;
+  const int i0 = KT == 64 ? (wave & 1) * 32 : wave * 32;
+  const int j0base = KT == 64 ? (wave >> 1) * 32 : 0;
+
+  f32x16 acc[3][NJ] = {};
+
+  // incremental decode for the LEADING gather position m = m0 + smx - 1
+  int dn = 0, dp = 0, dq = 0;
+  {
+    const long mf = m_begin + smx - 1;
+    const long m_first = mf < 0 ? 0 : mf;
+    dn = (int)(m_first / ((long)Ho * Wo));
+    const int pq = (int)(m_first % ((long)Ho * Wo));
+    dp = pq / Wo;
+    dq = pq % Wo;
+  }
+  auto advance = [&](int by) {
+    dq += by;
+    while (dq >= Wo) {
+      dq -= Wo;
+      if (++dp == Ho) {
+        dp = 0;
+        ++dn;
+      }
+    }
+  };
+
+  short8 vdy[4], vx[6];
+  int qg[6];  // q of each gathered position (for the row-boundary masks)
+  auto load_m = [&](long m0) {
+    if (do_dy) {
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const long m = m0 + sm + mi;
+        vdy[mi] = (m < m_end) ? *reinterpret_cast<const short8*>(
+                                    dy + m * KO + k0 + sk)
+                              : short8{};
+      }
+    }
+    if (do_x) {
+      int n_ = dn, p_ = dp, q_ = dq;
+#pragma unroll
+      for (int i = 0; i < 6; ++i) {
+        const long m = m0 + smx - 1 + i;
+        const int ih = p_ + r_ - 1;  // stride 1, pad 1
+        const int iw = q_;           // center tap (s=1)
+        qg[i] = q_;
+        const bool ok = m >= 0 && m < Mtot && (unsigned)ih < (unsigned)Hi;
+        vx[i] = ok ? *reinterpret_cast<const short8*>(
+                         x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + skx)
+                   : short8{};
+        if (i < 5 && m >= 0 && ++q_ == Wo) {
+          q_ = 0;
+          if (++p_ == Ho) {
+            p_ = 0;
+            ++n_;
+          }
+        }
+      }
+      advance(WGM);
+    }
+  };
+  auto stage_m = [&]() {
+    if (do_dy) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+      }
+    }
+    if (do_x) {
+      // copy s slot j <- center gather index j + s; boundary slots zeroed
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        T16* ldsS = lds + (KT + s * 64) * LDM;
+        short keep[4];
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) {
+          const int qt = qg[jj + 1];  // q of target output m
+          const bool dead = (s == 0 && qt == 0) || (s == 2 && qt == Wo - 1);
+          keep[jj] = dead ? (short)0 : (short)-1;
+        }
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          short4v pk = {(short)(vx[s + 0][e] & keep[0]),
+                        (short)(vx[s + 1][e] & keep[1]),
+                        (short)(vx[s + 2][e] & keep[2]),
+                        (short)(vx[s + 3][e] & keep[3])};
+          *reinterpret_cast<short4v*>(
+              reinterpret_cast<short*>(ldsS + (skx + e) * LDM + smx)) = pk;
+        }
+      }
+    }
+  };
+
+  load_m(m_begin);
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
+    __syncthreads();
+    stage_m();
+    __syncthreads();
+    if (m0 + WGM < m_end) load_m(m0 + WGM);
+    const T16* ldsDyT = lds;
+#pragma unroll
+    for (int kk = 0; kk < WGM; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsDyT + (i0 + li) * LDM + kk + kh * 8);
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        const T16* ldsXT = lds + (KT + s * 64) * LDM;
+#pragma unroll
+        for (int jj = 0; jj < NJ; ++jj) {
+          const short8 bf = *reinterpret_cast<const short8*>(
+              ldsXT + (j0base + jj * 32 + li) * LDM + kk + kh * 8);
+          acc[s][jj] = Mfma32<T16>::run(af, bf, acc[s][jj]);
+        }
+      }
+    }
+  }
+
+  float* slab = dw + (long)blockIdx.z * ((long)KO * CI * 9);
+#pragma unroll
+  for (int s = 0; s < 3; ++s) {
+    const int rs = r_ * 3 + s;
+#pragma unroll
+    for (int jj = 0; jj < NJ; ++jj) {
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+        const int c_abs = c0 + j0base + jj * 32 + li;
+        slab[((long)(k0 + i0 + i) * CI + c_abs) * 9 + rs] = acc[s][jj][reg];
+      }
+    }
+  }
+}
+
 // out[zc][e] = sum over this block-row's z-range of part[z][e] — the
 // z-parallel first level of the two-level reduce (small E can't fill the
 // chip with element-parallelism alone)
@@ -887,6 +1063,48 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                          cur_stream(), (const T16*)x.data_ptr(),
                          (const T16*)dy.data_ptr(), part.data_ptr<float>(),
                          N, Hi, Wi, CI, KO, wshift, m_per_chunk);
+    });
+    if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
+    return;
+  }
+  // 3x3/s1/p1: s-grouped kernel (3 taps per stage) — MI355X_WGRAD_S3=0
+  // falls back to the tap-per-block kernel
+  static const bool s3_on = [] {
+    const char* e = getenv("MI355X_WGRAD_S3");
+    return !e || e[0] != '0';
+  }();
+  if (s3_on && R == 3 && S == 3 && stride == 1 && pad == 1 && Hi == Ho &&
+      Wi == Wo) {
+    static const long cap3 = [] {
+      const char* e = getenv("MI355X_WGRAD_CHUNKS");
+      return e ? atol(e) : 0L;
+    }();
+    int nchunks = (int)std::max<long>(std::min<long>(cdiv_l(M, 4096), 4096),
+                                      1);
+    if (cap3 > 0) nchunks = (int)std::min<long>(nchunks, cap3);
+    long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
+    nchunks = (int)cdiv_l(M, m_per_chunk);
+    static const long ktf = [] {  // A/B: force KT=64 (occ 3 vs 2)
+      const char* e = getenv("MI355X_WGRAD_S3_KT");
+      return e ? atol(e) : 0L;
+    }();
+    const int KT = ktf == 64 ? 64 : (KO % 128 == 0 ? 128 : 64);
+    dim3 grid(KO / KT, 3 * (CI / 64), nchunks);
+    const long E = (long)KO * 9 * CI;
+    at::Tensor part = nchunks > 1
+                          ? at::empty({nchunks, E}, dw.options())
+                          : dw;
+    DISPATCH_16(x, T16, {
+      if (KT == 128)
+        hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 128>), grid, dim3(256),
+                           0, cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
+      else
+        hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64>), grid, dim3(256),
+                           0, cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
     });
     if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
     return;
