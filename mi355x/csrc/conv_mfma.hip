@@ -509,11 +509,12 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
 
   f32x16 acc[3][NJ] = {};
 
-  // incremental decode for the LEADING gather position m = m0 + smx - 1
+  // incremental decode for the CENTER gather position m = m0 + smx (the
+  // i=0 halo position is derived from it: clamping a -1 start desyncs the
+  // per-step advance for the smx==0 thread — measured as a wrong dw tile)
   int dn = 0, dp = 0, dq = 0;
   {
-    const long mf = m_begin + smx - 1;
-    const long m_first = mf < 0 ? 0 : mf;
+    const long m_first = m_begin + smx;
     dn = (int)(m_first / ((long)Ho * Wo));
     const int pq = (int)(m_first % ((long)Ho * Wo));
     dp = pq / Wo;
@@ -543,18 +544,31 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
       }
     }
     if (do_x) {
-      int n_ = dn, p_ = dp, q_ = dq;
-#pragma unroll
-      for (int i = 0; i < 6; ++i) {
-        const long m = m0 + smx - 1 + i;
+      int n_ = dn, p_ = dp, q_ = dq;  // position m0 + smx (center start)
+      {
+        // i = 0 halo: column q-1 of the SAME row; when q == 0 the only
+        // consumer slot (copy_0 at a q==0 target) is dead-masked, so a
+        // zero stands in
         const int ih = p_ + r_ - 1;  // stride 1, pad 1
-        const int iw = q_;           // center tap (s=1)
-        qg[i] = q_;
-        const bool ok = m >= 0 && m < Mtot && (unsigned)ih < (unsigned)Hi;
-        vx[i] = ok ? *reinterpret_cast<const short8*>(
-                         x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + skx)
+        qg[0] = q_ - 1;
+        const bool ok = m0 + smx - 1 >= 0 && q_ > 0 &&
+                        (unsigned)ih < (unsigned)Hi;
+        vx[0] = ok ? *reinterpret_cast<const short8*>(
+                         x + (((long)n_ * Hi + ih) * Wi + (q_ - 1)) * CI +
+                         c0 + skx)
                    : short8{};
-        if (i < 5 && m >= 0 && ++q_ == Wo) {
+      }
+#pragma unroll
+      for (int i = 1; i < 6; ++i) {
+        const long m = m0 + smx - 1 + i;
+        const int ih = p_ + r_ - 1;
+        qg[i] = q_;
+        const bool ok = m < Mtot && (unsigned)ih < (unsigned)Hi;
+        vx[i] = ok ? *reinterpret_cast<const short8*>(
+                         x + (((long)n_ * Hi + ih) * Wi + q_) * CI + c0 +
+                         skx)
+                   : short8{};
+        if (i < 5 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
             p_ = 0;
