@@ -40,6 +40,7 @@ CASES = [
     ((2, 32, 32, 16), 32, 3, 1, 1),
     ((2, 16, 16, 32), 64, 3, 2, 1),
     ((2, 8, 8, 64), 64, 3, 1, 1),
+    ((2, 64, 64, 64), 64, 3, 1, 1),
     ((2, 8, 8, 128), 128, 1, 1, 0),
     ((3, 7, 7, 64), 192, 3, 2, 1),
     ((4, 16, 16, 64), 128, 3, 2, 1),

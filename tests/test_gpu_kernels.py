@@ -41,6 +41,7 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 16, 16, 32), 64, 1, 2, 0, False, None),  # 1x1 shortcut
     ((2, 33, 33, 8), 16, 7, 2, 3, False, None),   # 7x7/2 stem, odd size
     ((2, 8, 8, 64), 64, 3, 1, 1, False, "relu"),  # MFMA path shape
+    ((2, 64, 64, 64), 64, 3, 1, 1, False, None),   # multi-m-chunk wgrad+reduce
     ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
     ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
