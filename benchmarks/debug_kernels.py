@@ -225,6 +225,9 @@ def pmcprobe():
     # layer1 wgrad MFMA (r18 shape)
     xw = torch.randn(256, 32, 32, 64, device="cuda").to(torch.bfloat16)
     dyw = torch.randn(256, 32, 32, 64, device="cuda").to(torch.bfloat16)
+    # conv gather-GEMM shapes (r18 layer1 fwd + dgrad at b1024)
+    xg = torch.randn(1024, 32, 32, 64, device="cuda").to(torch.bfloat16)
+    wg = (torch.randn(64, 3, 3, 64, device="cuda") * 0.1).to(torch.bfloat16)
     yp, ip = O.ext().maxpool_fwd(xp, 3, 2, 1)
     st = O.ext().bn_stats(xb)
     mean = st[0] / xb.numel() * 256
@@ -236,6 +239,7 @@ def pmcprobe():
         O.ext().maxpool_bwd(yp.clone(), ip, 112, 112, 3, 2, 1)
         O.ext().bn_apply(xb, mean, inv, g1, torch.zeros_like(g1), e, 1)
         O.ext().conv2d_wgrad(xw, dyw, 3, 3, 1, 1)
+        O.ext().conv2d_fwd(xg, wg, e, 1, 1, 0, 3, 3)
     torch.cuda.synchronize()
     print("pmcprobe done")
 
