@@ -93,7 +93,17 @@ class DistributedDataParallel(nn.Module):
         use_native = (comm == "rccl"
                       and self.flat.flat_param.is_cuda
                       and self.world_size > 1)
-        self.comm = _NativeComm() if use_native else _TorchComm(process_group)
+        if use_native:
+            try:
+                self.comm = _NativeComm()
+            except Exception as e:  # construction-time failure only; fall
+                # back to torch.distributed (also RCCL underneath on ROCm)
+                import warnings
+                warnings.warn(f"native RCCL comm init failed ({e}); "
+                              "falling back to torch.distributed")
+                self.comm = _TorchComm(process_group)
+        else:
+            self.comm = _TorchComm(process_group)
         self._works: list = []
         self._ready: dict[int, int] = {}
         self._launched = 0
