@@ -281,6 +281,141 @@ __global__ void conv_stats_reduce(const float* __restrict__ slab,
 }  // namespace
 
 // ---------------------------------------------------------------------------
+// ---------------------------------------------------------------------------
+// Patch gather-GEMM for 3x3/stride-1/pad-1 (fwd and dgrad): the geometric
+// input patch covering a 64-output tile (its q-rows +- 1 halo row, with
+// zeroed pad columns) is staged into LDS ONCE per c-chunk; all NINE taps
+// read it by address offset — the per-k-step global re-gather of the
+// plain kernel (9x input traffic, 18 barriers/c-chunk) becomes one
+// cooperative stage + 2 barriers. The B (weight) fragments live in
+// REGISTERS, prefetched one tap ahead straight from L2 — no B tile in
+// LDS, no B barriers. Vertical pad / image-crossing / M-tail are handled
+// by a per-(lane, tap-row) address select onto a zeroed LDS stub (never a
+// branch around the read).
+//   block: 64m x 64k; 4 waves as 2m x 2k, one 32x32 acc tile each.
+//   patch: [NR rows][Wi+2 cols][72] (stride 72: 16B-aligned b128 reads,
+//   36-dword lane stride -> worst 2-way bank conflicts).
+// ---------------------------------------------------------------------------
+constexpr int PCS = 72;  // patch col stride (elements)
+
+template <typename T16, bool TRANS>
+__global__ __launch_bounds__(256, 2) void conv_patch_gemm(
+    const T16* __restrict__ in,   // [N, Hi, Wi, CI] (dgrad: dy, CI=KO)
+    const T16* __restrict__ wgt,  // fwd: [KO, 9*CI]; dgrad: wflip strides
+    const float* __restrict__ bias, T16* __restrict__ out,
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    const long b_row_stride, const long b_rs_stride, const int act,
+    const int has_bias, const int NR) {
+  extern __shared__ __attribute__((aligned(16))) char psmem[];
+  T16* patch = reinterpret_cast<T16*>(psmem);
+  T16* zstub = patch + (long)NR * (Wi + 2) * PCS;  // 72 zero elements
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Hi * Wi;  // Ho==Hi, Wo==Wi (s1p1)
+  const int Wo = Wi, Ho = Hi;
+  const long bm0 = (long)blockIdx.x * 64;
+  const int k0 = blockIdx.y * 64;
+  const long gr0 = bm0 / Wo - 1;  // first staged global row (n*Ho + p)
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int wm = (wave & 1) * 32;   // wave's m-half
+  const int wk = (wave >> 1) * 32;  // wave's k-half
+
+  // per-lane A-row coordinates (fixed across the whole kernel)
+  const long m_lane = bm0 + wm + li;
+  const long grl = m_lane / Wo;
+  const int p_lane = (int)(grl % Ho);
+  const int prow = (int)(grl - gr0);
+  const int pcol = (int)(m_lane - grl * Wo) + 1;
+  const bool m_ok = m_lane < Mtot;
+  const long lane_base = ((long)prow * (Wi + 2) + pcol) * PCS;
+
+  if (tid < 64) zstub[tid] = T16{};  // first barrier publishes it
+
+  const int cchunks = CI / BK;
+  f32x16 acc = {};
+
+  // B fragments in registers: lane (li,kh) holds w[k0+wk+li] elements
+  // (tap, cc*64 + kk + kh*8); prefetched one tap ahead
+  const T16* wrow = wgt + (long)(k0 + wk + li) * b_row_stride;
+  auto load_b = [&](int tap, int cc, short8 (&bf)[4]) {
+    const T16* wp = wrow + (long)tap * b_rs_stride + cc * BK + kh * 8;
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk)
+      bf[kk] = *reinterpret_cast<const short8*>(wp + kk * 16);
+  };
+
+  short8 bf0[4], bf1[4];
+  load_b(0, 0, bf0);
+
+  for (int cc = 0; cc < cchunks; ++cc) {
+    // tap-0 B fragments for this c-chunk: issued before the staging
+    // barriers, so the load is long done when tap 0's MFMAs need it
+    if (cc > 0) load_b(0, cc, bf0);
+    __syncthreads();  // previous c-chunk's patch reads done
+    // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi ----
+    const int ngroups = NR * (Wi + 2) * (BK / 8);
+    for (int i = tid; i < ngroups; i += 256) {
+      const int g = i % (BK / 8);
+      const int ce = i / (BK / 8);
+      const int col = ce % (Wi + 2);
+      const int row = ce / (Wi + 2);
+      const long gr = gr0 + row;
+      const int iw = col - 1;
+      short8 v = {};
+      if (gr >= 0 && gr < (long)N * Ho && (unsigned)iw < (unsigned)Wi) {
+        const long nn = gr / Ho, pp = gr % Ho;
+        v = *reinterpret_cast<const short8*>(
+            in + ((nn * Hi + pp) * Wi + iw) * CI + cc * BK + g * 8);
+      }
+      *reinterpret_cast<short8*>(patch + ((long)row * (Wi + 2) + col) * PCS +
+                                 g * 8) = v;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      // tap row validity is per-lane but constant over s and kk
+      const int pv = TRANS ? p_lane + 1 - r : p_lane + r - 1;
+      const bool rv = m_ok && (unsigned)pv < (unsigned)Ho;
+#pragma unroll
+      for (int s = 0; s < 3; ++s) {
+        const int tap = r * 3 + s;
+        const long off = TRANS
+                             ? ((long)(1 - r) * (Wi + 2) + (1 - s)) * PCS
+                             : ((long)(r - 1) * (Wi + 2) + (s - 1)) * PCS;
+        const T16* abase = rv ? patch + lane_base + off : zstub;
+        // prefetch next tap's B fragments under this tap's MFMAs
+        short8 (&bfc)[4] = (tap & 1) ? bf1 : bf0;
+        short8 (&bfn)[4] = (tap & 1) ? bf0 : bf1;
+        if (tap < 8) load_b(tap + 1, cc, bfn);
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+          const short8 af = *reinterpret_cast<const short8*>(
+              abase + kk * 16 + kh * 8);
+          acc = Mfma32<T16>::run(af, bfc[kk], acc);
+        }
+      }
+    }
+  }
+
+  // ---- epilogue: bias + act + store ----
+  const float bv = has_bias ? bias[k0 + wk + li] : 0.f;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mtot) {
+      float v = acc[reg] + bv;
+      if (act == 1) v = fmaxf(v, 0.f);
+      out[m_out * KO + k0 + wk + li] = F16<T16>::from_f32(v);
+    }
+  }
+}
+
 // wgrad: dw[KO, kg] = sum_m dy[m, KO] * A_im2col[m, kg]. MFMA reduces over
 // its register-minor k dim, which here is m — both operands are m-major in
 // memory, so both tiles are staged TRANSPOSED into LDS ([k][m] / [kg][m]),
@@ -1228,6 +1363,28 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     });
     return;
   }
+  // 3x3/s1/p1 patch kernel: one geometric LDS patch per c-chunk serves
+  // all 9 taps (MI355X_CONV_PATCH=0 falls back to the per-tap gather)
+  static const bool patch_on = [] {
+    const char* e = getenv("MI355X_CONV_PATCH");
+    return !e || e[0] != '0';
+  }();
+  if (patch_on && !(stats.defined() && stats.numel() > 0) && R == 3 &&
+      S == 3 && stride == 1 && pad == 1 && Ho == Hi && Wo == Wi &&
+      Wo <= 64) {
+    const int NR = 64 / Wo + 3;
+    const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS) * 2;
+    dim3 pgrid_((unsigned)cdiv_l(M, 64), KO / 64);
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
+                         smem, cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)w.data_ptr(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO,
+                         (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
+    });
+    return;
+  }
   // BN=128 halves barriers per MFMA but also halves the grid — only use
   // it when M is large enough to keep the chip full at BM=128 tiles
   const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
@@ -1279,6 +1436,24 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
+  static const bool patch_on = [] {
+    const char* e = getenv("MI355X_CONV_PATCH");
+    return !e || e[0] != '0';
+  }();
+  if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && P == H &&
+      Q == W && W <= 64) {
+    const int NR = 64 / W + 3;
+    const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS) * 2;
+    dim3 pgrid_((unsigned)cdiv_l(M, 64), CI / 64);
+    DISPATCH_16(dy, T16, {
+      hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
+                         smem, cur_stream(), (const T16*)dy.data_ptr(),
+                         (const T16*)wflip.data_ptr(), nullptr,
+                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, (long)KO,
+                         (long)CI * KO, 0, 0, NR);
+    });
+    return;
+  }
   const bool wide = CI % 128 == 0 && cdiv_l(M, BM) * (CI / 128) >= 1024;
   dim3 grid((unsigned)cdiv_l(M, BM), CI / (wide ? 128 : 64));
   at::Tensor zp = conv_zero_page(dy);
