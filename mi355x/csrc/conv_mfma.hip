@@ -292,7 +292,10 @@ __global__ void conv_stats_reduce(const float* __restrict__ slab,
 // LDS, no B barriers. Vertical pad / image-crossing / M-tail are handled
 // by a per-(lane, tap-row) address select onto a zeroed LDS stub (never a
 // branch around the read).
-//   block: 64m x 64k; 4 waves as 2m x 2k, one 32x32 acc tile each.
+//   block: 128m x 64k; 4 waves each own 32 m-rows x the full 64 k as TWO
+//   32x32 acc tiles — one tile per wave makes every MFMA a serial
+//   16-cycle dependency chain (measured 260-290 TF); two interleave to
+//   the 8-cycle cadence.
 //   patch: [NR rows][Wi+2 cols][72] (stride 72: 16B-aligned b128 reads,
 //   36-dword lane stride -> worst 2-way bank conflicts).
 // ---------------------------------------------------------------------------
@@ -313,7 +316,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Hi * Wi;  // Ho==Hi, Wo==Wi (s1p1)
   const int Wo = Wi, Ho = Hi;
-  const long bm0 = (long)blockIdx.x * 64;
+  const long bm0 = (long)blockIdx.x * 128;
   const int k0 = blockIdx.y * 64;
   const long gr0 = bm0 / Wo - 1;  // first staged global row (n*Ho + p)
 
@@ -321,8 +324,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   const int lane = tid & 63;
   const int li = lane & 31;
   const int kh = lane >> 5;
-  const int wm = (wave & 1) * 32;   // wave's m-half
-  const int wk = (wave >> 1) * 32;  // wave's k-half
+  const int wm = wave * 32;  // wave's m-rows (full 64 k per wave)
 
   // per-lane A-row coordinates (fixed across the whole kernel)
   const long m_lane = bm0 + wm + li;
@@ -336,19 +338,22 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   if (tid < 64) zstub[tid] = T16{};  // first barrier publishes it
 
   const int cchunks = CI / BK;
-  f32x16 acc = {};
+  f32x16 acc[2] = {};
 
-  // B fragments in registers: lane (li,kh) holds w[k0+wk+li] elements
-  // (tap, cc*64 + kk + kh*8); prefetched one tap ahead
-  const T16* wrow = wgt + (long)(k0 + wk + li) * b_row_stride;
-  auto load_b = [&](int tap, int cc, short8 (&bf)[4]) {
-    const T16* wp = wrow + (long)tap * b_rs_stride + cc * BK + kh * 8;
+  // B fragments in registers: lane (li,kh) holds rows k0+li and k0+32+li,
+  // elements (tap, cc*64 + kk + kh*8); prefetched one tap ahead
+  const T16* wrow0 = wgt + (long)(k0 + li) * b_row_stride;
+  const T16* wrow1 = wgt + (long)(k0 + 32 + li) * b_row_stride;
+  auto load_b = [&](int tap, int cc, short8 (&bf)[2][4]) {
+    const long o = (long)tap * b_rs_stride + cc * BK + kh * 8;
 #pragma unroll
-    for (int kk = 0; kk < 4; ++kk)
-      bf[kk] = *reinterpret_cast<const short8*>(wp + kk * 16);
+    for (int kk = 0; kk < 4; ++kk) {
+      bf[0][kk] = *reinterpret_cast<const short8*>(wrow0 + o + kk * 16);
+      bf[1][kk] = *reinterpret_cast<const short8*>(wrow1 + o + kk * 16);
+    }
   };
 
-  short8 bf0[4], bf1[4];
+  short8 bf0[2][4], bf1[2][4];
   load_b(0, 0, bf0);
 
   for (int cc = 0; cc < cchunks; ++cc) {
@@ -389,29 +394,35 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
                              : ((long)(r - 1) * (Wi + 2) + (s - 1)) * PCS;
         const T16* abase = rv ? patch + lane_base + off : zstub;
         // prefetch next tap's B fragments under this tap's MFMAs
-        short8 (&bfc)[4] = (tap & 1) ? bf1 : bf0;
-        short8 (&bfn)[4] = (tap & 1) ? bf0 : bf1;
+        short8 (&bfc)[2][4] = (tap & 1) ? bf1 : bf0;
+        short8 (&bfn)[2][4] = (tap & 1) ? bf0 : bf1;
         if (tap < 8) load_b(tap + 1, cc, bfn);
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
           const short8 af = *reinterpret_cast<const short8*>(
               abase + kk * 16 + kh * 8);
-          acc = Mfma32<T16>::run(af, bfc[kk], acc);
+          acc[0] = Mfma32<T16>::run(af, bfc[0][kk], acc[0]);
+          acc[1] = Mfma32<T16>::run(af, bfc[1][kk], acc[1]);
         }
       }
     }
   }
 
   // ---- epilogue: bias + act + store ----
-  const float bv = has_bias ? bias[k0 + wk + li] : 0.f;
+  float bv[2];
+  bv[0] = has_bias ? bias[k0 + li] : 0.f;
+  bv[1] = has_bias ? bias[k0 + 32 + li] : 0.f;
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
     const long m_out = bm0 + wm + row;
     if (m_out < Mtot) {
-      float v = acc[reg] + bv;
-      if (act == 1) v = fmaxf(v, 0.f);
-      out[m_out * KO + k0 + wk + li] = F16<T16>::from_f32(v);
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2) {
+        float v = acc[t2][reg] + bv[t2];
+        if (act == 1) v = fmaxf(v, 0.f);
+        out[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
+      }
     }
   }
 }
@@ -1372,9 +1383,9 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   if (patch_on && !(stats.defined() && stats.numel() > 0) && R == 3 &&
       S == 3 && stride == 1 && pad == 1 && Ho == Hi && Wo == Wi &&
       Wo <= 64) {
-    const int NR = 64 / Wo + 3;
+    const int NR = 128 / Wo + 3;
     const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS) * 2;
-    dim3 pgrid_((unsigned)cdiv_l(M, 64), KO / 64);
+    dim3 pgrid_((unsigned)cdiv_l(M, 128), KO / 64);
     DISPATCH_16(x, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
                          smem, cur_stream(), (const T16*)x.data_ptr(),
@@ -1442,9 +1453,9 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   }();
   if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && P == H &&
       Q == W && W <= 64) {
-    const int NR = 64 / W + 3;
+    const int NR = 128 / W + 3;
     const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS) * 2;
-    dim3 pgrid_((unsigned)cdiv_l(M, 64), CI / 64);
+    dim3 pgrid_((unsigned)cdiv_l(M, 128), CI / 64);
     DISPATCH_16(dy, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
                          smem, cur_stream(), (const T16*)dy.data_ptr(),
