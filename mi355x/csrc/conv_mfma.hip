@@ -312,6 +312,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   extern __shared__ __attribute__((aligned(16))) char psmem[];
   T16* patch = reinterpret_cast<T16*>(psmem);
   T16* zstub = patch + (long)NR * (Wi + 2) * PCS;  // 72 zero elements
+  T16* ldsB = zstub + PCS;  // [64][LDK] weight tile for the current tap
 
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Hi * Wi;  // Ho==Hi, Wo==Wi (s1p1)
@@ -340,26 +341,28 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   const int cchunks = CI / BK;
   f32x16 acc[2] = {};
 
-  // B fragments in registers: lane (li,kh) holds rows k0+li and k0+32+li,
-  // elements (tap, cc*64 + kk + kh*8); prefetched one tap ahead
-  const T16* wrow0 = wgt + (long)(k0 + li) * b_row_stride;
-  const T16* wrow1 = wgt + (long)(k0 + 32 + li) * b_row_stride;
-  auto load_b = [&](int tap, int cc, short8 (&bf)[2][4]) {
-    const long o = (long)tap * b_rs_stride + cc * BK + kh * 8;
-#pragma unroll
-    for (int kk = 0; kk < 4; ++kk) {
-      bf[0][kk] = *reinterpret_cast<const short8*>(wrow0 + o + kk * 16);
-      bf[1][kk] = *reinterpret_cast<const short8*>(wrow1 + o + kk * 16);
-    }
+  // B staged through LDS, cooperatively and COALESCED (per-lane scattered
+  // weight-row loads measured 86.6% WAIT_ANY — a wave's worth of strided
+  // 64B reads with 1-tap prefetch never hides L2 latency; the block-wide
+  // burst + barrier pipeline does). 4 threads per B row, 16 elements each.
+  const int sb_n = tid >> 2;
+  const int sb_c = (tid & 3) * 16;
+  const T16* wrowB = wgt + (long)(k0 + sb_n) * b_row_stride + sb_c;
+  short8 breg[2];
+  auto load_b = [&](int tap, int cc) {
+    const T16* wp = wrowB + (long)tap * b_rs_stride + cc * BK;
+    breg[0] = *reinterpret_cast<const short8*>(wp);
+    breg[1] = *reinterpret_cast<const short8*>(wp + 8);
+  };
+  auto stage_b = [&]() {
+    short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
+    *reinterpret_cast<short8*>(pb) = breg[0];
+    *reinterpret_cast<short8*>(pb + 8) = breg[1];
   };
 
-  short8 bf0[2][4], bf1[2][4];
-  load_b(0, 0, bf0);
+  load_b(0, 0);
 
   for (int cc = 0; cc < cchunks; ++cc) {
-    // tap-0 B fragments for this c-chunk: issued before the staging
-    // barriers, so the load is long done when tap 0's MFMAs need it
-    if (cc > 0) load_b(0, cc, bf0);
     __syncthreads();  // previous c-chunk's patch reads done
     // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi ----
     const int ngroups = NR * (Wi + 2) * (BK / 8);
@@ -393,16 +396,24 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
                              ? ((long)(1 - r) * (Wi + 2) + (1 - s)) * PCS
                              : ((long)(r - 1) * (Wi + 2) + (s - 1)) * PCS;
         const T16* abase = rv ? patch + lane_base + off : zstub;
-        // prefetch next tap's B fragments under this tap's MFMAs
-        short8 (&bfc)[2][4] = (tap & 1) ? bf1 : bf0;
-        short8 (&bfn)[2][4] = (tap & 1) ? bf0 : bf1;
-        if (tap < 8) load_b(tap + 1, cc, bfn);
+        __syncthreads();  // previous tap's MFMAs done reading ldsB
+        stage_b();
+        __syncthreads();
+        // next tap's (or next c-chunk's) B burst rides under the MFMAs
+        if (tap < 8)
+          load_b(tap + 1, cc);
+        else if (cc + 1 < cchunks)
+          load_b(0, cc + 1);
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
           const short8 af = *reinterpret_cast<const short8*>(
               abase + kk * 16 + kh * 8);
-          acc[0] = Mfma32<T16>::run(af, bfc[0][kk], acc[0]);
-          acc[1] = Mfma32<T16>::run(af, bfc[1][kk], acc[1]);
+          const short8 b0 = *reinterpret_cast<const short8*>(
+              ldsB + li * LDK + kk * 16 + kh * 8);
+          const short8 b1 = *reinterpret_cast<const short8*>(
+              ldsB + (32 + li) * LDK + kk * 16 + kh * 8);
+          acc[0] = Mfma32<T16>::run(af, b0, acc[0]);
+          acc[1] = Mfma32<T16>::run(af, b1, acc[1]);
         }
       }
     }
@@ -1384,7 +1395,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
       S == 3 && stride == 1 && pad == 1 && Ho == Hi && Wo == Wi &&
       Wo <= 64) {
     const int NR = 128 / Wo + 3;
-    const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS) * 2;
+    const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), KO / 64);
     DISPATCH_16(x, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
@@ -1454,7 +1465,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && P == H &&
       Q == W && W <= 64) {
     const int NR = 128 / W + 3;
-    const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS) * 2;
+    const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), CI / 64);
     DISPATCH_16(dy, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
