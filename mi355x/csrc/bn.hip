@@ -144,7 +144,10 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
   }
   const long e0 = (long)blockIdx.x * e_per_block;
   const long e1 = min(E, e0 + e_per_block);
-  for (long e = e0 + (long)tid * 8; e < e1; e += 256 * 8) {
+  constexpr long ESTEP = 256 * 8;
+  // body for one 16B group (channel group is e-invariant: strides are
+  // multiples of 2048 and 2048 % C == 0)
+  auto body = [&](long e) {
     if (!BWD) {
       const short8 vx = *reinterpret_cast<const short8*>(x + e);
 #pragma unroll
@@ -167,7 +170,16 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
         s1[u] += d;
       }
     }
+  };
+  // 4 sub-streams per iteration: one outstanding 16B load per thread caps
+  // at ~2.4 TB/s by Little's law (512 blocks x 256 thr x 16 B in flight);
+  // four batched loads quadruple the in-flight bytes
+  long e = e0 + (long)tid * 8;
+  for (; e + 3 * ESTEP < e1; e += 4 * ESTEP) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) body(e + v * ESTEP);
   }
+  for (; e < e1; e += ESTEP) body(e);
   // wave-level pre-reduction: lanes whose 8-channel group repeats within
   // the wave (group stride C/8 lanes) fold via shfl_xor before touching
   // LDS — cuts LDS-atomic collisions by 512/C
