@@ -620,6 +620,144 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
   }
 }
 
+// 128x128-output-tile wgrad for KO%128 && CI%128 (the ResNet-50 1x1
+// bottleneck convs carry ~45% of its FLOPs and can't use the s-grouped
+// kernel): each wave owns FOUR 32x32 acc tiles (2 KO x 2 CI), so one
+// dy+x stage feeds 16 MFMAs per wave per m-step instead of 8, and each
+// fragment read is reused twice. Same transposed-staging/split-M/slab
+// machinery as conv_wgrad_mfma_kernel.
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    float* __restrict__ dw,  // chunk slabs of [KO, R*S*CI]
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
+    const int Ho, const int Wo, const int R, const int S, const int stride,
+    const int pad, const long m_per_chunk, const int nchunks) {
+  __shared__ T16 lds[(128 + 128) * LDM];
+
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const int k0 = blockIdx.x * 128;
+  const int cblocks = CI / 128;
+  const int c0 = (blockIdx.y % cblocks) * 128;
+  const int s_ = (blockIdx.y / cblocks) % S;
+  const int r_ = blockIdx.y / (cblocks * S);
+  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+
+  // staging: all 256 threads stage dy (4m x 8k over 128 k-rows) and x
+  // (4m x 8c over 128 c-rows) with the same (m-group, row-group) map
+  const int sm = (tid & 15) * 4;
+  const int sk = (tid >> 4) * 8;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int i0 = (wave & 1) * 32;   // KO sub-offset
+  const int j0 = (wave >> 1) * 32;  // CI sub-offset
+
+  f32x16 acc[2][2] = {};
+
+  int dn = 0, dp = 0, dq = 0;
+  {
+    const long m_first = m_begin + sm;
+    dn = (int)(m_first / ((long)Ho * Wo));
+    const int pq = (int)(m_first % ((long)Ho * Wo));
+    dp = pq / Wo;
+    dq = pq % Wo;
+  }
+  auto advance = [&](int by) {
+    dq += by;
+    while (dq >= Wo) {
+      dq -= Wo;
+      if (++dp == Ho) {
+        dp = 0;
+        ++dn;
+      }
+    }
+  };
+
+  short8 vdy[4], vx0[4];
+  auto load_m = [&](long m0) {
+    int n_ = dn, p_ = dp, q_ = dq;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      const long m = m0 + sm + mi;
+      vdy[mi] = m < m_end ? *reinterpret_cast<const short8*>(
+                                dy + m * KO + k0 + sk)
+                          : short8{};
+      const int ih = p_ * stride - pad + r_;
+      const int iw = q_ * stride - pad + s_;
+      const bool ok = m < m_end && (unsigned)ih < (unsigned)Hi &&
+                      (unsigned)iw < (unsigned)Wi;
+      const T16* xp = x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
+      vx0[mi] = ok ? *reinterpret_cast<const short8*>(xp) : short8{};
+      if (mi < 3 && ++q_ == Wo) {
+        q_ = 0;
+        if (++p_ == Ho) {
+          p_ = 0;
+          ++n_;
+        }
+      }
+    }
+    advance(WGM);
+  };
+  auto stage_m = [&]() {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+      *reinterpret_cast<short4v*>(
+          reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+      short4v px0 = {vx0[0][e], vx0[1][e], vx0[2][e], vx0[3][e]};
+      *reinterpret_cast<short4v*>(reinterpret_cast<short*>(
+          lds + (128 + sk + e) * LDM + sm)) = px0;
+    }
+  };
+
+  load_m(m_begin);
+  for (long m0 = m_begin; m0 < m_end; m0 += WGM) {
+    __syncthreads();
+    stage_m();
+    __syncthreads();
+    if (m0 + WGM < m_end) load_m(m0 + WGM);
+    const T16* ldsDyT = lds;
+    const T16* ldsXT = lds + 128 * LDM;
+#pragma unroll
+    for (int kk = 0; kk < WGM; kk += 16) {
+      short8 af[2], bf[2];
+#pragma unroll
+      for (int a = 0; a < 2; ++a)
+        af[a] = *reinterpret_cast<const short8*>(
+            ldsDyT + (i0 + a * 64 + li) * LDM + kk + kh * 8);
+#pragma unroll
+      for (int b = 0; b < 2; ++b)
+        bf[b] = *reinterpret_cast<const short8*>(
+            ldsXT + (j0 + b * 64 + li) * LDM + kk + kh * 8);
+#pragma unroll
+      for (int a = 0; a < 2; ++a)
+#pragma unroll
+        for (int b = 0; b < 2; ++b)
+          acc[a][b] = Mfma32<T16>::run(af[a], bf[b], acc[a][b]);
+    }
+  }
+
+  const int rs = r_ * S + s_;
+  const long RS = (long)R * S;
+  float* slab = dw + (long)blockIdx.z * ((long)KO * CI * RS);
+#pragma unroll
+  for (int a = 0; a < 2; ++a)
+#pragma unroll
+    for (int b = 0; b < 2; ++b)
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+        const int c_abs = c0 + j0 + b * 64 + li;
+        slab[((long)(k0 + i0 + a * 64 + i) * CI + c_abs) * RS + rs] =
+            acc[a][b][reg];
+      }
+}
+
 // s-grouped wgrad for the dominant 3x3/stride-1/pad-1 convs: ONE block
 // computes all 3 s-taps of one filter row r from ONE dy+x stage. The
 // three x operand images are m-SHIFTED copies of the same center gather
@@ -1276,6 +1414,33 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                            0, cur_stream(), (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(), part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
+    });
+    if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
+    return;
+  }
+  // KO%128 && CI%128 (notably every r50 1x1 bottleneck conv): 128x128
+  // output tile, 4 acc tiles per wave — 2x the MFMA per staged byte of
+  // the 64-wide-CI kernel (which measured ~283 TF on these shapes)
+  static const bool t128_on = [] {
+    const char* e = getenv("MI355X_WGRAD_T128");
+    return !e || e[0] != '0';
+  }();
+  if (t128_on && KO % 128 == 0 && CI % 128 == 0) {
+    int nchunks = (int)std::max<long>(std::min<long>(cdiv_l(M, 4096), 4096),
+                                      1);
+    long m_per_chunk = cdiv_l(cdiv_l(M, nchunks), WGM) * WGM;
+    nchunks = (int)cdiv_l(M, m_per_chunk);
+    dim3 grid(KO / 128, (unsigned)(R * S * (CI / 128)), nchunks);
+    const long E = (long)KO * R * S * CI;
+    at::Tensor part = nchunks > 1
+                          ? at::empty({nchunks, E}, dw.options())
+                          : dw;
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16>), grid, dim3(256), 0,
+                         cur_stream(), (const T16*)x.data_ptr(),
+                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                         (int)stride, (int)pad, m_per_chunk, nchunks);
     });
     if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
     return;
