@@ -42,6 +42,9 @@ CASES = [
     ((2, 8, 8, 64), 64, 3, 1, 1),
     ((2, 64, 64, 64), 64, 3, 1, 1),
     ((2, 8, 8, 128), 128, 1, 1, 0),
+    ((2, 10, 10, 256), 128, 1, 1, 0),
+    ((2, 16, 16, 128), 256, 1, 2, 0),
+    ((2, 15, 15, 128), 128, 3, 2, 1),
     ((3, 7, 7, 64), 192, 3, 2, 1),
     ((4, 16, 16, 64), 128, 3, 2, 1),
     ((2, 9, 9, 256), 512, 1, 2, 0),

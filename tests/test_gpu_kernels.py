@@ -43,6 +43,9 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 8, 8, 64), 64, 3, 1, 1, False, "relu"),  # MFMA path shape
     ((2, 64, 64, 64), 64, 3, 1, 1, False, None),   # multi-m-chunk wgrad+reduce
     ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
+    ((2, 10, 10, 256), 128, 1, 1, 0, False, None), # t128 1x1, 2 c-blocks
+    ((2, 16, 16, 128), 256, 1, 2, 0, False, None), # t128 1x1 stride-2 (r50 ds)
+    ((2, 15, 15, 128), 128, 3, 2, 1, False, None), # t128 3x3/s2/p1 (r50 conv2)
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
     ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
     ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # stem7 7x7/2 (dot2 kernel)
