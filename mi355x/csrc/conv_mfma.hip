@@ -1559,7 +1559,10 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   if (patch_on && !(stats.defined() && stats.numel() > 0) && R == 3 &&
       S == 3 && stride == 1 && pad == 1 && Ho == Hi && Wo == Wi &&
       Wo <= 64) {
-    const int NR = 128 / Wo + 3;
+    // rows spanned by a 128-output tile starting mid-row, +2 halo rows:
+    // 128/Wo+3 was ONE short for Wo in {56,28,14,7} (tap r=2 then read
+    // past the patch -> diverging ResNet-50 training)
+    const int NR = (Wo + 126) / Wo + 3;
     const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), KO / 64);
     DISPATCH_16(x, T16, {
@@ -1629,7 +1632,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   }();
   if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && P == H &&
       Q == W && W <= 64) {
-    const int NR = 128 / W + 3;
+    const int NR = (W + 126) / W + 3;  // see fwd launcher comment
     const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), CI / 64);
     DISPATCH_16(dy, T16, {

@@ -42,6 +42,8 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 33, 33, 8), 16, 7, 2, 3, False, None),   # 7x7/2 stem, odd size
     ((2, 8, 8, 64), 64, 3, 1, 1, False, "relu"),  # MFMA path shape
     ((2, 64, 64, 64), 64, 3, 1, 1, False, None),   # multi-m-chunk wgrad+reduce
+    ((3, 7, 7, 64), 64, 3, 1, 1, False, None),     # patch: Wo=7 (odd row span)
+    ((2, 14, 14, 128), 128, 3, 1, 1, False, None), # patch: Wo=14
     ((2, 8, 8, 128), 128, 1, 1, 0, False, None),  # MFMA 1x1
     ((2, 10, 10, 256), 128, 1, 1, 0, False, None), # t128 1x1, 2 c-blocks
     ((2, 16, 16, 128), 256, 1, 2, 0, False, None), # t128 1x1 stride-2 (r50 ds)
