@@ -78,12 +78,35 @@ class BatchNorm2d(nn.Module):
         self.bias = nn.Parameter(torch.zeros(num_features))
         self.register_buffer("running_mean", torch.zeros(num_features))
         self.register_buffer("running_var", torch.ones(num_features))
-        self.register_buffer("num_batches_tracked", torch.tensor(0, dtype=torch.long))
+        # a host-side int, not a device buffer: the torch-style device
+        # tensor cost one captured GPU kernel per BN layer per step (20
+        # launches/step on ResNet-18) just to count. state_dict key parity
+        # is kept via the save/load hooks below.
+        self._nbt = 0
         self.process_group = None  # set by sync_bn.enable()
+
+    @property
+    def num_batches_tracked(self):
+        return torch.tensor(self._nbt, dtype=torch.long)
+
+    @num_batches_tracked.setter
+    def num_batches_tracked(self, v):
+        self._nbt = int(v)
+
+    def _save_to_state_dict(self, destination, prefix, keep_vars):
+        super()._save_to_state_dict(destination, prefix, keep_vars)
+        destination[prefix + "num_batches_tracked"] = torch.tensor(
+            self._nbt, dtype=torch.long)
+
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        key = prefix + "num_batches_tracked"
+        if key in state_dict:
+            self._nbt = int(state_dict.pop(key))
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
     def forward(self, x, residual=None):
         if self.training:
-            self.num_batches_tracked += 1
+            self._nbt += 1
         return ops.batch_norm(x, self.weight, self.bias, self.running_mean,
                               self.running_var, self.training, self.momentum,
                               residual, self.act, self.process_group if self.training else None)
@@ -107,7 +130,7 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
             and conv.act is None):
         from mi355x.ops import functional as F_
 
-        bn.num_batches_tracked += 1
+        bn._nbt += 1
 
         y, stats = F_.conv2d_with_stats(x, conv.weight, conv.stride,
                                         conv.padding)
