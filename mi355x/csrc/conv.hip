@@ -568,7 +568,7 @@ typedef float float2v __attribute__((ext_vector_type(2)));
 template <typename T16>
 __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
     const T16* __restrict__ x, const T16* __restrict__ dy,
-    float* __restrict__ part,  // [gridDim.x][K*7*7*3]
+    float* __restrict__ part,  // [gridDim.x*4][K*7*7*3]
     int N, int H, int W, int K, int Ho, int Wo, int pad, int rows_per_chunk) {
   constexpr int ROWS = 7, APAIR = 11;
   extern __shared__ char smem[];
@@ -583,13 +583,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
   const long row0 = (long)blockIdx.x * rows_per_chunk;
   const long row1 = min(nrows, row0 + rows_per_chunk);
 
-  // waves split FILTER ROWS, not the q-range: a full acc[7][11] per wave
-  // was 154 VGPRs -> occupancy 2 and 61% WAIT_ANY; two rows per wave is
-  // 44 and each wave walks every output of the row reading dy from LDS
-  // (cheap). Waves then write DISJOINT rows of ONE slab per chunk.
-  const int r_lo = wv * 2;                       // waves 0..2: 2 rows,
-  const int r_hi = min(ROWS, r_lo + 2);          // wave 3: row 6 only
-  float2v acc[2][APAIR] = {};
+  float2v acc[ROWS][APAIR] = {};
+
+  const int qper = (Wo + 3) / 4;
+  const int q0 = wv * qper, q1 = min(Wo, q0 + qper);
 
   for (long row = row0; row < row1; ++row) {
     const int p = (int)(row % Ho);
@@ -610,41 +607,36 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
     for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
     __syncthreads();
 
-    for (int q = 0; q < Wo; ++q) {
+    for (int q = q0; q < q1; ++q) {
       const float dyv = (k < K) ? F16<T16>::to_f32(dys[q * K + k]) : 0.f;
       float2v d2;
       d2.x = dyv;
       d2.y = dyv;
 #pragma unroll
-      for (int ri = 0; ri < 2; ++ri) {
-        const int r = r_lo + ri;
-        if (r >= ROWS) break;
+      for (int r = 0; r < ROWS; ++r) {
         const float2v* xrow = reinterpret_cast<const float2v*>(
             reinterpret_cast<const char*>(xs + r * sstride) + 24 * q);
 #pragma unroll
-        for (int u = 0; u < APAIR; ++u) acc[ri][u] += d2 * xrow[u];
+        for (int u = 0; u < APAIR; ++u) acc[r][u] += d2 * xrow[u];
       }
     }
   }
 
   if (k < K) {
     const long E = (long)K * 7 * 7 * 3;
-    float* slab = part + (long)blockIdx.x * E;
+    float* slab = part + ((long)blockIdx.x * 4 + wv) * E;
 #pragma unroll
-    for (int ri = 0; ri < 2; ++ri) {
-      const int r = r_lo + ri;
-      if (r >= ROWS) break;
+    for (int r = 0; r < ROWS; ++r)
 #pragma unroll
       for (int u = 0; u < APAIR; ++u) {
         const int e0 = 2 * u, e1 = 2 * u + 1;
         const int s0 = e0 / 3, c0 = e0 % 3;
-        slab[(((long)k * 3 + c0) * 7 + r) * 7 + s0] = acc[ri][u].x;
+        slab[(((long)k * 3 + c0) * 7 + r) * 7 + s0] = acc[r][u].x;
         if (e1 < 21) {
           const int s1 = e1 / 3, c1 = e1 % 3;
-          slab[(((long)k * 3 + c1) * 7 + r) * 7 + s1] = acc[ri][u].y;
+          slab[(((long)k * 3 + c1) * 7 + r) * 7 + s1] = acc[r][u].y;
         }
       }
-    }
   }
 }
 
@@ -1094,7 +1086,7 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
         long nc = std::min<long>(512, nrows);
         const int rows_per_chunk = (int)cdiv_l(nrows, nc);
         nc = cdiv_l(nrows, rows_per_chunk);
-        auto partl = at::empty({nc, E}, x.options().dtype(at::kFloat));
+        auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));
         const int selems = (2 * Q + 5) * 3;
         const int sstride = (selems + 4) & ~3;
         const size_t smem = (size_t)7 * sstride * 4 + (size_t)Q * K * 2;
@@ -1106,7 +1098,7 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                              partl.data_ptr<float>(), N, H, W, K, P, Q,
                              (int)pad, rows_per_chunk);
         });
-        wgrad_reduce_launch(partl, dw, E, nc);
+        wgrad_reduce_launch(partl, dw, E, nc * 4);
         return dw;
       }
       auto part = at::empty({nchunks * 4, E}, x.options().dtype(at::kFloat));
