@@ -544,6 +544,127 @@ __global__ __launch_bounds__(256) void conv_fwd_stem7_lds(
   }
 }
 
+// ---- 3x3/stride-1 CIFAR stem, LDS-staged (C=3, K<=64) ------------------
+// Same strategy as the 7x7 stem: the register-window kernels were
+// latency-bound on per-output wave-uniform loads (conv_fwd_smallc 224 us,
+// conv_wgrad_smallc 301 us per call at b1024). One block stages the
+// 3-row input strip for one output row as f32 (converted once), then
+// taps are plain 4B-aligned LDS reads: fwd lane-per-k with 27 register
+// weights; wgrad q-split waves with a 27-element f32 accumulator.
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_fwd_stem3_lds(
+    const T16* __restrict__ x, const T16* __restrict__ wgt,
+    const float* __restrict__ bias, T16* __restrict__ y, int N, int H,
+    int W, int K, int Ho, int Wo, long wrow_stride, int act, int has_bias) {
+  extern __shared__ __attribute__((aligned(16))) char smem3[];
+  float* xs = reinterpret_cast<float*>(smem3);
+  const int selems = (Wo + 2) * 3;
+  const int sstride = (selems + 4) & ~3;
+
+  const int p = blockIdx.x % Ho;
+  const int n = blockIdx.x / Ho;
+  for (int i = threadIdx.x; i < 3 * sstride; i += 256) {
+    const int rr = i / sstride;
+    const int e = i - rr * sstride;
+    const int cs = e / 3, c = e - cs * 3;
+    const int ih = p - 1 + rr, iw = cs - 1;
+    const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
+                    (unsigned)iw < (unsigned)W;
+    xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
+               : 0.f;
+  }
+
+  const int k = threadIdx.x & 63;
+  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+  float wr[27];
+  {
+    const T16* wk = wgt + (long)min(k, K - 1) * wrow_stride;
+#pragma unroll
+    for (int i = 0; i < 27; ++i) wr[i] = F16<T16>::to_f32(wk[i]);
+  }
+  const float bk = has_bias ? bias[min(k, K - 1)] : 0.f;
+  __syncthreads();
+
+  const int qper = (Wo + 3) / 4;
+  const int q0 = wv * qper, q1 = min(Wo, q0 + qper);
+  const long mrow = ((long)n * Ho + p) * Wo;
+  for (int q = q0; q < q1; ++q) {
+    float acc = bk;
+#pragma unroll
+    for (int r = 0; r < 3; ++r) {
+      const float* row = xs + r * sstride + 3 * q;
+#pragma unroll
+      for (int i = 0; i < 9; ++i) acc += row[i] * wr[r * 9 + i];
+    }
+    if (act == 1) acc = fmaxf(acc, 0.f);
+    if (k < K) y[(mrow + q) * K + k] = F16<T16>::from_f32(acc);
+  }
+}
+
+template <typename T16>
+__global__ __launch_bounds__(256) void conv_wgrad_stem3_lds(
+    const T16* __restrict__ x, const T16* __restrict__ dy,
+    float* __restrict__ part,  // [gridDim.x*4][K*3*3*3]
+    int N, int H, int W, int K, int Ho, int Wo, int rows_per_chunk) {
+  extern __shared__ __attribute__((aligned(16))) char smem3[];
+  const int selems = (Wo + 2) * 3;
+  const int sstride = (selems + 4) & ~3;
+  float* xs = reinterpret_cast<float*>(smem3);
+  T16* dys = reinterpret_cast<T16*>(smem3 + 3 * sstride * 4);
+
+  const int k = threadIdx.x & 63;
+  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
+  const long nrows = (long)N * Ho;
+  const long row0 = (long)blockIdx.x * rows_per_chunk;
+  const long row1 = min(nrows, row0 + rows_per_chunk);
+
+  float acc[3][9] = {};
+  const int qper = (Wo + 3) / 4;
+  const int q0 = wv * qper, q1 = min(Wo, q0 + qper);
+
+  for (long row = row0; row < row1; ++row) {
+    const int p = (int)(row % Ho);
+    const int n = (int)(row / Ho);
+    __syncthreads();
+    for (int i = threadIdx.x; i < 3 * sstride; i += 256) {
+      const int rr = i / sstride;
+      const int e = i - rr * sstride;
+      const int cs = e / 3, c = e - cs * 3;
+      const int ih = p - 1 + rr, iw = cs - 1;
+      const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
+                      (unsigned)iw < (unsigned)W;
+      xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
+                 : 0.f;
+    }
+    const T16* dyrow = dy + row * Wo * K;
+    for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    __syncthreads();
+
+    for (int q = q0; q < q1; ++q) {
+      const float dyv = (k < K) ? F16<T16>::to_f32(dys[q * K + k]) : 0.f;
+#pragma unroll
+      for (int r = 0; r < 3; ++r) {
+        const float* row_ = xs + r * sstride + 3 * q;
+#pragma unroll
+        for (int i = 0; i < 9; ++i) acc[r][i] += dyv * row_[i];
+      }
+    }
+  }
+
+  if (k < K) {
+    const long E = (long)K * 27;
+    float* slab = part + ((long)blockIdx.x * 4 + wv) * E;
+#pragma unroll
+    for (int r = 0; r < 3; ++r)
+#pragma unroll
+      for (int i = 0; i < 9; ++i) {
+        const int s = i / 3, c = i % 3;
+        // parameter layout [K, C, R, S]
+        slab[(((long)k * 3 + c) * 3 + r) * 3 + s] = acc[r][i];
+      }
+  }
+}
+
 // ---- 7x7/stride-2 stem wgrad (C=3, K<=64) ------------------------------
 // Lane = output channel k; block (chunk, r) accumulates filter row r.
 // dw[k, r, s, c] += dy[m, k] * x[m, tap(r,s,c)] over the block's m-range.
@@ -945,6 +1066,24 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     });
     return y;
   }
+  // CIFAR stem 3x3/s1/p1: LDS-staged variant (see conv_fwd_stem3_lds)
+  if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      P == H && Q == W) {
+    const long wrow = w.dim() == 2 ? w.size(1) : 27L;
+    const int selems = (Q + 2) * 3;
+    const int sstride = (selems + 4) & ~3;
+    const size_t smem = (size_t)3 * sstride * 4;
+    const int has_bias = bias.numel() > 0;
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((conv_fwd_stem3_lds<T16>), dim3((unsigned)(N * P)),
+                         dim3(256), smem, cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         (T16*)y.data_ptr(), N, H, W, K, P, Q, wrow,
+                         (int)act, has_bias);
+    });
+    return y;
+  }
   const bool smallc_fwd = K <= 64 && R == 3 && S == 3 && (C == 3 || C == 6) &&
                           (stride == 1 || stride == 2);
   if (smallc_fwd) {
@@ -1063,6 +1202,28 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     // with plain stores (nchunks==1) or via wgrad_reduce_chunks (nchunks>1)
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad);
+    return dw;
+  }
+  if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      P == H && Q == W) {
+    const long nrows = (long)N * P;
+    long nc = std::min<long>(512, nrows);
+    const int rows_per_chunk = (int)cdiv_l(nrows, nc);
+    nc = cdiv_l(nrows, rows_per_chunk);
+    const long E = (long)K * 27;
+    auto dw = at::empty({K, 3L, 3L, 3L}, x.options().dtype(at::kFloat));
+    auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));
+    const int selems = (Q + 2) * 3;
+    const int sstride = (selems + 4) & ~3;
+    const size_t smem = (size_t)3 * sstride * 4 + (size_t)Q * K * 2;
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((conv_wgrad_stem3_lds<T16>), dim3((unsigned)nc),
+                         dim3(256), smem, cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)dy.data_ptr(),
+                         partl.data_ptr<float>(), N, H, W, K, P, Q,
+                         rows_per_chunk);
+    });
+    wgrad_reduce_launch(partl, dw, E, nc * 4);
     return dw;
   }
   const bool smallc = K <= 64 &&
