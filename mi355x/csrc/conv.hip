@@ -277,17 +277,7 @@ __global__ __launch_bounds__(256) void conv_fwd_smallc(
   }
 }
 
-// ---- 7x7/stride-2 ImageNet stem fwd (C=3, K<=64) -----------------------
-// Lane = output channel k; all 64 lanes compute the SAME output pixel, so
-// every x address is wave-uniform (scalar-unit friendly). The 7 input rows
-// live in per-row CIRCULAR buffers of 12 packed bf16/f16 PAIRS; the q-walk
-// advances 3 pairs (= 6 elements = stride 2 x C 3) per output, so the
-// rotation phase cycles with period 4 and the q-loop is unrolled x4 with
-// COMPILE-TIME buffer indices (guide rule 20: no runtime-indexed register
-// arrays, no data shifting). Each tap pair is one v_dot2c_f32_bf16 —
-// 7 rows x 11 pairs = 77 dot2 per output vs the GENC gather path's
-// per-element staging (measured 12.3 ms/call on the ResNet-50 stem; this
-// kernel replaces it).
+// ---- packed bf16/f16 dot2 helpers (stem kernels) -----------------------
 template <typename T16>
 struct Dot2;
 template <>
@@ -315,157 +305,6 @@ __device__ __forceinline__ typename Dot2<T16>::v2 pack2(T16 a, T16 b) {
   typename Dot2<T16>::v2 r;
   __builtin_memcpy(&r, &u, 4);
   return r;
-}
-
-template <typename T16, bool LDSW>
-__global__ __launch_bounds__(256) void conv_fwd_stem7(
-    const T16* __restrict__ x, const T16* __restrict__ wgt,
-    const float* __restrict__ bias, const T16* __restrict__ zpage,
-    T16* __restrict__ y, int N, int H, int W, int K, int Ho, int Wo,
-    int pad, long wrow_stride, int act, int has_bias, long m_per_chunk,
-    long Mtot) {
-  using V2 = typename Dot2<T16>::v2;
-  constexpr int ROWS = 7, WPAIR = 11, BUFP = 12;
-  const int k = threadIdx.x & 63;
-  // readfirstlane: wv is uniform within a wave, but divergence analysis
-  // can't see that through threadIdx.x — forcing it scalar moves the whole
-  // m/n/p/q walk and every x address into SGPRs (VGPR budget is what caps
-  // occupancy here)
-  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
-
-  // per-lane packed weights: wp[r][u] = (w[3s+c = 2u], w[2u+1]); last pair's
-  // hi half is zero (21 taps/row -> 10.5 pairs). LDSW=true keeps them in
-  // LDS instead of 77 VGPRs (occupancy 2 -> 3; lanes read stride-77 dwords
-  // = conflict-free) at the cost of one ds_read per dot2.
-  __shared__ V2 lwp[LDSW ? 64 * ROWS * WPAIR : 1];
-  V2 wp[LDSW ? 1 : ROWS][WPAIR];
-  {
-    const T16* wk = wgt + (long)min(k, K - 1) * wrow_stride;
-    const T16 z{};
-    if constexpr (LDSW) {
-      if (wv == 0) {
-#pragma unroll
-        for (int r = 0; r < ROWS; ++r) {
-          const T16* wr = wk + r * 21;
-#pragma unroll
-          for (int u = 0; u < WPAIR - 1; ++u)
-            lwp[(k * ROWS + r) * WPAIR + u] = pack2(wr[2 * u], wr[2 * u + 1]);
-          lwp[(k * ROWS + r) * WPAIR + WPAIR - 1] = pack2(wr[20], z);
-        }
-      }
-      __syncthreads();
-    } else {
-#pragma unroll
-      for (int r = 0; r < ROWS; ++r) {
-        const T16* wr = wk + r * 21;
-#pragma unroll
-        for (int u = 0; u < WPAIR - 1; ++u)
-          wp[r][u] = pack2(wr[2 * u], wr[2 * u + 1]);
-        wp[r][WPAIR - 1] = pack2(wr[20], z);
-      }
-    }
-  }
-  const float bk = has_bias ? bias[min(k, K - 1)] : 0.f;
-
-  const long m_begin = (long)blockIdx.x * m_per_chunk;
-  const long m_end = min(Mtot, m_begin + m_per_chunk);
-  const long per_wave = (m_end - m_begin + 3) / 4;
-  long m = m_begin + wv * per_wave;
-  const long w1 = min(m_end, m + per_wave);
-  if (m >= w1) return;
-
-  int q = (int)(m % Wo);
-  long np = m / Wo;
-  int p = (int)(np % Ho);
-  int n = (int)(np / Ho);
-
-  V2 buf[ROWS][BUFP];
-
-  while (m < w1) {
-    // ---- segment (n, p, q..): row bases + full refill --------------
-    const int ih0 = 2 * p - pad;
-    const T16* rowb[ROWS];
-#pragma unroll
-    for (int r = 0; r < ROWS; ++r) {
-      const int ih = ih0 + r;
-      rowb[r] = ((unsigned)ih < (unsigned)H)
-                    ? x + ((long)n * H + ih) * W * 3
-                    : nullptr;
-    }
-    const int iw0s = 2 * q - pad;  // stream column origin for this segment
-#pragma unroll
-    for (int r = 0; r < ROWS; ++r) {
-      const T16* rb = rowb[r];
-#pragma unroll
-      for (int t = 0; t < WPAIR; ++t) {
-        const int e0 = 2 * t, e1 = 2 * t + 1;
-        const int col0 = e0 / 3, c0 = e0 % 3;
-        const int col1 = e1 / 3, c1 = e1 % 3;
-        const int iwa = iw0s + col0, iwb = iw0s + col1;
-        const T16* pa = (rb && (unsigned)iwa < (unsigned)W)
-                            ? rb + (long)iwa * 3 + c0 : zpage;
-        const T16* pb = (rb && (unsigned)iwb < (unsigned)W)
-                            ? rb + (long)iwb * 3 + c1 : zpage + 1;
-        buf[r][t] = pack2(*pa, *pb);
-      }
-    }
-
-    // ---- steady state: 4-phase unrolled q-walk ---------------------
-#define STEM7_PHASE(PH)                                                      \
-  {                                                                          \
-    float acc = bk;                                                          \
-    _Pragma("unroll") for (int r = 0; r < ROWS; ++r) {                       \
-      _Pragma("unroll") for (int u = 0; u < WPAIR; ++u) {                    \
-        const V2 wv_ = LDSW ? lwp[(k * ROWS + r) * WPAIR + u] : wp[LDSW ? 0 : r][u]; \
-        acc = Dot2<T16>::fma(buf[r][(3 * (PH) + u) % BUFP], wv_, acc);       \
-      }                                                                      \
-    }                                                                        \
-    if (act == 1) acc = fmaxf(acc, 0.f);                                     \
-    if (k < K) y[m * K + k] = F16<T16>::from_f32(acc);                       \
-    ++m;                                                                     \
-    ++q;                                                                     \
-    if (q == Wo || m == w1) goto segment_done;                               \
-    /* fresh pairs for the next output: elements at columns               */ \
-    /* iw1, iw1, iw1+1, iw1+1, iw1+1, iw1+2 with c = 1,2,0,1,2,0          */ \
-    {                                                                        \
-      const int iw1 = 2 * (q - 1) + 7 - pad;                                 \
-      _Pragma("unroll") for (int r = 0; r < ROWS; ++r) {                     \
-        const T16* rb = rowb[r];                                             \
-        const T16* p0 = (rb && (unsigned)(iw1) < (unsigned)W)                \
-                            ? rb + (long)(iw1)*3 + 1 : zpage;                \
-        const T16* p1 = (rb && (unsigned)(iw1) < (unsigned)W)                \
-                            ? rb + (long)(iw1)*3 + 2 : zpage + 1;            \
-        const T16* p2 = (rb && (unsigned)(iw1 + 1) < (unsigned)W)            \
-                            ? rb + (long)(iw1 + 1) * 3 : zpage + 2;          \
-        const T16* p3 = (rb && (unsigned)(iw1 + 1) < (unsigned)W)            \
-                            ? rb + (long)(iw1 + 1) * 3 + 1 : zpage + 3;      \
-        const T16* p4 = (rb && (unsigned)(iw1 + 1) < (unsigned)W)            \
-                            ? rb + (long)(iw1 + 1) * 3 + 2 : zpage + 4;      \
-        const T16* p5 = (rb && (unsigned)(iw1 + 2) < (unsigned)W)            \
-                            ? rb + (long)(iw1 + 2) * 3 : zpage + 5;          \
-        buf[r][(3 * (PH) + 11) % BUFP] = pack2(*p0, *p1);                    \
-        buf[r][(3 * (PH) + 12) % BUFP] = pack2(*p2, *p3);                    \
-        buf[r][(3 * (PH) + 13) % BUFP] = pack2(*p4, *p5);                    \
-      }                                                                      \
-    }                                                                        \
-  }
-
-    for (;;) {
-      STEM7_PHASE(0)
-      STEM7_PHASE(1)
-      STEM7_PHASE(2)
-      STEM7_PHASE(3)
-    }
-#undef STEM7_PHASE
-  segment_done:
-    if (q == Wo) {
-      q = 0;
-      if (++p == Ho) {
-        p = 0;
-        ++n;
-      }
-    }
-  }
 }
 
 // ---- 7x7/stride-2 stem fwd v3: cooperative LDS staging -----------------
@@ -761,123 +600,6 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
   }
 }
 
-template <typename T16>
-__global__ __launch_bounds__(256) void conv_wgrad_stem7(
-    const T16* __restrict__ x, const T16* __restrict__ dy,
-    const T16* __restrict__ zpage,
-    float* __restrict__ part,  // [gridDim.x*4][K*R*S*C] (r-slice per block)
-    int N, int H, int W, int K, int Ho, int Wo, int pad, long m_per_chunk,
-    long Mtot) {
-  constexpr int C_ = 3, S_ = 7, APAIR = 11, BUFP = 18;
-  const int r = blockIdx.y;
-  const int k = threadIdx.x & 63;
-  const int wv = __builtin_amdgcn_readfirstlane(threadIdx.x >> 6);
-
-  float2v acc[APAIR] = {};
-  const long m_begin = (long)blockIdx.x * m_per_chunk;
-  const long m_end = min(Mtot, m_begin + m_per_chunk);
-  const long per_wave = (m_end - m_begin + 3) / 4;
-  long m = m_begin + wv * per_wave;
-  const long w1 = min(m_end, m + per_wave);
-
-  if (m < w1) {
-    int q = (int)(m % Wo);
-    long np = m / Wo;
-    int p = (int)(np % Ho);
-    int n = (int)(np / Ho);
-
-    float2v buf[BUFP];
-
-    auto ld = [&](const T16* rb, int iw, int c) -> float {
-      const T16* src = (rb && (unsigned)iw < (unsigned)W)
-                           ? rb + (long)iw * C_ + c
-                           : zpage;
-      return F16<T16>::to_f32(*src);
-    };
-
-    while (m < w1) {
-      const int ih = 2 * p - pad + r;
-      const T16* rb = ((unsigned)ih < (unsigned)H)
-                          ? x + ((long)n * H + ih) * W * C_
-                          : nullptr;
-      const int iw0s = 2 * q - pad;  // stream column origin
-      // prefill pairs 0..13 (this output's window + the next one's)
-#pragma unroll
-      for (int t = 0; t < 14; ++t) {
-        const int e0 = 2 * t, e1 = 2 * t + 1;
-        float2v v;
-        v.x = ld(rb, iw0s + e0 / 3, e0 % 3);
-        v.y = ld(rb, iw0s + e1 / 3, e1 % 3);
-        buf[t] = v;
-      }
-
-#define WG7_PHASE(PH)                                                        \
-  {                                                                          \
-    const float dyv = (k < K) ? F16<T16>::to_f32(dy[m * K + k]) : 0.f;       \
-    float2v d2;                                                              \
-    d2.x = dyv;                                                              \
-    d2.y = dyv;                                                              \
-    _Pragma("unroll") for (int u = 0; u < APAIR; ++u) {                      \
-      acc[u] += d2 * buf[(3 * (PH) + u) % BUFP];                             \
-    }                                                                        \
-    ++m;                                                                     \
-    ++q;                                                                     \
-    if (q == Wo || m == w1) goto seg_done;                                   \
-    {                                                                        \
-      /* fresh pairs (3j+14..16) for output j+2: elements at columns      */ \
-      /* iwf, iwf, iwf+1, iwf+1, iwf+1, iwf+2 with c = 1,2,0,1,2,0        */ \
-      const int iwf = 2 * (q - 1) + 9 - pad;                                 \
-      float2v f0, f1, f2;                                                    \
-      f0.x = ld(rb, iwf, 1);                                                 \
-      f0.y = ld(rb, iwf, 2);                                                 \
-      f1.x = ld(rb, iwf + 1, 0);                                             \
-      f1.y = ld(rb, iwf + 1, 1);                                             \
-      f2.x = ld(rb, iwf + 1, 2);                                             \
-      f2.y = ld(rb, iwf + 2, 0);                                             \
-      buf[(3 * (PH) + 14) % BUFP] = f0;                                      \
-      buf[(3 * (PH) + 15) % BUFP] = f1;                                      \
-      buf[(3 * (PH) + 16) % BUFP] = f2;                                      \
-    }                                                                        \
-  }
-
-      for (;;) {
-        WG7_PHASE(0)
-        WG7_PHASE(1)
-        WG7_PHASE(2)
-        WG7_PHASE(3)
-        WG7_PHASE(4)
-        WG7_PHASE(5)
-      }
-#undef WG7_PHASE
-    seg_done:
-      if (q == Wo) {
-        q = 0;
-        if (++p == Ho) {
-          p = 0;
-          ++n;
-        }
-      }
-    }
-  }
-
-  if (k < K) {
-    const long E = (long)K * S_ * S_ * C_;
-    float* slab = part + ((long)blockIdx.x * 4 + wv) * E;
-    // store in the parameter layout [K, C, R, S]: tap index 3s+c = pair
-    // element 2u (+1); last pair's hi half is the dead 22nd slot
-#pragma unroll
-    for (int u = 0; u < APAIR; ++u) {
-      const int e0 = 2 * u, e1 = 2 * u + 1;
-      const int s0 = e0 / 3, c0 = e0 % 3;
-      slab[(((long)k * C_ + c0) * S_ + r) * S_ + s0] = acc[u].x;
-      if (e1 < 21) {
-        const int s1 = e1 / 3, c1 = e1 % 3;
-        slab[(((long)k * C_ + c1) * S_ + r) * S_ + s1] = acc[u].y;
-      }
-    }
-  }
-}
-
 // ---- small-C wgrad (stems: C=3/6, K<=64) -------------------------------
 // lane = output channel k, all S_*C_ taps of one filter row r in REGISTERS
 // (compile-time C_/S_ keep the accumulator array in VGPRs — guide rule
@@ -1024,63 +746,16 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     nchunks = cdiv_l(M, m_per_chunk);
     const int has_bias = bias.numel() > 0;
     at::Tensor zp = conv_zero_page(x);
-    static const bool v2 = [] {  // fallback knob: register-window variant
-      const char* e = getenv("MI355X_STEM7_V2");
-      return e && e[0] == '1';
-    }();
-    if (!v2) {
-      const int selems = (2 * Q + 5) * 3;
-      const int sstride = (selems + 4) & ~3;
-      const size_t smem = (size_t)7 * sstride * x.element_size();
-      DISPATCH_16(x, T16, {
-        hipLaunchKernelGGL((conv_fwd_stem7_lds<T16>), dim3((unsigned)(N * P)),
-                           dim3(256), smem, cur_stream(),
-                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
-                           has_bias ? bias.data_ptr<float>() : nullptr,
-                           (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)pad,
-                           wrow, (int)act, has_bias);
-      });
-      return y;
-    }
-    static const bool ldsw = [] {  // A/B knob: weights in LDS vs VGPR
-      const char* e = getenv("MI355X_STEM7_LDSW");
-      return !e || e[0] == '1';  // default on (occupancy 2 -> 3)
-    }();
-    DISPATCH_16(x, T16, {
-      if (ldsw)
-        hipLaunchKernelGGL((conv_fwd_stem7<T16, true>), dim3(nchunks),
-                           dim3(256), 0, cur_stream(),
-                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
-                           has_bias ? bias.data_ptr<float>() : nullptr,
-                           (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N,
-                           H, W, K, P, Q, (int)pad, wrow, (int)act, has_bias,
-                           m_per_chunk, M);
-      else
-        hipLaunchKernelGGL((conv_fwd_stem7<T16, false>), dim3(nchunks),
-                           dim3(256), 0, cur_stream(),
-                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
-                           has_bias ? bias.data_ptr<float>() : nullptr,
-                           (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N,
-                           H, W, K, P, Q, (int)pad, wrow, (int)act, has_bias,
-                           m_per_chunk, M);
-    });
-    return y;
-  }
-  // CIFAR stem 3x3/s1/p1: LDS-staged variant (see conv_fwd_stem3_lds)
-  if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&
-      P == H && Q == W) {
-    const long wrow = w.dim() == 2 ? w.size(1) : 27L;
-    const int selems = (Q + 2) * 3;
+    const int selems = (2 * Q + 5) * 3;
     const int sstride = (selems + 4) & ~3;
-    const size_t smem = (size_t)3 * sstride * 4;
-    const int has_bias = bias.numel() > 0;
+    const size_t smem = (size_t)7 * sstride * x.element_size();
     DISPATCH_16(x, T16, {
-      hipLaunchKernelGGL((conv_fwd_stem3_lds<T16>), dim3((unsigned)(N * P)),
+      hipLaunchKernelGGL((conv_fwd_stem7_lds<T16>), dim3((unsigned)(N * P)),
                          dim3(256), smem, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (T16*)y.data_ptr(), N, H, W, K, P, Q, wrow,
-                         (int)act, has_bias);
+                         (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)pad,
+                         wrow, (int)act, has_bias);
     });
     return y;
   }
@@ -1236,43 +911,26 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     nchunks = cdiv_l(M, m_per_chunk);
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     if (C == 3 && S == 7 && stride == 2 && pad <= 4) {
-      static const bool v2 = [] {
-        const char* e = getenv("MI355X_STEM7_V2");
-        return e && e[0] == '1';
-      }();
-      if (!v2) {
-        const long nrows = (long)N * P;
-        // 512 blocks x 4 waves at occupancy 2 fills the chip exactly once;
-        // fewer chunks = 4x less partial-slab traffic for the reduce
-        long nc = std::min<long>(512, nrows);
-        const int rows_per_chunk = (int)cdiv_l(nrows, nc);
-        nc = cdiv_l(nrows, rows_per_chunk);
-        auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));
-        const int selems = (2 * Q + 5) * 3;
-        const int sstride = (selems + 4) & ~3;
-        const size_t smem = (size_t)7 * sstride * 4 + (size_t)Q * K * 2;
-        DISPATCH_16(x, T16, {
-          hipLaunchKernelGGL((conv_wgrad_stem7_lds<T16>), dim3((unsigned)nc),
-                             dim3(256), smem, cur_stream(),
-                             (const T16*)x.data_ptr(),
-                             (const T16*)dy.data_ptr(),
-                             partl.data_ptr<float>(), N, H, W, K, P, Q,
-                             (int)pad, rows_per_chunk);
-        });
-        wgrad_reduce_launch(partl, dw, E, nc * 4);
-        return dw;
-      }
-      auto part = at::empty({nchunks * 4, E}, x.options().dtype(at::kFloat));
-      at::Tensor zpw = conv_zero_page(x);
-      dim3 grid((unsigned)nchunks, R);
+
+      const long nrows = (long)N * P;
+      // 512 blocks x 4 waves at occupancy 2 fills the chip exactly once;
+      // fewer chunks = 4x less partial-slab traffic for the reduce
+      long nc = std::min<long>(512, nrows);
+      const int rows_per_chunk = (int)cdiv_l(nrows, nc);
+      nc = cdiv_l(nrows, rows_per_chunk);
+      auto partl = at::empty({nc * 4, E}, x.options().dtype(at::kFloat));
+      const int selems = (2 * Q + 5) * 3;
+      const int sstride = (selems + 4) & ~3;
+      const size_t smem = (size_t)7 * sstride * 4 + (size_t)Q * K * 2;
       DISPATCH_16(x, T16, {
-        hipLaunchKernelGGL((conv_wgrad_stem7<T16>), grid, dim3(256), 0,
-                           cur_stream(), (const T16*)x.data_ptr(),
+        hipLaunchKernelGGL((conv_wgrad_stem7_lds<T16>), dim3((unsigned)nc),
+                           dim3(256), smem, cur_stream(),
+                           (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(),
-                           (const T16*)zpw.data_ptr(), part.data_ptr<float>(),
-                           N, H, W, K, P, Q, (int)pad, m_per_chunk, M);
+                           partl.data_ptr<float>(), N, H, W, K, P, Q,
+                           (int)pad, rows_per_chunk);
       });
-      wgrad_reduce_launch(part, dw, E, nchunks * 4);
+      wgrad_reduce_launch(partl, dw, E, nc * 4);
       return dw;
     }
     // empty, not zeros: every slab element is written (acc starts at 0 and
