@@ -485,13 +485,11 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
   const bool do_dy = tid < 2 * KT;
   const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
   const int t = do_dy ? tid : 0;
-  // k/c varies FASTEST across lanes: adjacent lanes read adjacent memory
-  // (the m-fast map issued 16B loads 1 KB apart per lane — uncoalesced)
-  const int sk = (t % (KT / 8)) * 8;     // k-element offset (dyT rows)
-  const int sm = (t / (KT / 8)) * 4;     // m offset (4 rows)
+  const int sm = (t & 15) * 4;           // m offset (4 rows)
+  const int sk = (t >> 4) * 8;           // k-element offset (dyT rows)
   const int tx = tid & 127;
-  const int skx = (tx & 7) * 8;
-  const int smx = (tx >> 3) * 4;
+  const int smx = (tx & 15) * 4;
+  const int skx = (tx >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -649,8 +647,8 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
 
   // staging: all 256 threads stage dy (4m x 8k over 128 k-rows) and x
   // (4m x 8c over 128 c-rows) with the same (m-group, row-group) map
-  const int sk = (tid & 15) * 8;  // k/c fastest: coalesced staging loads
-  const int sm = (tid >> 4) * 4;
+  const int sm = (tid & 15) * 4;
+  const int sk = (tid >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -790,12 +788,11 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
   const bool do_dy = tid < 2 * KT;
   const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
   const int t = do_dy ? tid : 0;
-  // k/c fastest across lanes (coalesced gathers; see the plain kernel)
-  const int sk = (t % (KT / 8)) * 8;
-  const int sm = (t / (KT / 8)) * 4;
+  const int sm = (t & 15) * 4;
+  const int sk = (t >> 4) * 8;
   const int tx = tid & 127;
-  const int skx = (tx & 7) * 8;
-  const int smx = (tx >> 3) * 4;
+  const int smx = (tx & 15) * 4;
+  const int skx = (tx >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
