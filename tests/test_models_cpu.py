@@ -112,3 +112,23 @@ def test_checkpoint_roundtrip(tmp_path):
 def test_build_model_names():
     for name in ["net", "resnet18", "resnet50"]:
         assert build_model(name) is not None
+
+
+def test_bn_num_batches_tracked_state_dict_parity():
+    """The host-side BN step counter must still round-trip through
+    state_dict under the torch key (torch checkpoints interchange)."""
+    from mi355x.models.layers import BatchNorm2d
+
+    bn = BatchNorm2d(8)
+    x = torch.randn(2, 4, 4, 8)
+    bn.train()
+    bn(x)
+    bn(x)
+    sd = bn.state_dict()
+    assert "num_batches_tracked" in sd
+    assert int(sd["num_batches_tracked"]) == 2
+
+    bn2 = BatchNorm2d(8)
+    bn2.load_state_dict(sd)
+    assert bn2._nbt == 2
+    assert int(bn2.num_batches_tracked) == 2
