@@ -1,11 +1,17 @@
 // NHWC conv2d: forward, dgrad, wgrad (SURVEY.md N5 — the reference's
 // cuDNN/MIOpen implicit-GEMM conv, rebuilt for CDNA4).
 //
-// v0: direct kernels, correctness-first, laid out so the C (channel-minor)
-// dimension is the coalesced/vectorized axis everywhere. The MFMA
-// implicit-GEMM path (conv_mfma.hip) replaces these for the hot shapes;
-// these remain the fallback for odd geometries and the numerics reference
-// on-device.
+// Three tiers live here:
+//  - direct kernels (correctness-first fallback for odd geometries and the
+//    on-device numerics reference), C channel-minor coalesced everywhere;
+//  - small-C register-window kernels (C=3/6 stems the MFMA path cannot
+//    tile: C % 64 != 0);
+//  - LDS-staged stem kernels (7x7/s2 ImageNet and 3x3/s1 CIFAR stems):
+//    one block stages the input strip for an output row cooperatively,
+//    taps are LDS reads, packed bf16 v_dot2c (7x7) or plain f32 FMA (3x3)
+//    accumulation. These replaced the register-window variants for those
+//    shapes after PMC showed them latency-bound (profiles/).
+// The MFMA implicit-GEMM path (conv_mfma.hip) takes every C%64==0 shape.
 #include "common.h"
 
 namespace {
