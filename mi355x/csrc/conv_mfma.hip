@@ -348,31 +348,22 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
   const int sb_n = tid >> 2;
   const int sb_c = (tid & 3) * 16;
   const T16* wrowB = wgt + (long)(k0 + sb_n) * b_row_stride + sb_c;
-  // two register sets (named: a runtime index would spill) and two LDS
-  // buffers, ping-ponged by GLOBAL tap parity: one barrier per tap — the
-  // write of tap g+1 lands in the buffer tap g's MFMAs are not reading
-  short8 brA[2], brB[2];
-  auto load_b = [&](int tap, int cc, short8 (&br)[2]) {
+  short8 breg[2];
+  auto load_b = [&](int tap, int cc) {
     const T16* wp = wrowB + (long)tap * b_rs_stride + cc * BK;
-    br[0] = *reinterpret_cast<const short8*>(wp);
-    br[1] = *reinterpret_cast<const short8*>(wp + 8);
+    breg[0] = *reinterpret_cast<const short8*>(wp);
+    breg[1] = *reinterpret_cast<const short8*>(wp + 8);
   };
-  auto stage_b = [&](int buf, short8 (&br)[2]) {
-    short* pb = reinterpret_cast<short*>(ldsB + (buf * 64 + sb_n) * LDK +
-                                         sb_c);
-    *reinterpret_cast<short8*>(pb) = br[0];
-    *reinterpret_cast<short8*>(pb + 8) = br[1];
+  auto stage_b = [&]() {
+    short* pb = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
+    *reinterpret_cast<short8*>(pb) = breg[0];
+    *reinterpret_cast<short8*>(pb + 8) = breg[1];
   };
 
-  load_b(0, 0, brA);
-  load_b(1, 0, brB);
+  load_b(0, 0);
 
   for (int cc = 0; cc < cchunks; ++cc) {
-    const int gpar = (cc * 9) & 1;  // LDS-buffer parity of this cc's tap 0
     __syncthreads();  // previous c-chunk's patch reads done
-    // brA holds B(cc,0), brB holds B(cc,1): primed pre-loop for cc 0 and
-    // by the previous cc's tap-7/8 tail loads afterwards
-    stage_b(gpar, brA);
     // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi ----
     const int ngroups = NR * (Wi + 2) * (BK / 8);
     for (int i = tid; i < ngroups; i += 256) {
@@ -405,35 +396,25 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
                              ? ((long)(1 - r) * (Wi + 2) + (1 - s)) * PCS
                              : ((long)(r - 1) * (Wi + 2) + (s - 1)) * PCS;
         const T16* abase = rv ? patch + lane_base + off : zstub;
-        // regs currently hold B(tap+1): write it to the OTHER LDS buffer
-        // while this tap's MFMAs read theirs; load B(tap+2) regs under it
-        short8 (&br_next)[2] = (tap & 1) ? brA : brB;
-        short8 (&br_load)[2] = (tap & 1) ? brB : brA;
-        if (tap < 8) stage_b((gpar ^ ((tap + 1) & 1)), br_next);
-        if (tap < 7)
-          load_b(tap + 2, cc, br_load);
-        else if (cc + 1 < cchunks) {
-          // prime the next c-chunk: B(cc+1,0) must land in brA (the head
-          // stages brA) and B(cc+1,1) in brB — explicit names, the parity
-          // refs would swap them
-          if (tap == 7)
-            load_b(0, cc + 1, brA);  // brA was staged above: free
-          else
-            load_b(1, cc + 1, brB);  // tap 8 never stages brB: free
-        }
-        const T16* ldsBt = ldsB + (gpar ^ (tap & 1)) * 64 * LDK;
+        __syncthreads();  // previous tap's MFMAs done reading ldsB
+        stage_b();
+        __syncthreads();
+        // next tap's (or next c-chunk's) B burst rides under the MFMAs
+        if (tap < 8)
+          load_b(tap + 1, cc);
+        else if (cc + 1 < cchunks)
+          load_b(0, cc + 1);
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
           const short8 af = *reinterpret_cast<const short8*>(
               abase + kk * 16 + kh * 8);
           const short8 b0 = *reinterpret_cast<const short8*>(
-              ldsBt + li * LDK + kk * 16 + kh * 8);
+              ldsB + li * LDK + kk * 16 + kh * 8);
           const short8 b1 = *reinterpret_cast<const short8*>(
-              ldsBt + (32 + li) * LDK + kk * 16 + kh * 8);
+              ldsB + (32 + li) * LDK + kk * 16 + kh * 8);
           acc[0] = Mfma32<T16>::run(af, b0, acc[0]);
           acc[1] = Mfma32<T16>::run(af, b1, acc[1]);
         }
-        __syncthreads();  // tap done; next tap's buffer fully written
       }
     }
   }
@@ -1584,7 +1565,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     // 128/Wo+3 was ONE short for Wo in {56,28,14,7} (tap r=2 then read
     // past the patch -> diverging ResNet-50 training)
     const int NR = (Wo + 126) / Wo + 3;
-    const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS + 2 * 64 * LDK) * 2;
+    const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), KO / 64);
     DISPATCH_16(x, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
@@ -1654,7 +1635,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && P == H &&
       Q == W && W <= 64) {
     const int NR = (W + 126) / W + 3;  // see fwd launcher comment
-    const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS + 2 * 64 * LDK) * 2;
+    const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), CI / 64);
     DISPATCH_16(dy, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
