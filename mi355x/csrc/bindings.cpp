@@ -25,16 +25,17 @@ at::Tensor bn_stats(at::Tensor x);
 at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
                        at::Tensor running_var, double m_total,
                        double momentum, double eps);
-at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
+std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                     at::Tensor gamma, at::Tensor beta, at::Tensor res,
                     long act);
 at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
-                         at::Tensor mean, at::Tensor invstd);
+                         at::Tensor mean, at::Tensor invstd,
+                         at::Tensor mask);
 std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
                                   at::Tensor mean, at::Tensor invstd,
                                   at::Tensor gamma, at::Tensor dgamma,
                                   at::Tensor dbeta, double m_total,
-                                  bool want_dres);
+                                  bool want_dres, at::Tensor mask);
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
                                     long pad);

@@ -127,7 +127,7 @@ __global__ void bn_bwd_dx_kernel(const T16* __restrict__ x,
 template <typename T16, bool BWD>
 __global__ __launch_bounds__(256) void bn_reduce_fast(
     const T16* __restrict__ x, const T16* __restrict__ dy,
-    const T16* __restrict__ y, const float* __restrict__ mean,
+    const unsigned char* __restrict__ msk, const float* __restrict__ mean,
     const float* __restrict__ invstd, float* __restrict__ out,  // [2,C]
     long E, int C, long e_per_block) {
   extern __shared__ __attribute__((aligned(16))) float lsum[];  // [2*C]
@@ -159,12 +159,11 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
     } else {
       const short8 vx = *reinterpret_cast<const short8*>(x + e);
       const short8 vd = *reinterpret_cast<const short8*>(dy + e);
-      short8 vy = {};
-      if (y) vy = *reinterpret_cast<const short8*>(y + e);
+      const unsigned mb = msk ? msk[e >> 3] : 0xffu;
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
         float d = s16_to_f32<T16>(vd[u]);
-        if (y && s16_to_f32<T16>(vy[u]) <= 0.f) d = 0.f;
+        if (!((mb >> u) & 1)) d = 0.f;
         const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
         s0[u] += d * xh;
         s1[u] += d;
@@ -216,7 +215,8 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
     const T16* __restrict__ x, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ beta, const T16* __restrict__ res,
-    T16* __restrict__ y, long E, int C, int act) {
+    T16* __restrict__ y, unsigned char* __restrict__ msk, long E, int C,
+    int act) {
   const int c0 = ((blockIdx.x * blockDim.x + threadIdx.x) * 8) % C;
   float sc[8], sh[8];
 #pragma unroll
@@ -231,21 +231,26 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
     short8 vr = {};
     if (res) vr = *reinterpret_cast<const short8*>(res + e);
     short8 o;
+    unsigned mb = 0;
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       float v = s16_to_f32<T16>(vx[u]) * sc[u] + sh[u];
       if (res) v += s16_to_f32<T16>(vr[u]);
       if (act == 1) v = fmaxf(v, 0.f);
       o[u] = f32_to_s16<T16>(v);
+      // mask bit mirrors the bf16-rounded "y > 0" test the backward used
+      // to make by re-reading y (1/16 the bytes)
+      if (s16_to_f32<T16>(o[u]) > 0.f) mb |= 1u << u;
     }
     *reinterpret_cast<short8*>(y + e) = o;
+    if (msk) msk[e >> 3] = (unsigned char)mb;
   }
 }
 
 template <typename T16>
 __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     const T16* __restrict__ x, const T16* __restrict__ dy,
-    const T16* __restrict__ y, const float* __restrict__ mean,
+    const unsigned char* __restrict__ msk, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
     T16* __restrict__ dx, T16* __restrict__ dres, long E, int C,
@@ -265,13 +270,12 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
        e += stride) {
     const short8 vx = *reinterpret_cast<const short8*>(x + e);
     const short8 vd = *reinterpret_cast<const short8*>(dy + e);
-    short8 vy = {};
-    if (y) vy = *reinterpret_cast<const short8*>(y + e);
+    const unsigned mb = msk ? msk[e >> 3] : 0xffu;
     short8 odx, ods;
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       float d = s16_to_f32<T16>(vd[u]);
-      if (y && s16_to_f32<T16>(vy[u]) <= 0.f) d = 0.f;
+      if (!((mb >> u) & 1)) d = 0.f;
       if (dres) ods[u] = f32_to_s16<T16>(d);
       const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
       odx[u] = f32_to_s16<T16>(g_[u] * (d - b_[u] - xh * a_[u]));
@@ -380,9 +384,13 @@ at::Tensor bn_stats(at::Tensor x) {
   return out;
 }
 
-at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
-                    at::Tensor gamma, at::Tensor beta, at::Tensor res,
-                    long act) {
+// returns {y, mask}: mask is a 1-bit-per-element "y > 0" map (u8, one
+// byte per 8 channel-elements) when act==relu on the fast path — the
+// backward reads it instead of re-reading y (1/16 the bytes). Empty
+// tensor otherwise (the fallback kernels keep the y-based masking).
+std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean,
+                                 at::Tensor invstd, at::Tensor gamma,
+                                 at::Tensor beta, at::Tensor res, long act) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
@@ -392,6 +400,9 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
   const int has_res = res.numel() > 0;
   if (bn_fast_ok(M, C)) {
     const long E = M * (long)C;
+    at::Tensor mask = act == 1
+                          ? at::empty({E / 8}, x.options().dtype(at::kByte))
+                          : at::Tensor();
     const int grid = (int)std::min<long>(cdiv_l(E, 256 * 8), 2048);
     DISPATCH_16(x, T16, {
       hipLaunchKernelGGL(bn_apply_fast<T16>, dim3(grid), dim3(256), 0,
@@ -399,9 +410,12 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), beta.data_ptr<float>(),
                          has_res ? (const T16*)res.data_ptr() : nullptr,
-                         (T16*)y.data_ptr(), E, C, (int)act);
+                         (T16*)y.data_ptr(),
+                         mask.defined() ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
+                         E, C, (int)act);
     });
-    return y;
+    return {y, mask.defined() ? mask : at::Tensor()};
   }
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL(bn_apply_kernel<T16>, dim3(ew_grid2(M * C)), dim3(256),
@@ -411,11 +425,12 @@ at::Tensor bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                        has_res ? (const T16*)res.data_ptr() : nullptr,
                        (T16*)y.data_ptr(), M, C, (int)act, has_res);
   });
-  return y;
+  return {y, at::Tensor()};
 }
 
 at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
-                         at::Tensor mean, at::Tensor invstd) {
+                         at::Tensor mean, at::Tensor invstd,
+                         at::Tensor mask) {
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
@@ -430,7 +445,8 @@ at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
       hipLaunchKernelGGL((bn_reduce_fast<T16, true>), grid, dim3(256),
                          2 * C * sizeof(float), cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)dy.data_ptr(),
-                         y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                         mask.numel() ? mask.data_ptr<unsigned char>()
+                                      : nullptr,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          out.data_ptr<float>(), E, C, e_per_block);
     });
@@ -456,7 +472,7 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
                                   at::Tensor mean, at::Tensor invstd,
                                   at::Tensor gamma, at::Tensor dgamma,
                                   at::Tensor dbeta, double m_total,
-                                  bool want_dres) {
+                                  bool want_dres, at::Tensor mask) {
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
@@ -470,7 +486,8 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
       hipLaunchKernelGGL(bn_bwd_dx_fast<T16>, dim3(grid), dim3(256), 0,
                          cur_stream(), (const T16*)x.data_ptr(),
                          (const T16*)dy.data_ptr(),
-                         y.numel() ? (const T16*)y.data_ptr() : nullptr,
+                         mask.numel() ? mask.data_ptr<unsigned char>()
+                                      : nullptr,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
                          dbeta.data_ptr<float>(), (T16*)dx.data_ptr(),

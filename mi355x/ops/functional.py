@@ -251,21 +251,25 @@ class _BNFn(torch.autograd.Function):
             float(m_total), momentum, BN_EPS)  # [2,C]: mean, invstd
         mean, invstd = mi[0], mi[1]
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
-        y = ext().bn_apply(x, mean, invstd, gamma.detach().float(),
-                           beta.detach().float(), res, act)
-        ctx.save_for_backward(x, y, mean, invstd, gamma)
+        y, mask = ext().bn_apply(x, mean, invstd, gamma.detach().float(),
+                                 beta.detach().float(), res, act)
+        # mask: 1-bit-per-element "y > 0" (fast path + relu) — backward
+        # reads it instead of re-reading y (1/16 the bytes)
+        if mask is None:
+            mask = torch.empty(0, device=x.device, dtype=torch.uint8)
+        ctx.save_for_backward(x, y, mean, invstd, gamma, mask)
         ctx.conf = (act, residual is not None, m_total, process_group)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, mean, invstd, gamma = ctx.saved_tensors
+        x, y, mean, invstd, gamma, mask = ctx.saved_tensors
         act, has_res, m_total, pg = ctx.conf
         dy = dy.contiguous()
         # ReLU mask fused into both backward kernels (no standalone pass)
         y_arg = y if act == _ACT_RELU else torch.empty(0, device=x.device,
                                                        dtype=x.dtype)
-        r = ext().bn_bwd_reduce(x, dy, y_arg, mean, invstd)  # [2,C]
+        r = ext().bn_bwd_reduce(x, dy, y_arg, mean, invstd, mask)  # [2,C]
         if pg is not None:
             import torch.distributed as dist
 
@@ -274,7 +278,7 @@ class _BNFn(torch.autograd.Function):
         want_dres = has_res and act == _ACT_RELU
         dx, dres = ext().bn_bwd_dx(x, dy, y_arg, mean, invstd,
                                    gamma.detach().float(), dgamma, dbeta,
-                                   float(m_total), want_dres)
+                                   float(m_total), want_dres, mask)
         if has_res and not want_dres:
             dres = dy  # no activation: residual grad is dy itself
         elif not has_res:
@@ -300,8 +304,9 @@ def batch_norm(x, gamma, beta, running_mean, running_var, training,
         mean = running_mean.float()
         invstd = torch.rsqrt(running_var.float() + BN_EPS)
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
-        return ext().bn_apply(x.contiguous(), mean, invstd, gamma.detach().float(),
-                              beta.detach().float(), res, a)
+        return ext().bn_apply(x.contiguous(), mean, invstd,
+                              gamma.detach().float(), beta.detach().float(),
+                              res, a)[0]
     xc = x.permute(0, 3, 1, 2)
     y = F.batch_norm(xc, running_mean, running_var, gamma, beta, training,
                      momentum, BN_EPS).permute(0, 2, 3, 1)
