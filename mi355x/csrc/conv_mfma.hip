@@ -52,7 +52,6 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
     const float* __restrict__ bias,  // [KO] or null
     const T16* __restrict__ zpage,   // >=256 zero elements (OOB gather target)
-    const T16* __restrict__ addin,   // optional epilogue accumulate input
     T16* __restrict__ out,         // [N*Ho*Wo, KO]
     float* __restrict__ stats_slab,  // null, or [gy][gx][2][BNT] partial
     const int N, const int Hi, const int Wi, const int CI, const int KO,  //  (sum,sumsq) of this block's output tile (conv->BN fusion)
@@ -227,8 +226,6 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
 #pragma unroll
       for (int tnt = 0; tnt < NT; ++tnt) {
         float v = acc[tnt][reg] + bv[tnt];
-        if (addin)
-          v += F16<T16>::to_f32(addin[m_out * KO + k0 + tnt * 32 + li]);
         if (act == 1) v = fmaxf(v, 0.f);
         if (stats_slab) {
           ssum[tnt] += v;
@@ -309,8 +306,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     const T16* __restrict__ in,   // [N, Hi, Wi, CI] (dgrad: dy, CI=KO)
     const T16* __restrict__ wgt,  // fwd: [KO, 9*CI]; dgrad: wflip strides
     const float* __restrict__ bias, T16* __restrict__ out,
-    const T16* __restrict__ addin,  // optional: out += addin (residual-
-    const int N, const int Hi, const int Wi, const int CI, const int KO,  // junction grad fused into the dgrad epilogue)
+    const int N, const int Hi, const int Wi, const int CI, const int KO,
     const long b_row_stride, const long b_rs_stride, const int act,
     const int has_bias, const int NR) {
   extern __shared__ __attribute__((aligned(16))) char psmem[];
@@ -435,8 +431,6 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
 #pragma unroll
       for (int t2 = 0; t2 < 2; ++t2) {
         float v = acc[t2][reg] + bv[t2];
-        if (addin)
-          v += F16<T16>::to_f32(addin[m_out * KO + k0 + t2 * 32 + li]);
         if (act == 1) v = fmaxf(v, 0.f);
         out[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
       }
@@ -1507,7 +1501,7 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
-                       (const T16*)zp.data_ptr(), nullptr,
+                       (const T16*)zp.data_ptr(),
                        (T16*)y.data_ptr(), nullptr, N, Hi, Wi, CI, KO, Ho,
                        Wo, (int)R, (int)S, (int)stride, (int)pad, KGP, 0,
                        (int)act, has_bias);
@@ -1578,7 +1572,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          smem, cur_stream(), (const T16*)x.data_ptr(),
                          (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (T16*)y.data_ptr(), nullptr, N, Hi, Wi, CI, KO,
+                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO,
                          (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
     });
     return;
@@ -1602,7 +1596,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(), nullptr,
+                         (const T16*)zp2.data_ptr(),
                          (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
                          KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
@@ -1612,7 +1606,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(), nullptr,
+                         (const T16*)zp2.data_ptr(),
                          (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
                          KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
@@ -1630,12 +1624,10 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
 // dgrad: in = dy[N,P,Q,KO], wflip = [R,S,CI,KO] (w[k,R-1-r,S-1-s,c]),
 // out = dx[N,H,W,CI]
 void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
-                            long R, long S, long stride, long pad,
-                            at::Tensor addin) {
+                            long R, long S, long stride, long pad) {
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
-  const bool has_add = addin.defined() && addin.numel() > 0;
   static const bool patch_on = [] {
     const char* e = getenv("MI355X_CONV_PATCH");
     return !e || e[0] != '0';
@@ -1649,9 +1641,7 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
       hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
                          smem, cur_stream(), (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
-                         (T16*)dx.data_ptr(),
-                         has_add ? (const T16*)addin.data_ptr() : nullptr,
-                         N, P, Q, KO, CI, (long)KO,
+                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, (long)KO,
                          (long)CI * KO, 0, 0, NR);
     });
     return;
@@ -1666,7 +1656,6 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
                          (const T16*)zp.data_ptr(),
-                         has_add ? (const T16*)addin.data_ptr() : nullptr,
                          (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
                          (int)R, (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);
@@ -1676,7 +1665,6 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                          (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
                          (const T16*)zp.data_ptr(),
-                         has_add ? (const T16*)addin.data_ptr() : nullptr,
                          (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
                          (int)R, (int)S, (int)stride, (int)pad, (long)KO,
                          (long)CI * KO, 0, 0);

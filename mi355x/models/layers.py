@@ -142,16 +142,6 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
     return bn(conv(x), residual=residual)
 
 
-def conv_bn_tap(conv: "Conv2d", bn: "BatchNorm2d", x):
-    """conv -> BN returning (out, tap). The block's shortcut consumes
-    `tap` instead of x directly, so the residual-junction gradient is
-    fused into this conv's dgrad epilogue (ops.functional._ConvTapFn)."""
-    from mi355x.ops import functional as F_
-
-    y, tap = F_.conv2d_tap(x, conv.weight, conv.stride, conv.padding)
-    return bn(y), tap
-
-
 class MaxPool2d(nn.Module):
     def __init__(self, kernel_size, stride=None, padding=0):
         super().__init__()

@@ -13,7 +13,7 @@ import torch
 from torch import nn
 
 from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, conv_bn,
-                     conv_bn_tap, to_model_layout)
+                     to_model_layout)
 from mi355x import ops
 
 
@@ -29,10 +29,8 @@ class BasicBlock(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        # the shortcut consumes conv1's TAP so the junction gradient fuses
-        # into conv1's dgrad (mi355x/models/layers.py conv_bn_tap)
-        out, tap = conv_bn_tap(self.conv1, self.bn1, x)
-        identity = tap if self.downsample is None else self.downsample(tap)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = conv_bn(self.conv1, self.bn1, x)
         out = conv_bn(self.conv2, self.bn2, out, residual=identity)
         return out
 
@@ -51,8 +49,8 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        out, tap = conv_bn_tap(self.conv1, self.bn1, x)
-        identity = tap if self.downsample is None else self.downsample(tap)
+        identity = x if self.downsample is None else self.downsample(x)
+        out = conv_bn(self.conv1, self.bn1, x)
         out = conv_bn(self.conv2, self.bn2, out)
         out = conv_bn(self.conv3, self.bn3, out, residual=identity)
         return out

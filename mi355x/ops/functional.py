@@ -145,7 +145,7 @@ class _ConvFn(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             wflip = _w16_conv_flip(ctx.weight_ref, dy.dtype)
             dx = ext().conv2d_dgrad(dy, wflip, stride, padding,
-                                    x.shape[1], x.shape[2], _EMPTY_BF16(x))
+                                    x.shape[1], x.shape[2])
         dw = None
         if ctx.needs_input_grad[1]:
             # wgrad kernels emit fp32 [K,C,R,S] directly (parameter layout)
@@ -155,56 +155,6 @@ class _ConvFn(torch.autograd.Function):
         if has_bias and ctx.needs_input_grad[2]:
             db = dy.float().sum(dim=(0, 1, 2))
         return dx, dw, db, None, None, None, None
-
-
-def _EMPTY_BF16(like):
-    return torch.empty(0, device=like.device, dtype=like.dtype)
-
-
-class _ConvTapFn(torch.autograd.Function):
-    """conv2d with a residual TAP: forward returns (y, tap) where tap is
-    the input passed through. A ResNet block feeds `tap` (not x) to its
-    shortcut/residual path, so the junction gradient arrives HERE as
-    d_tap and is fused into the dgrad epilogue (dx = dgrad(dy) + d_tap) —
-    autograd's separate full-tensor add at the junction disappears
-    (measured 8 adds/step on r18, 16 on r50 at 35-83 us each)."""
-
-    @staticmethod
-    def forward(ctx, x, weight, stride, padding):
-        w16 = _w16_conv(weight, x.dtype)
-        e = torch.empty(0, device=x.device)
-        y = ext().conv2d_fwd(x, w16, e, stride, padding, _ACT_NONE,
-                             weight.shape[2], weight.shape[3])
-        ctx.save_for_backward(x, w16)
-        ctx.weight_ref = weight
-        ctx.conf = (stride, padding, weight.shape)
-        return y, x.view_as(x)
-
-    @staticmethod
-    def backward(ctx, dy, dtap):
-        x, w16 = ctx.saved_tensors
-        stride, padding, wshape = ctx.conf
-        dy = dy.contiguous()
-        dx = None
-        if ctx.needs_input_grad[0]:
-            wflip = _w16_conv_flip(ctx.weight_ref, dy.dtype)
-            add = dtap.contiguous() if dtap is not None else _EMPTY_BF16(x)
-            dx = ext().conv2d_dgrad(dy, wflip, stride, padding,
-                                    x.shape[1], x.shape[2], add)
-        dw = None
-        if ctx.needs_input_grad[1]:
-            dw = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride,
-                                    padding)
-        return dx, dw, None, None
-
-
-def conv2d_tap(x, weight, stride=1, padding=0):
-    """conv2d returning (y, tap): route the block's shortcut through `tap`
-    so its gradient fuses into this conv's dgrad (GPU training only; on
-    CPU the plain double-use add semantics apply)."""
-    if x.is_cuda and x.requires_grad:
-        return _ConvTapFn.apply(x, weight, stride, padding)
-    return conv2d(x, weight, None, stride, padding, None), x
 
 
 def conv2d_with_stats(x, weight, stride=1, padding=0):
