@@ -340,16 +340,38 @@ __global__ __launch_bounds__(256) void conv_fwd_stem7_lds(
   const int n = blockIdx.x / Ho;
   const int ih0 = 2 * p - pad;
 
-  // ---- cooperative strip fill: strip col cs = iw + pad ----
+  // ---- cooperative strip fill. Strip element (iw+pad)*3+c maps to the
+  // CONTIGUOUS x row element iw*3+c, so the interior is short8 block
+  // copies (the elementwise /3-decode version was ~19 serial 2B loads
+  // per thread); pads and invalid rows are vector-zeroed.
   const T16 z{};
-  for (int i = threadIdx.x; i < ROWS * sstride; i += 256) {
-    const int rr = i / sstride;
-    const int e = i - rr * sstride;
-    const int cs = e / 3, c = e - cs * 3;
-    const int ih = ih0 + rr, iw = cs - pad;
-    const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
-                    (unsigned)iw < (unsigned)W;
-    xs[i] = ok ? x[(((long)n * H + ih) * W + iw) * 3 + c] : z;
+  const int row3 = 3 * W;             // contiguous payload per row
+  const int off3 = 3 * pad;           // strip elements before the payload
+  const int nv8 = (row3 + 7) / 8;
+  for (int t = threadIdx.x; t < ROWS * nv8; t += 256) {
+    const int rr = t / nv8;
+    const int j8 = (t - rr * nv8) * 8;
+    const int ih = ih0 + rr;
+    T16* dst = xs + (long)rr * sstride + off3 + j8;
+    if ((row3 & 7) == 0 && (unsigned)ih < (unsigned)H && j8 + 8 <= row3) {
+      *reinterpret_cast<short8*>(dst) = *reinterpret_cast<const short8*>(
+          x + ((long)n * H + ih) * row3 + j8);
+    } else {
+#pragma unroll
+      for (int u = 0; u < 8; ++u) {
+        const int e = j8 + u;
+        dst[u] = (e < row3 && (unsigned)ih < (unsigned)H)
+                     ? x[((long)n * H + ih) * row3 + e]
+                     : z;
+      }
+    }
+  }
+  // zero the pad columns (and the alignment tail) of every row
+  for (int t = threadIdx.x; t < ROWS * (sstride - row3); t += 256) {
+    const int nz = sstride - row3;
+    const int rr = t / nz;
+    const int e = t - rr * nz;
+    xs[(long)rr * sstride + (e < off3 ? e : row3 + e)] = z;
   }
 
   const int k = threadIdx.x & 63;
@@ -408,15 +430,37 @@ __global__ __launch_bounds__(256) void conv_fwd_stem3_lds(
 
   const int p = blockIdx.x % Ho;
   const int n = blockIdx.x / Ho;
-  for (int i = threadIdx.x; i < 3 * sstride; i += 256) {
-    const int rr = i / sstride;
-    const int e = i - rr * sstride;
-    const int cs = e / 3, c = e - cs * 3;
-    const int ih = p - 1 + rr, iw = cs - 1;
-    const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
-                    (unsigned)iw < (unsigned)W;
-    xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
-               : 0.f;
+  {
+    // vectorized fill (see the 7x7 stem comment: strip interior is one
+    // contiguous x row)
+    const int row3 = 3 * W, off3 = 3;
+    const int nv8 = (row3 + 7) / 8;
+    for (int t = threadIdx.x; t < 3 * nv8; t += 256) {
+      const int rr = t / nv8;
+      const int j8 = (t - rr * nv8) * 8;
+      const int ih = p - 1 + rr;
+      float* dst = xs + (long)rr * sstride + off3 + j8;
+      if ((row3 & 7) == 0 && (unsigned)ih < (unsigned)H && j8 + 8 <= row3) {
+        const short8 v = *reinterpret_cast<const short8*>(
+            x + ((long)n * H + ih) * row3 + j8);
+#pragma unroll
+        for (int u = 0; u < 8; ++u) dst[u] = s16_to_f32<T16>(v[u]);
+      } else {
+#pragma unroll
+        for (int u = 0; u < 8; ++u) {
+          const int e = j8 + u;
+          dst[u] = (e < row3 && (unsigned)ih < (unsigned)H)
+                       ? F16<T16>::to_f32(x[((long)n * H + ih) * row3 + e])
+                       : 0.f;
+        }
+      }
+    }
+    for (int t = threadIdx.x; t < 3 * (sstride - row3); t += 256) {
+      const int nz = sstride - row3;
+      const int rr = t / nz;
+      const int e = t - rr * nz;
+      xs[(long)rr * sstride + (e < off3 ? e : row3 + e)] = 0.f;
+    }
   }
 
   const int k = threadIdx.x & 63;
@@ -471,15 +515,36 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem3_lds(
     const int p = (int)(row % Ho);
     const int n = (int)(row / Ho);
     __syncthreads();
-    for (int i = threadIdx.x; i < 3 * sstride; i += 256) {
-      const int rr = i / sstride;
-      const int e = i - rr * sstride;
-      const int cs = e / 3, c = e - cs * 3;
-      const int ih = p - 1 + rr, iw = cs - 1;
-      const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
-                      (unsigned)iw < (unsigned)W;
-      xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
-                 : 0.f;
+    {
+      const int row3 = 3 * W, off3 = 3;
+      const int nv8 = (row3 + 7) / 8;
+      for (int t = threadIdx.x; t < 3 * nv8; t += 256) {
+        const int rr = t / nv8;
+        const int j8 = (t - rr * nv8) * 8;
+        const int ih = p - 1 + rr;
+        float* dst = xs + (long)rr * sstride + off3 + j8;
+        if ((row3 & 7) == 0 && (unsigned)ih < (unsigned)H && j8 + 8 <= row3) {
+          const short8 v = *reinterpret_cast<const short8*>(
+              x + ((long)n * H + ih) * row3 + j8);
+#pragma unroll
+          for (int u = 0; u < 8; ++u) dst[u] = s16_to_f32<T16>(v[u]);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) {
+            const int e = j8 + u;
+            dst[u] = (e < row3 && (unsigned)ih < (unsigned)H)
+                         ? F16<T16>::to_f32(
+                               x[((long)n * H + ih) * row3 + e])
+                         : 0.f;
+          }
+        }
+      }
+      for (int t = threadIdx.x; t < 3 * (sstride - row3); t += 256) {
+        const int nz = sstride - row3;
+        const int rr = t / nz;
+        const int e = t - rr * nz;
+        xs[(long)rr * sstride + (e < off3 ? e : row3 + e)] = 0.f;
+      }
     }
     const T16* dyrow = dy + row * Wo * K;
     for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
@@ -559,15 +624,38 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
     const int n = (int)(row / Ho);
     const int ih0 = 2 * p - pad;
     __syncthreads();  // previous row's reads done before overwrite
-    for (int i = threadIdx.x; i < ROWS * sstride; i += 256) {
-      const int rr = i / sstride;
-      const int e = i - rr * sstride;
-      const int cs = e / 3, c = e - cs * 3;
-      const int ih = ih0 + rr, iw = cs - pad;
-      const bool ok = e < selems && (unsigned)ih < (unsigned)H &&
-                      (unsigned)iw < (unsigned)W;
-      xs[i] = ok ? F16<T16>::to_f32(x[(((long)n * H + ih) * W + iw) * 3 + c])
-                 : 0.f;
+    {
+      // vectorized fill: strip interior is a contiguous x row (see the
+      // fwd stem kernel comment), short8-loaded and converted
+      const int row3 = 3 * W, off3 = 3 * pad;
+      const int nv8 = (row3 + 7) / 8;
+      for (int t = threadIdx.x; t < ROWS * nv8; t += 256) {
+        const int rr = t / nv8;
+        const int j8 = (t - rr * nv8) * 8;
+        const int ih = ih0 + rr;
+        float* dst = xs + (long)rr * sstride + off3 + j8;
+        if ((row3 & 7) == 0 && (unsigned)ih < (unsigned)H && j8 + 8 <= row3) {
+          const short8 v = *reinterpret_cast<const short8*>(
+              x + ((long)n * H + ih) * row3 + j8);
+#pragma unroll
+          for (int u = 0; u < 8; ++u) dst[u] = s16_to_f32<T16>(v[u]);
+        } else {
+#pragma unroll
+          for (int u = 0; u < 8; ++u) {
+            const int e = j8 + u;
+            dst[u] = (e < row3 && (unsigned)ih < (unsigned)H)
+                         ? F16<T16>::to_f32(
+                               x[((long)n * H + ih) * row3 + e])
+                         : 0.f;
+          }
+        }
+      }
+      for (int t = threadIdx.x; t < ROWS * (sstride - row3); t += 256) {
+        const int nz = sstride - row3;
+        const int rr = t / nz;
+        const int e = t - rr * nz;
+        xs[(long)rr * sstride + (e < off3 ? e : row3 + e)] = 0.f;
+      }
     }
     const T16* dyrow = dy + row * Wo * K;
     for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
