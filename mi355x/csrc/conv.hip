@@ -547,7 +547,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem3_lds(
       }
     }
     const T16* dyrow = dy + row * Wo * K;
-    for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    if (((Wo * K) & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < Wo * K; i += 256 * 8)
+        *reinterpret_cast<short8*>(dys + i) =
+            *reinterpret_cast<const short8*>(dyrow + i);
+    } else {
+      for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    }
     __syncthreads();
 
     for (int q = q0; q < q1; ++q) {
@@ -658,7 +664,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_stem7_lds(
       }
     }
     const T16* dyrow = dy + row * Wo * K;
-    for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    if (((Wo * K) & 7) == 0) {
+      for (int i = threadIdx.x * 8; i < Wo * K; i += 256 * 8)
+        *reinterpret_cast<short8*>(dys + i) =
+            *reinterpret_cast<const short8*>(dyrow + i);
+    } else {
+      for (int i = threadIdx.x; i < Wo * K; i += 256) dys[i] = dyrow[i];
+    }
     __syncthreads();
 
     for (int q = q0; q < q1; ++q) {
