@@ -34,8 +34,10 @@ def parse_args():
     ap.add_argument("--steps", type=int, default=30)
     ap.add_argument("--warmup", type=int, default=10)
     ap.add_argument("--batch", type=int,
-                    default=int(os.environ.get("MI355X_BENCH_BATCH", "1024")),
-                    help="per-GPU batch size")
+                    default=int(os.environ.get("MI355X_BENCH_BATCH", "4096")),
+                    help="per-GPU batch size (measured sweep: 94.5k img/s "
+                         "at b1024 -> 107.6k at b4096, peak 108.4k at "
+                         "b8192 on 1xMI355X; 288 GB HBM3E fits far more)")
     ap.add_argument("--model", default=os.environ.get("MI355X_BENCH_MODEL",
                                                       "resnet18"))
     ap.add_argument("--size", type=int,
