@@ -865,6 +865,24 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
     });
     return y;
   }
+  // CIFAR stem 3x3/s1/p1: LDS-staged variant (conv_fwd_stem3_lds)
+  if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&
+      P == H && Q == W) {
+    const long wrow = w.dim() == 2 ? w.size(1) : 27L;
+    const int selems = (Q + 2) * 3;
+    const int sstride = (selems + 4) & ~3;
+    const size_t smem = (size_t)3 * sstride * 4;
+    const int has_bias = bias.numel() > 0;
+    DISPATCH_16(x, T16, {
+      hipLaunchKernelGGL((conv_fwd_stem3_lds<T16>), dim3((unsigned)(N * P)),
+                         dim3(256), smem, cur_stream(),
+                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                         has_bias ? bias.data_ptr<float>() : nullptr,
+                         (T16*)y.data_ptr(), N, H, W, K, P, Q, wrow,
+                         (int)act, has_bias);
+    });
+    return y;
+  }
   const bool smallc_fwd = K <= 64 && R == 3 && S == 3 && (C == 3 || C == 6) &&
                           (stride == 1 || stride == 2);
   if (smallc_fwd) {
