@@ -225,8 +225,7 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
     sh[u] = beta[c0 + u] - mean[c0 + u] * sc[u];
   }
   const long stride = (long)gridDim.x * blockDim.x * 8;
-  for (long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; e < E;
-       e += stride) {
+  auto body = [&](long e) {
     const short8 vx = *reinterpret_cast<const short8*>(x + e);
     short8 vr = {};
     if (res) vr = *reinterpret_cast<const short8*>(res + e);
@@ -244,7 +243,15 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
     }
     *reinterpret_cast<short8*>(y + e) = o;
     if (msk) msk[e >> 3] = (unsigned char)mb;
+  };
+  // two batched sub-streams per iteration (same in-flight-bytes rationale
+  // as bn_reduce_fast; the channel group is stride-invariant)
+  long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  for (; e + stride < E; e += 2 * stride) {
+    body(e);
+    body(e + stride);
   }
+  for (; e < E; e += stride) body(e);
 }
 
 template <typename T16>
@@ -266,8 +273,7 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     b_[u] = dbeta[c0 + u] * inv_m;
   }
   const long stride = (long)gridDim.x * blockDim.x * 8;
-  for (long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8; e < E;
-       e += stride) {
+  auto body = [&](long e) {
     const short8 vx = *reinterpret_cast<const short8*>(x + e);
     const short8 vd = *reinterpret_cast<const short8*>(dy + e);
     const unsigned mb = msk ? msk[e >> 3] : 0xffu;
@@ -282,7 +288,13 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     }
     *reinterpret_cast<short8*>(dx + e) = odx;
     if (dres) *reinterpret_cast<short8*>(dres + e) = ods;
+  };
+  long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  for (; e + stride < E; e += 2 * stride) {
+    body(e);
+    body(e + stride);
   }
+  for (; e < E; e += stride) body(e);
 }
 
 // one tiny kernel replacing the Python stats glue (mean/var/invstd/
