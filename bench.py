@@ -37,7 +37,7 @@ def parse_args():
                     default=int(os.environ["MI355X_BENCH_BATCH"])
                     if "MI355X_BENCH_BATCH" in os.environ else None,
                     help="per-GPU batch size; default 8192 for 32-px "
-                         "models, 1024 at 224 px (measured sweep: 94.5k "
+                         "models, 2048 at 224 px (measured sweep: 94.5k "
                          "img/s at b1024 -> 107.6k at b4096 -> 111.5k at "
                          "b8192 eager on 1xMI355X)")
     ap.add_argument("--model", default=os.environ.get("MI355X_BENCH_MODEL",
@@ -63,7 +63,7 @@ def parse_args():
 def main():
     args = parse_args()
     if args.batch is None:
-        args.batch = 8192 if args.size < 224 else 1024
+        args.batch = 8192 if args.size < 224 else 2048
     use_cuda = torch.cuda.is_available()
     world = comm.env_world_size()
     distributed = world > 1
