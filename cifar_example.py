@@ -12,6 +12,8 @@ Env overrides (benchmark configs; CLI shape unchanged):
   MI355X_SYNTHETIC=1                        synthetic 32x32 data (no download)
   MI355X_BATCH / MI355X_EPOCHS / MI355X_STEPS / MI355X_LR
   MI355X_DEVICE=cpu|cuda                    (default cpu, like the reference)
+  MI355X_CKPT=path                          checkpoint path (default the
+                                            reference's ./cifar_net.pth)
 """
 
 import os
@@ -81,7 +83,7 @@ def main():
             break
     print(f"Finished Training ({steps} steps, {time.time() - t0:.1f}s)")
 
-    PATH = "./cifar_net.pth"
+    PATH = os.environ.get("MI355X_CKPT", "./cifar_net.pth")
     torch.save(net.state_dict(), PATH)
 
     correct = 0

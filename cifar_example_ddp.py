@@ -16,7 +16,7 @@ Launch:  python -m mi355x.launcher --nproc-per-node 8 cifar_example_ddp.py
 
 Env overrides (benchmark configs; CLI shape unchanged): MI355X_MODEL,
 MI355X_SYNTHETIC, MI355X_BATCH, MI355X_EPOCHS, MI355X_STEPS, MI355X_LR,
-MI355X_SYNC_BN=1.
+MI355X_SYNC_BN=1, MI355X_CKPT (checkpoint path, default ./cifar_net.pth).
 """
 
 import argparse
@@ -112,7 +112,7 @@ def main(args):
     if args.rank == 0:
         print(f"Finished Training ({steps} steps, {time.time() - t0:.1f}s)")
 
-    PATH = "./cifar_net.pth"
+    PATH = os.environ.get("MI355X_CKPT", "./cifar_net.pth")
     if args.rank == 0:  # rank-gated (fixes reference quirk 7); keys keep
         torch.save(net.state_dict(), PATH)  # the 'module.' prefix
 

@@ -27,7 +27,7 @@ at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
                        double momentum, double eps);
 std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invstd,
                     at::Tensor gamma, at::Tensor beta, at::Tensor res,
-                    long act);
+                    long act, bool want_mask);
 at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
                          at::Tensor mean, at::Tensor invstd,
                          at::Tensor mask);

@@ -397,12 +397,14 @@ at::Tensor bn_stats(at::Tensor x) {
 }
 
 // returns {y, mask}: mask is a 1-bit-per-element "y > 0" map (u8, one
-// byte per 8 channel-elements) when act==relu on the fast path — the
-// backward reads it instead of re-reading y (1/16 the bytes). Empty
-// tensor otherwise (the fallback kernels keep the y-based masking).
+// byte per 8 channel-elements) when act==relu AND want_mask (training)
+// on the fast path — the backward reads it instead of re-reading y (1/16
+// the bytes). Empty tensor otherwise (eval callers skip the allocation
+// and mask stores entirely; the fallback kernels keep y-based masking).
 std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean,
                                  at::Tensor invstd, at::Tensor gamma,
-                                 at::Tensor beta, at::Tensor res, long act) {
+                                 at::Tensor beta, at::Tensor res, long act,
+                                 bool want_mask) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
@@ -412,7 +414,7 @@ std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean,
   const int has_res = res.numel() > 0;
   if (bn_fast_ok(M, C)) {
     const long E = M * (long)C;
-    at::Tensor mask = act == 1
+    at::Tensor mask = (act == 1 && want_mask)
                           ? at::empty({E / 8}, x.options().dtype(at::kByte))
                           : at::Tensor();
     const int grid = (int)std::min<long>(cdiv_l(E, 256 * 8), 2048);

@@ -1,7 +1,7 @@
 // Linear layer GEMMs. The classifier GEMMs in this
-// workload are small ([B,512]x[512,10] for ResNet-18/CIFAR); v0 uses
-// simple coalesced direct kernels. The MFMA tile path (gemm_mfma.hip)
-// takes over for large shapes.
+// workload are small ([B,512]x[512,10] for ResNet-18/CIFAR); these are
+// simple coalesced direct kernels, kept as the fallback for shapes the
+// MFMA tile path (gemm_mfma.hip) does not cover.
 #include "common.h"
 
 namespace {

@@ -217,6 +217,9 @@ std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
+  // relative argmax is one byte (r*kw+s) — larger windows would alias
+  TORCH_CHECK(kernel * kernel <= 256, "maxpool window ", kernel, "x", kernel,
+              " exceeds the byte-packed argmax range (kernel^2 must be <=256)");
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   const int P = (H + 2 * pad - kernel) / stride + 1;
   const int Q = (W + 2 * pad - kernel) / stride + 1;
@@ -245,6 +248,8 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
                        long kernel, long stride, long pad) {
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
+  TORCH_CHECK(kernel * kernel <= 256, "maxpool window ", kernel, "x", kernel,
+              " exceeds the byte-packed argmax range (kernel^2 must be <=256)");
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
   DISPATCH_16(dy, T16, {

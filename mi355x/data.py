@@ -282,15 +282,23 @@ class DataLoader:
             slot_ev[s] = ev
             return dx, dy, ev
 
+        def handover(p):
+            dx, dy, ev = p
+            cur = torch.cuda.current_stream(self.device)
+            cur.wait_event(ev)
+            # dx/dy were allocated on the copy stream; tell the caching
+            # allocator they are consumed on the compute stream, else a
+            # freed batch could be reused (and overwritten by a later H2D
+            # copy) while compute still reads it.
+            dx.record_stream(cur)
+            dy.record_stream(cur)
+            return dx, dy
+
         pending = None
         for k, b in enumerate(batches):
             nxt = stage(b, k & 1)
             if pending is not None:
-                dx, dy, ev = pending
-                torch.cuda.current_stream(self.device).wait_event(ev)
-                yield dx, dy
+                yield handover(pending)
             pending = nxt
         if pending is not None:
-            dx, dy, ev = pending
-            torch.cuda.current_stream(self.device).wait_event(ev)
-            yield dx, dy
+            yield handover(pending)

@@ -98,11 +98,19 @@ class BatchNorm2d(nn.Module):
         destination[prefix + "num_batches_tracked"] = torch.tensor(
             self._nbt, dtype=torch.long)
 
-    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+    def _load_from_state_dict(self, state_dict, prefix, local_metadata,
+                              strict, missing_keys, unexpected_keys,
+                              error_msgs):
+        # read (never pop — the caller's dict must stay loadable twice) and
+        # clear the hook's unexpected-key report for the host-side counter
         key = prefix + "num_batches_tracked"
         if key in state_dict:
-            self._nbt = int(state_dict.pop(key))
-        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
+            self._nbt = int(state_dict[key])
+        super()._load_from_state_dict(state_dict, prefix, local_metadata,
+                                      strict, missing_keys, unexpected_keys,
+                                      error_msgs)
+        if key in unexpected_keys:
+            unexpected_keys.remove(key)
 
     def forward(self, x, residual=None):
         if self.training:

@@ -252,7 +252,7 @@ class _BNFn(torch.autograd.Function):
         mean, invstd = mi[0], mi[1]
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
         y, mask = ext().bn_apply(x, mean, invstd, gamma.detach().float(),
-                                 beta.detach().float(), res, act)
+                                 beta.detach().float(), res, act, True)
         # mask: 1-bit-per-element "y > 0" (fast path + relu) — backward
         # reads it instead of re-reading y (1/16 the bytes)
         if mask is None:
@@ -304,9 +304,10 @@ def batch_norm(x, gamma, beta, running_mean, running_var, training,
         mean = running_mean.float()
         invstd = torch.rsqrt(running_var.float() + BN_EPS)
         res = residual if residual is not None else torch.empty(0, device=x.device, dtype=x.dtype)
+        # eval: no backward, skip the relu-mask allocation + stores
         return ext().bn_apply(x.contiguous(), mean, invstd,
                               gamma.detach().float(), beta.detach().float(),
-                              res, a)[0]
+                              res, a, False)[0]
     xc = x.permute(0, 3, 1, 2)
     y = F.batch_norm(xc, running_mean, running_var, gamma, beta, training,
                      momentum, BN_EPS).permute(0, 2, 3, 1)
