@@ -47,7 +47,11 @@ def init_process_group(backend: str | None = None, timeout_s: int = 300):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
     os.environ.setdefault("MASTER_PORT", "29500")
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        # MI355X_BACKEND=gloo lets multi-rank rehearsals share one GPU
+        # (RCCL refuses two ranks on the same device)
+        backend = os.environ.get(
+            "MI355X_BACKEND",
+            "nccl" if torch.cuda.is_available() else "gloo")
     if backend == "nccl":
         torch.cuda.set_device(local % max(torch.cuda.device_count(), 1))
     if not dist.is_initialized():

@@ -27,6 +27,7 @@ tests run the identical reducer logic).
 
 from __future__ import annotations
 
+import contextlib
 import os
 
 import torch
@@ -76,8 +77,13 @@ class DistributedDataParallel(nn.Module):
     def __init__(self, module: nn.Module, flat: FlatState | None = None,
                  bucket_mb: float | None = None,
                  first_bucket_mb: float = 1.0,
-                 process_group=None, comm: str | None = None):
+                 process_group=None, comm: str | None = None,
+                 find_unused_parameters: bool | None = None):
         super().__init__()
+        if find_unused_parameters is None:
+            find_unused_parameters = (
+                os.environ.get("MI355X_FIND_UNUSED", "0") == "1")
+        self.find_unused_parameters = find_unused_parameters
         if bucket_mb is None:
             bucket_mb = float(os.environ.get("MI355X_BUCKET_MB", "25"))
         self.module = module
@@ -93,21 +99,28 @@ class DistributedDataParallel(nn.Module):
         use_native = (comm == "rccl"
                       and self.flat.flat_param.is_cuda
                       and self.world_size > 1)
+        if use_native and not self._native_feasible_everywhere():
+            # the fallback decision must be COLLECTIVE: one rank silently
+            # dropping to torch.distributed while the others construct the
+            # native communicator would deadlock in ncclCommInitRank
+            import warnings
+            warnings.warn("native RCCL comm unavailable on at least one "
+                          "rank; all ranks falling back to torch.distributed")
+            use_native = False
         if use_native:
-            try:
-                self.comm = _NativeComm()
-            except Exception as e:  # construction-time failure only; fall
-                # back to torch.distributed (also RCCL underneath on ROCm)
-                import warnings
-                warnings.warn(f"native RCCL comm init failed ({e}); "
-                              "falling back to torch.distributed")
-                self.comm = _TorchComm(process_group)
+            # past the agreement point a construction failure raises loudly
+            # (a crash torchrun/our launcher tears down beats a silent
+            # divergent fallback that hangs the job)
+            self.comm = _NativeComm()
         else:
             self.comm = _TorchComm(process_group)
         self._works: list = []
         self._ready: dict[int, int] = {}
+        self._ready_ids: set[int] = set()
         self._launched = 0
         self._hooks = []
+        self._sync_enabled = True
+        self._names = {id(p): n for n, p in module.named_parameters()}
         if self.world_size > 1:
             self._broadcast_initial_state()
             for p in self.flat.params:
@@ -115,19 +128,58 @@ class DistributedDataParallel(nn.Module):
                     p.register_post_accumulate_grad_hook(self._mark_ready))
         self._reset_bucket_state()
 
+    def _native_feasible_everywhere(self) -> bool:
+        """All-rank MIN-reduce of local native-comm feasibility (extension
+        importable + RCCL symbols present) over the already-initialized
+        torch.distributed group, so every rank takes the same comm path."""
+        ok = 1
+        try:
+            from mi355x.ops import ext
+            e = ext()
+            ok = int(hasattr(e, "RcclComm") and hasattr(e, "rccl_get_unique_id"))
+        except Exception:
+            ok = 0
+        backend = dist.get_backend(self.process_group)
+        flag = torch.tensor(
+            [ok], dtype=torch.int32,
+            device=self.flat.flat_param.device if backend == "nccl" else "cpu")
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=self.process_group)
+        return bool(int(flag.item()))
+
     # -- init-time sync ----------------------------------------------------
     def _broadcast_initial_state(self):
         self.comm.broadcast(self.flat.flat_param, 0)
+        # buffers (BN running stats): batched into one flat broadcast per
+        # (device, dtype) group instead of one tiny collective per buffer
+        # (ResNet-50 has 106 of them)
+        groups: dict[tuple, list[torch.Tensor]] = {}
         for buf in self.module.buffers():
-            self.comm.broadcast(buf, 0)
+            if buf.numel() == 0:
+                continue
+            groups.setdefault((buf.device, buf.dtype), []).append(buf)
+        for bufs in groups.values():
+            if len(bufs) == 1:
+                self.comm.broadcast(bufs[0], 0)
+                continue
+            flatb = torch.cat([b.detach().reshape(-1) for b in bufs])
+            self.comm.broadcast(flatb, 0)
+            off = 0
+            for b in bufs:
+                n = b.numel()
+                b.detach().copy_(flatb[off:off + n].view_as(b))
+                off += n
 
     # -- per-iteration machinery -------------------------------------------
     def _reset_bucket_state(self):
         self._ready = {b.index: 0 for b in self.flat.buckets}
+        self._ready_ids = set()
         self._launched = 0
         self._works = []
 
     def _mark_ready(self, param: torch.Tensor):
+        if not self._sync_enabled:
+            return
+        self._ready_ids.add(id(param))
         b = self.flat.bucket_of[id(param)]
         self._ready[b.index] += 1
         # launch complete buckets in fixed index order (cross-rank safety)
@@ -142,17 +194,49 @@ class DistributedDataParallel(nn.Module):
             self._launched += 1
 
     def forward(self, *args, **kwargs):
-        if self.world_size > 1:
+        if self.world_size > 1 and self._sync_enabled:
             self._reset_bucket_state()
         return self.module(*args, **kwargs)
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Skip gradient all-reduce inside this context (gradient
+        accumulation, torch-DDP semantics): grads accumulate locally into
+        the flat buffer; the first backward outside the context reduces
+        the accumulated sums."""
+        old = self._sync_enabled
+        self._sync_enabled = False
+        try:
+            yield
+        finally:
+            self._sync_enabled = old
 
     def finish_grad_sync(self):
         """Wait for all in-flight bucket all-reduces (call before the
         optimizer step). Gradients are left as SUMS over ranks; consume
         self.grad_scale in the optimizer (or scale explicitly)."""
-        if self.world_size > 1:
-            assert self._launched == len(self.flat.buckets), \
-                "backward did not produce grads for every bucket"
+        if self.world_size > 1 and self._sync_enabled:
+            if self._launched < len(self.flat.buckets):
+                if not self.find_unused_parameters:
+                    missing = [self._names.get(id(p), "<unnamed>")
+                               for b in self.flat.buckets
+                               for p in b.params
+                               if id(p) not in self._ready_ids]
+                    raise RuntimeError(
+                        "backward produced no gradient for parameter(s) "
+                        f"{missing}; pass find_unused_parameters=True (or "
+                        "MI355X_FIND_UNUSED=1) if the model has frozen or "
+                        "conditionally-used parameters")
+                # flush: launch the remaining buckets in the same fixed
+                # index order every rank uses. Params that got no grad
+                # contribute their current flat-grad content (zeros after
+                # zero_grad) — summing zeros matches torch DDP's
+                # find_unused_parameters behavior.
+                for b in self.flat.buckets[self._launched:]:
+                    w = self.comm.all_reduce_async(self.flat.grad_slice(b))
+                    if w is not None:
+                        self._works.append(w)
+                self._launched = len(self.flat.buckets)
             self.comm.finish(self._works)
         self._works = []
 

@@ -16,13 +16,29 @@ import socket
 import time
 
 RCCL_PORT_OFFSET = 17
+UID_BYTES = 128  # sizeof(ncclUniqueId), asserted against the extension
 
 
-def _serve_id(addr: str, port: int, payload: bytes, world: int):
+def _bind_server(addr: str, port: int, world: int) -> socket.socket:
+    """Bind the rendezvous listen socket SYNCHRONOUSLY on rank 0 so a port
+    conflict raises immediately there (a bind failure inside a background
+    thread would leave every other rank timing out with no cause visible)."""
     srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
     srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
-    srv.bind((addr, port))
+    try:
+        srv.bind((addr, port))
+    except OSError as e:
+        srv.close()
+        raise OSError(
+            f"RCCL rendezvous: rank 0 could not bind {addr}:{port} "
+            f"(MASTER_PORT+{RCCL_PORT_OFFSET}): {e}. Another process is "
+            "holding the port — change MASTER_PORT or kill the holder."
+        ) from e
     srv.listen(world)
+    return srv
+
+
+def _serve_id(srv: socket.socket, payload: bytes, world: int):
     served = 0
     while served < world - 1:
         conn, _ = srv.accept()
@@ -32,24 +48,30 @@ def _serve_id(addr: str, port: int, payload: bytes, world: int):
     srv.close()
 
 
-def _fetch_id(addr: str, port: int, timeout_s: float = 120.0) -> bytes:
+def _fetch_id(addr: str, port: int, timeout_s: float | None = None) -> bytes:
+    if timeout_s is None:
+        timeout_s = float(os.environ.get("MI355X_RDZV_TIMEOUT", "120"))
     deadline = time.time() + timeout_s
     while True:
         try:
             s = socket.create_connection((addr, port), timeout=5)
             buf = b""
-            while len(buf) < 128:
-                chunk = s.recv(128 - len(buf))
+            while len(buf) < UID_BYTES:
+                chunk = s.recv(UID_BYTES - len(buf))
                 if not chunk:
                     break
                 buf += chunk
             s.close()
-            if len(buf) == 128:
+            if len(buf) == UID_BYTES:
                 return buf
         except OSError:
             pass
         if time.time() > deadline:
-            raise TimeoutError("RCCL rendezvous: could not fetch ncclUniqueId")
+            raise TimeoutError(
+                f"RCCL rendezvous: could not fetch ncclUniqueId from "
+                f"{addr}:{port} within {timeout_s:.0f}s (override with "
+                "MI355X_RDZV_TIMEOUT). Rank 0 is down or unreachable — "
+                "check its log for a bind/init error.")
         time.sleep(0.2)
 
 
@@ -69,9 +91,11 @@ class NativeComm:
         if world > 1:
             if rank == 0:
                 uid = ext().rccl_get_unique_id()
+                assert len(uid) == UID_BYTES, \
+                    f"ncclUniqueId is {len(uid)} bytes, expected {UID_BYTES}"
+                srv = _bind_server(master_addr, port, world)
                 t = threading.Thread(target=_serve_id,
-                                     args=(master_addr, port, uid, world),
-                                     daemon=True)
+                                     args=(srv, uid, world), daemon=True)
                 t.start()
             else:
                 uid = _fetch_id(master_addr, port)

@@ -26,6 +26,8 @@ def test_cifar_example_serial_cpu(tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     assert "Finished Training" in r.stdout
     assert "Accuracy of the network" in r.stdout
+    assert (tmp_path / "net.pth").exists()  # MI355X_CKPT honored
+    assert not os.path.exists(os.path.join(REPO, "cifar_net.pth"))
 
 
 def test_cifar_example_ddp_2proc_cpu(tmp_path):
@@ -40,3 +42,4 @@ def test_cifar_example_ddp_2proc_cpu(tmp_path):
     }, timeout=600)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "Accuracy on the test set" in r.stdout
+    assert (tmp_path / "net_ddp.pth").exists()  # MI355X_CKPT honored
