@@ -94,9 +94,32 @@ def main():
     else:
         flat = FlatState(net)
         gscale = 1.0
-    # fp16 (BASELINE config 5): static loss scaling, graph-safe; the
-    # 1/scale is folded into the fused SGD grad_scale
-    loss_scale = 256.0 if args.fp16 else 1.0
+
+    # graph decision up front: dynamic loss scaling takes a host-side
+    # branch per step, so the graphed path keeps a static scale instead
+    want_graph = (str(args.graph) == "1"
+                  or (str(args.graph) == "auto" and args.batch < 2048))
+    if distributed and os.environ.get("MI355X_GRAPH_DIST", "0") != "1":
+        # world>1 graph capture (RCCL collectives inside hipGraph) is
+        # untested on this pool — off unless explicitly requested
+        want_graph = False
+    use_graph = want_graph and use_cuda
+
+    # fp16 (BASELINE config 5): DYNAMIC loss scaling via amp.GradScaler
+    # (halve on overflow, grow on clean streaks), inf/nan scan over the
+    # flat fp32 grad buffer post-all-reduce; 1/scale folds into the fused
+    # SGD grad_scale. Graphed runs fall back to a static 256.0 scale.
+    scaler = None
+    loss_scale = 1.0
+    if args.fp16:
+        if use_graph:
+            loss_scale = 256.0
+        else:
+            from mi355x import amp
+            scaler = amp.GradScaler(
+                init_scale=float(os.environ.get("MI355X_INIT_SCALE", 2.0**14)),
+                growth_interval=int(os.environ.get("MI355X_GROWTH_INTERVAL",
+                                                   "200")))
     optimizer = optim.SGD(flat, lr=0.1, momentum=0.9,
                           grad_scale=gscale / loss_scale)
 
@@ -112,20 +135,21 @@ def main():
         optimizer.zero_grad()
         out = net(x)
         loss = cross_entropy(out, y)
+        if scaler is not None:
+            used = scaler.scale_value  # capture: step_ok may grow it
+            (loss * used).backward()
+            if distributed:
+                net.finish_grad_sync()
+            if scaler.step_ok(flat.flat_grad):
+                optimizer.grad_scale = gscale / used
+                optimizer.step()
+            return loss
         (loss * loss_scale).backward() if loss_scale != 1.0 else loss.backward()
         if distributed:
             net.finish_grad_sync()
         optimizer.step()
         return loss
 
-    # world>1 graph capture (RCCL collectives inside hipGraph) is untested
-    # on this pool — default it off for the multi-GPU scaling run unless
-    # explicitly requested (MI355X_GRAPH_DIST=1)
-    want_graph = (str(args.graph) == "1"
-                  or (str(args.graph) == "auto" and args.batch < 2048))
-    if distributed and os.environ.get("MI355X_GRAPH_DIST", "0") != "1":
-        want_graph = False
-    use_graph = want_graph and use_cuda
     if use_graph:
       try:
         # capture one full train step (launch-bound at CIFAR sizes: ~200
@@ -208,6 +232,9 @@ def main():
                 "parallelism": f"dp{world}",
                 "sync_bn": bool(args.sync_bn),
                 "hip_graph": bool(use_graph),
+                "loss_scaling": (None if not args.fp16 else
+                                 "static256" if scaler is None else
+                                 f"dynamic(final={scaler.scale_value:g})"),
                 "loss_final": round(float(loss.item()), 4),
             },
         }))

@@ -16,7 +16,8 @@ Launch:  python -m mi355x.launcher --nproc-per-node 8 cifar_example_ddp.py
 
 Env overrides (benchmark configs; CLI shape unchanged): MI355X_MODEL,
 MI355X_SYNTHETIC, MI355X_BATCH, MI355X_EPOCHS, MI355X_STEPS, MI355X_LR,
-MI355X_SYNC_BN=1, MI355X_CKPT (checkpoint path, default ./cifar_net.pth).
+MI355X_SYNC_BN=1, MI355X_FP16=1 (fp16 compute + dynamic loss scaling),
+MI355X_CKPT (checkpoint path, default ./cifar_net.pth).
 """
 
 import argparse
@@ -75,6 +76,13 @@ def main(args):
     if os.environ.get("MI355X_SYNC_BN", "0") == "1":
         sync_bn.enable(net)
     net = DistributedDataParallel(net)
+    # BASELINE config 5: fp16 compute + dynamic loss scaling (overflow
+    # skips the step and halves the scale; clean streaks grow it)
+    scaler = None
+    if os.environ.get("MI355X_FP16", "0") == "1":
+        from mi355x import amp
+        amp.set_compute_dtype(torch.float16)
+        scaler = amp.GradScaler()
     optimizer = optim.SGD(net.flat, lr=lr, momentum=0.9,
                           grad_scale=net.grad_scale)
 
@@ -93,9 +101,17 @@ def main(args):
             optimizer.zero_grad()
             outputs = net(inputs)
             loss = cross_entropy(outputs, labels)
-            loss.backward()
-            net.finish_grad_sync()
-            optimizer.step()
+            if scaler is not None:
+                used = scaler.scale_value
+                (loss * used).backward()
+                net.finish_grad_sync()
+                if scaler.step_ok(net.flat.flat_grad):
+                    optimizer.grad_scale = net.grad_scale / used
+                    optimizer.step()
+            else:
+                loss.backward()
+                net.finish_grad_sync()
+                optimizer.step()
 
             if meter is not None:
                 meter.step(inputs.shape[0])

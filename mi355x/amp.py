@@ -65,10 +65,13 @@ class GradScaler:
         return 1.0 / self.scale_value
 
     def step_ok(self, flat_grad: torch.Tensor) -> bool:
-        """True if flat_grad is finite; updates the dynamic scale."""
-        finite = bool(torch.isfinite(flat_grad.sum()).item()) and bool(
-            torch.isfinite(flat_grad).all().item()
-        )
+        """True if flat_grad is finite; updates the dynamic scale.
+
+        Callers must capture the scale BEFORE calling (this may grow it):
+            used = scaler.scale_value; (loss*used).backward(); ...
+            if scaler.step_ok(flat): opt.grad_scale = base/used; opt.step()
+        """
+        finite = bool(torch.isfinite(flat_grad).all().item())
         if finite:
             self._clean_steps += 1
             if self._clean_steps >= self.growth_interval:

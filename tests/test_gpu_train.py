@@ -88,6 +88,44 @@ def test_resnet18_fp16_trains_on_gpu():
     assert losses[-1] < losses[0] * 1.05, losses
 
 
+def test_fp16_dynamic_scaling_overflow_recovery():
+    """BASELINE config 5 on hardware: start with an absurd loss scale so the
+    fp32 flat grads overflow; the scaler must skip those steps (params
+    untouched), halve down to a workable scale, then train normally."""
+    from mi355x import amp
+
+    dev = torch.device("cuda")
+    torch.manual_seed(0)
+    with amp.autocast(torch.float16):
+        net = build_model("resnet18").to(dev)
+        flat = FlatState(net)
+        opt = optim.SGD(flat, lr=0.05, momentum=0.9)
+        scaler = amp.GradScaler(init_scale=2.0**40, growth_interval=10_000)
+        g = torch.Generator().manual_seed(0)
+        x = torch.randn(16, 3, 32, 32, generator=g).to(dev)
+        y = torch.randint(0, 10, (16,), generator=g).to(dev)
+        p0 = flat.flat_param.clone()
+        skips = 0
+        losses = []
+        for _ in range(30):
+            opt.zero_grad()
+            used = scaler.scale_value
+            loss = cross_entropy(net(x), y)
+            (loss * used).backward()
+            if scaler.step_ok(flat.flat_grad):
+                opt.grad_scale = 1.0 / used
+                opt.step()
+                losses.append(loss.item())
+            else:
+                skips += 1
+                if not losses:  # nothing applied yet: params must be frozen
+                    torch.testing.assert_close(flat.flat_param, p0,
+                                               rtol=0, atol=0)
+    assert skips >= 1, "2^40 scale should overflow fp32 grads at least once"
+    assert scaler.scale_value < 2.0**40
+    assert len(losses) >= 5 and losses[-1] < losses[0], (skips, losses)
+
+
 def test_device_dataloader_prefetch_roundtrip():
     """The prefetching H2D loader (mi355x/data.py device= path) delivers
     exactly the dataset's tensors, in sampler order, on the GPU."""
