@@ -32,7 +32,11 @@ def get_datasets():
     if os.environ.get("MI355X_SYNTHETIC", "0") == "1":
         return (SyntheticImageDataset(50000, seed=1),
                 SyntheticImageDataset(10000, seed=2))
-    return CIFAR10("./data", train=True), CIFAR10("./data", train=False)
+    # download=True bootstraps a clean box, like the reference's
+    # torchvision.datasets.CIFAR10(root='./data', download=True)
+    # (/root/reference/cifar_example.py:40-44)
+    return (CIFAR10("./data", train=True, download=True),
+            CIFAR10("./data", train=False, download=True))
 
 
 def main():
@@ -43,9 +47,13 @@ def main():
     max_steps = int(os.environ.get("MI355X_STEPS", "0"))  # 0 = full epochs
 
     trainset, testset = get_datasets()
+    # num_workers=2 mirrors the reference surface (cifar_example.py:47,52);
+    # our loader needs no worker processes (see mi355x/data.py)
     trainloader = DataLoader(trainset, batch_size=batch, shuffle=True,
+                             num_workers=2,
                              device=device if device.type == "cuda" else None)
     testloader = DataLoader(testset, batch_size=batch, shuffle=False,
+                            num_workers=2,
                             device=device if device.type == "cuda" else None)
 
     net = build_model(os.environ.get("MI355X_MODEL", "net")).to(device)

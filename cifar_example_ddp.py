@@ -44,16 +44,23 @@ def init_distributed(args):
     comm.barrier()
 
 
-def get_datasets():
+def get_datasets(args):
     if os.environ.get("MI355X_SYNTHETIC", "0") == "1":
         return (SyntheticImageDataset(50000, seed=1),
                 SyntheticImageDataset(10000, seed=2))
+    # rank-0-gated download + barrier — a conscious fix of reference quirk
+    # 11, where every rank races on the download because dataset setup
+    # precedes init_distributed (/root/reference/cifar_example_ddp.py:67-69)
+    if args.rank == 0:
+        from mi355x.data import download_cifar10
+        download_cifar10("./data")
+    comm.barrier()
     return CIFAR10("./data", train=True), CIFAR10("./data", train=False)
 
 
 def main(args):
-    trainset, testset = get_datasets()
     init_distributed(args)
+    trainset, testset = get_datasets(args)
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", args.gpu) if use_cuda else torch.device("cpu")
 

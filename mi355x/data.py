@@ -37,6 +37,7 @@ import torch
 
 __all__ = [
     "CIFAR10",
+    "download_cifar10",
     "SyntheticImageDataset",
     "DistributedSampler",
     "DataLoader",
@@ -51,10 +52,66 @@ _CIFAR_DIR = "cifar-10-batches-py"
 _TRAIN_BATCHES = [f"data_batch_{i}" for i in range(1, 6)]
 _TEST_BATCHES = ["test_batch"]
 
+CIFAR10_URL = "https://www.cs.toronto.edu/~kriz/cifar-10-python.tar.gz"
+CIFAR10_MD5 = "c58f30108f718f92721af3b95e74349a"
+
 CIFAR10_CLASSES = (
     "plane", "car", "bird", "cat", "deer",
     "dog", "frog", "horse", "ship", "truck",
 )
+
+
+def _md5(path: str) -> str:
+    import hashlib
+    h = hashlib.md5()
+    with open(path, "rb") as f:
+        for chunk in iter(lambda: f.read(1 << 20), b""):
+            h.update(chunk)
+    return h.hexdigest()
+
+
+def _batches_present(base: str) -> bool:
+    return all(os.path.exists(os.path.join(base, n))
+               for n in _TRAIN_BATCHES + _TEST_BATCHES)
+
+
+def download_cifar10(root: str, url: str = CIFAR10_URL,
+                     md5: str = CIFAR10_MD5) -> None:
+    """Fetch + verify + extract CIFAR-10 into the torchvision on-disk layout
+    (``<root>/cifar-10-batches-py/``) — the bootstrap the reference gets
+    from torchvision.datasets.CIFAR10(download=True)
+    (/root/reference/cifar_example.py:40-44). No-op if the batches are
+    already in place. Callers in distributed jobs must gate this on rank 0
+    and barrier (consciously fixing reference quirk 11: all ranks racing on
+    the download, /root/reference/cifar_example_ddp.py:67-69)."""
+    import tarfile
+    import urllib.request
+
+    base = os.path.join(root, _CIFAR_DIR)
+    if _batches_present(base):
+        return
+    os.makedirs(root, exist_ok=True)
+    tgz = os.path.join(root, os.path.basename(url) or "cifar-10-python.tar.gz")
+    if not (os.path.exists(tgz) and _md5(tgz) == md5):
+        part = tgz + ".part"
+        urllib.request.urlretrieve(url, part)
+        os.replace(part, tgz)
+    got = _md5(tgz)
+    if got != md5:
+        raise RuntimeError(
+            f"CIFAR-10 archive checksum mismatch at {tgz}: got {got}, "
+            f"expected {md5} — delete the file and retry")
+    with tarfile.open(tgz, "r:gz") as tf:
+        try:
+            tf.extractall(root, filter="data")
+        except TypeError:  # tarfile without extraction filters
+            for m in tf.getmembers():
+                if m.name.startswith(("/", "..")) or ".." in m.name.split("/"):
+                    raise RuntimeError(f"unsafe path in archive: {m.name}")
+            tf.extractall(root)
+    if not _batches_present(base):
+        raise RuntimeError(
+            f"CIFAR-10 archive extracted but batches missing under {base}")
 
 
 class CIFAR10:
@@ -63,12 +120,18 @@ class CIFAR10:
     Accepts the same on-disk layout torchvision downloads
     (``<root>/cifar-10-batches-py/data_batch_{1..5}``, ``test_batch``; each a
     pickle with ``b"data"`` as (N, 3072) uint8 row-major RRR...GGG...BBB and
-    ``b"labels"`` a list of ints).  Samples come back as float32 (3, 32, 32)
-    tensors normalized to [-1, 1] — identical math to the reference's
-    ToTensor + Normalize((0.5,)*3, (0.5,)*3) (cifar_example.py:38-40).
+    ``b"labels"`` a list of ints).  ``download=True`` bootstraps a clean box
+    via :func:`download_cifar10` (reference surface:
+    torchvision.datasets.CIFAR10(root, download=True),
+    /root/reference/cifar_example.py:40-44).  Samples come back as float32
+    (3, 32, 32) tensors normalized to [-1, 1] — identical math to the
+    reference's ToTensor + Normalize((0.5,)*3, (0.5,)*3)
+    (cifar_example.py:38-40).
     """
 
-    def __init__(self, root: str, train: bool = True):
+    def __init__(self, root: str, train: bool = True, download: bool = False):
+        if download:
+            download_cifar10(root)
         base = os.path.join(root, _CIFAR_DIR)
         names = _TRAIN_BATCHES if train else _TEST_BATCHES
         datas, labels = [], []
@@ -199,7 +262,15 @@ class DataLoader:
 
     def __init__(self, dataset, batch_size: int = 1, shuffle: bool = False,
                  sampler: Optional[Sequence[int]] = None,
-                 drop_last: bool = False, device=None, seed: int = 0):
+                 drop_last: bool = False, device=None, seed: int = 0,
+                 num_workers: int = 0):
+        # num_workers is accepted for reference-surface parity
+        # (cifar_example.py:47,52 passes num_workers=2). Worker PROCESSES
+        # are deliberately not spawned: our datasets are fully materialized
+        # tensors (no per-item decode), so batch gathering is one indexed
+        # copy and the GPU path overlaps H2D on a copy stream instead —
+        # there is no decode work to parallelize.
+        self.num_workers = num_workers
         self.dataset = dataset
         self.batch_size = batch_size
         self.shuffle = shuffle
