@@ -71,7 +71,11 @@ def main():
         rank, world, local = comm.init_process_group()
     else:
         rank, local = 0, 0
-    device = torch.device("cuda", local) if use_cuda else torch.device("cpu")
+    # modulo so oversubscribed layouts (more ranks than GPUs — rehearsal
+    # on a 1-GPU box with gloo) land on a valid ordinal instead of dying
+    # with "invalid device ordinal"
+    device = (torch.device("cuda", local % torch.cuda.device_count())
+              if use_cuda else torch.device("cpu"))
     if args.fp16:
         from mi355x import amp
         amp.set_compute_dtype(torch.float16)

@@ -62,7 +62,8 @@ def main(args):
     init_distributed(args)
     trainset, testset = get_datasets(args)
     use_cuda = torch.cuda.is_available()
-    device = torch.device("cuda", args.gpu) if use_cuda else torch.device("cpu")
+    device = (torch.device("cuda", args.gpu % torch.cuda.device_count())
+              if use_cuda else torch.device("cpu"))
 
     batch = int(os.environ.get("MI355X_BATCH", "4"))
     epochs = int(os.environ.get("MI355X_EPOCHS", "2"))
