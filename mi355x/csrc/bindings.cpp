@@ -44,8 +44,10 @@ at::Tensor maxpool_bwd(at::Tensor dy, at::Tensor idx, long H, long W,
 at::Tensor global_avg_pool(at::Tensor x);
 
 at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act);
-at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w);
+at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wt);
 at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy);
+at::Tensor gemm_nt_mfma(at::Tensor A, at::Tensor B, at::Tensor bias, long act,
+                        bool out32);
 
 std::vector<at::Tensor> cross_entropy_fwd(at::Tensor logits,
                                           at::Tensor target);
@@ -78,6 +80,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("linear_fwd", &linear_fwd);
   m.def("linear_dgrad", &linear_dgrad);
   m.def("linear_wgrad", &linear_wgrad);
+  m.def("gemm_nt_mfma", &gemm_nt_mfma);
   m.def("cross_entropy_fwd", &cross_entropy_fwd);
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
 }

@@ -306,6 +306,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     const T16* __restrict__ in,   // [N, Hi, Wi, CI] (dgrad: dy, CI=KO)
     const T16* __restrict__ wgt,  // fwd: [KO, 9*CI]; dgrad: wflip strides
     const float* __restrict__ bias, T16* __restrict__ out,
+    float* __restrict__ stats_slab,  // null, or per-block (sum,sumsq) rows
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const long b_row_stride, const long b_rs_stride, const int act,
     const int has_bias, const int NR) {
@@ -419,10 +420,11 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     }
   }
 
-  // ---- epilogue: bias + act + store ----
+  // ---- epilogue: bias + act + store (+ optional BN-stats partials) ----
   float bv[2];
   bv[0] = has_bias ? bias[k0 + li] : 0.f;
   bv[1] = has_bias ? bias[k0 + 32 + li] : 0.f;
+  float ssum[2] = {}, ssq[2] = {};
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
@@ -432,9 +434,31 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
       for (int t2 = 0; t2 < 2; ++t2) {
         float v = acc[t2][reg] + bv[t2];
         if (act == 1) v = fmaxf(v, 0.f);
+        if (stats_slab) {
+          ssum[t2] += v;
+          ssq[t2] += v * v;
+        }
         out[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
       }
     }
+  }
+  if (stats_slab) {
+    // conv->BN fusion (same fold as conv_gather_gemm): 8 lanes per
+    // channel (4 waves x 2 kh halves) reduce through LDS, one slab row
+    // per block — replaces BN's separate bn_stats pass
+    float* lsum = reinterpret_cast<float*>(psmem);
+    __syncthreads();
+    for (int t = tid; t < 2 * 64; t += 256) lsum[t] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int t2 = 0; t2 < 2; ++t2) {
+      atomicAdd(lsum + t2 * 32 + li, ssum[t2]);
+      atomicAdd(lsum + 64 + t2 * 32 + li, ssq[t2]);
+    }
+    __syncthreads();
+    float* slab =
+        stats_slab + ((long)blockIdx.y * gridDim.x + blockIdx.x) * 2 * 64;
+    for (int t = tid; t < 2 * 64; t += 256) slab[t] = lsum[t];
   }
 }
 
@@ -1558,23 +1582,36 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     const char* e = getenv("MI355X_CONV_PATCH");
     return !e || e[0] != '0';
   }();
-  if (patch_on && !(stats.defined() && stats.numel() > 0) && R == 3 &&
-      S == 3 && stride == 1 && pad == 1 && Ho == Hi && Wo == Wi &&
-      Wo <= 64) {
+  if (patch_on && R == 3 && S == 3 && stride == 1 && pad == 1 && Ho == Hi &&
+      Wo == Wi && Wo <= 64) {
     // rows spanned by a 128-output tile starting mid-row, +2 halo rows:
     // 128/Wo+3 was ONE short for Wo in {56,28,14,7} (tap r=2 then read
     // past the patch -> diverging ResNet-50 training)
     const int NR = (Wo + 126) / Wo + 3;
     const size_t smem = ((size_t)NR * (Wi + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), KO / 64);
+    at::Tensor pslab;
+    float* pslab_ptr = nullptr;
+    if (stats.defined() && stats.numel() > 0) {
+      pslab = at::empty({(long)pgrid_.y * pgrid_.x * 128},
+                        x.options().dtype(at::kFloat));
+      pslab_ptr = pslab.data_ptr<float>();
+    }
     DISPATCH_16(x, T16, {
       hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
                          smem, cur_stream(), (const T16*)x.data_ptr(),
                          (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (T16*)y.data_ptr(), N, Hi, Wi, CI, KO,
+                         (T16*)y.data_ptr(), pslab_ptr, N, Hi, Wi, CI, KO,
                          (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
     });
+    if (pslab_ptr) {
+      const int bx_per_block = (int)cdiv_l(pgrid_.x, 64);
+      dim3 rgrid((unsigned)cdiv_l(pgrid_.x, bx_per_block), pgrid_.y);
+      hipLaunchKernelGGL(conv_stats_reduce, rgrid, dim3(256), 0,
+                         cur_stream(), pslab_ptr, stats.data_ptr<float>(),
+                         pgrid_.x, 128, KO, bx_per_block);
+    }
     return;
   }
   // BN=128 halves barriers per MFMA but also halves the grid — only use
@@ -1641,8 +1678,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
       hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
                          smem, cur_stream(), (const T16*)dy.data_ptr(),
                          (const T16*)wflip.data_ptr(), nullptr,
-                         (T16*)dx.data_ptr(), N, P, Q, KO, CI, (long)KO,
-                         (long)CI * KO, 0, 0, NR);
+                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
+                         (long)KO, (long)CI * KO, 0, 0, NR);
     });
     return;
   }

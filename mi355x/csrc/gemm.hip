@@ -87,12 +87,19 @@ inline int ggrid(long n) { return (int)std::min<long>(cdiv_l(n, 256), 4096); }
 
 }  // namespace
 
+at::Tensor gemm_nt_mfma(at::Tensor A, at::Tensor B, at::Tensor bias, long act,
+                        bool out32);  // gemm_mfma.hip
+
 at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
   const int M = x.size(0), K = x.size(1), N = w.size(0);
   TORCH_CHECK(w.size(1) == K, "linear weight/input mismatch");
+  // MFMA tile path once there is at least one K-slice and a few M-tiles;
+  // direct kernel keeps the sub-MFMA shapes (K % 8 != 0 or tiny M)
+  if (K % 8 == 0 && K >= 64 && M >= 128)
+    return gemm_nt_mfma(x, w, bias, act, false);
   auto y = at::empty({M, N}, x.options());
   const int has_bias = bias.numel() > 0;
   DISPATCH_16(x, T16, {
@@ -105,10 +112,17 @@ at::Tensor linear_fwd(at::Tensor x, at::Tensor w, at::Tensor bias, long act) {
   return y;
 }
 
-at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w) {
+// wt (optional, may be empty): cached [K,N] transpose of w — the MFMA
+// dgrad runs dx = dy @ w as gemm_nt(dy, wT) when the N reduction is
+// MFMA-sized.
+at::Tensor linear_dgrad(at::Tensor dy, at::Tensor w, at::Tensor wt) {
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
   const int M = dy.size(0), N = dy.size(1), K = w.size(1);
+  if (wt.numel() > 0 && N % 8 == 0 && N >= 64 && M >= 128) {
+    auto none = at::empty({0}, dy.options().dtype(at::kFloat));
+    return gemm_nt_mfma(dy, wt, none, 0, false);
+  }
   auto dx = at::empty({M, K}, dy.options());
   DISPATCH_16(dy, T16, {
     hipLaunchKernelGGL(linear_dgrad_kernel<T16>, dim3(ggrid((long)M * K)),
@@ -123,6 +137,14 @@ at::Tensor linear_wgrad(at::Tensor x, at::Tensor dy) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   const int M = x.size(0), K = x.size(1), N = dy.size(1);
+  // dw = dy^T @ x as gemm_nt(dyT, xT) with the M reduction in the MFMA
+  // K-slot; the two transposes are tiny next to the M-long reduction
+  if (M % 8 == 0 && M >= 256 && N >= 32 && K >= 64) {
+    auto dyT = dy.t().contiguous();
+    auto xT = x.t().contiguous();
+    auto none = at::empty({0}, x.options().dtype(at::kFloat));
+    return gemm_nt_mfma(dyT, xT, none, 0, true);
+  }
   const long total = (long)N * K;
   const long xy_blocks = cdiv_l(total, 256);
   int nchunks = (int)std::min<long>(std::max<long>(512 / xy_blocks, 1),

@@ -96,6 +96,18 @@ def _w16_linear(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
     return w16
 
 
+def _w16_linear_t(weight: torch.Tensor, dtype: torch.dtype) -> torch.Tensor:
+    """fp32 [N,K] parameter -> cached 16-bit [K,N] transpose (the MFMA
+    dgrad's B operand: dx = dy @ w == gemm_nt(dy, wT))."""
+    key = (weight._version, _cache_epoch, dtype)
+    cache = getattr(weight, "_mi355x_wT", None)
+    if cache is not None and cache[0] == key:
+        return cache[1]
+    wt = weight.detach().to(dtype).t().contiguous()
+    weight._mi355x_wT = (key, wt)
+    return wt
+
+
 def _relu_mask_bwd(dy: torch.Tensor, y: torch.Tensor) -> torch.Tensor:
     """dy masked by y>0, via the extension's elementwise kernel."""
     return ext().relu_bwd(dy.contiguous(), y)
@@ -193,6 +205,7 @@ class _LinearFn(torch.autograd.Function):
         b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
         y = ext().linear_fwd(x, w16, b32, act)
         ctx.save_for_backward(x, w16, y)
+        ctx.weight_ref = weight  # for the cached dgrad transpose
         ctx.conf = (act, bias is not None)
         return y
 
@@ -203,7 +216,10 @@ class _LinearFn(torch.autograd.Function):
         dy = dy.contiguous()
         if act == _ACT_RELU:
             dy = _relu_mask_bwd(dy, y)
-        dx = ext().linear_dgrad(dy, w16) if ctx.needs_input_grad[0] else None
+        dx = None
+        if ctx.needs_input_grad[0]:
+            wt = _w16_linear_t(ctx.weight_ref, dy.dtype)
+            dx = ext().linear_dgrad(dy, w16, wt)
         dw = ext().linear_wgrad(x, dy) if ctx.needs_input_grad[1] else None
         db = dy.float().sum(dim=0) if (has_bias and ctx.needs_input_grad[2]) else None
         return dx, dw, db, None

@@ -24,6 +24,7 @@ SRC = [
     "mi355x/csrc/bn.hip",
     "mi355x/csrc/pool.hip",
     "mi355x/csrc/gemm.hip",
+    "mi355x/csrc/gemm_mfma.hip",
     "mi355x/csrc/loss.hip",
     "mi355x/csrc/rccl_comm.cpp",
 ]

@@ -92,6 +92,11 @@ def test_conv2d_fwd_bwd(shape, K, ksz, stride, pad, bias, act):
     (4, 84, 120, True, "relu"),
     (32, 10, 512, True, None),
     (8, 33, 100, False, None),  # K % 8 != 0 scalar path
+    # MFMA tile path (gemm_mfma.hip): exact tiles and M/N/K tails
+    (256, 64, 64, True, None),
+    (300, 100, 128, True, "relu"),   # M tail, N tail, dgrad falls back
+    (512, 1000, 256, False, None),   # ResNet-50-classifier-like, MFMA dgrad
+    (256, 120, 400, True, None),     # partial final K-slice (400 = 6*64+16)
 ])
 def test_linear_fwd_bwd(M, N, K, bias, act):
     xc, xg = _pair((M, K), requires_grad=True)
