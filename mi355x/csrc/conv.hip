@@ -828,6 +828,19 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                             long R, long S, long stride, long pad);
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                             long R, long S, long stride, long pad);
+// stem_mfma.hip — MFMA GEMM stem (7x7/s2/C=3), replaces the dot2 kernels
+void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                               at::Tensor y, long pad, long act);
+void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+                                 long pad);
+
+static bool stem_gemm_on() {
+  static const bool v = [] {
+    const char* e = getenv("MI355X_STEM_GEMM");
+    return !e || e[0] != '0';
+  }();
+  return v;
+}
 
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act, long kR, long kS) {
@@ -844,6 +857,10 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   // dedicated ImageNet-stem kernel: 7x7/s2/C=3 (dot2-packed sliding rows)
   const bool stem7 = K <= 64 && C == 3 && R == 7 && S == 7 && stride == 2 &&
                      pad <= 3;
+  if (stem7 && w.dim() == 2 && K % 64 == 0 && stem_gemm_on()) {
+    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act);
+    return y;
+  }
   if (stem7) {
     const long M = (long)N * P * Q;
     const long wrow = w.dim() == 2 ? w.size(1) : (long)R * S * C;
@@ -1035,7 +1052,10 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     nchunks = cdiv_l(M, m_per_chunk);
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     if (C == 3 && S == 7 && stride == 2 && pad <= 4) {
-
+      if (K % 64 == 0 && R == 7 && pad <= 3 && stem_gemm_on()) {
+        conv_wgrad_stem_gemm_launch(x, dy, dw, pad);
+        return dw;
+      }
       const long nrows = (long)N * P;
       // 512 blocks x 4 waves at occupancy 2 fills the chip exactly once;
       // fewer chunks = 4x less partial-slab traffic for the reduce

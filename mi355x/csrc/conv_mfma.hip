@@ -236,24 +236,50 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     }
   }
   if (stats_slab) {
-    // conv->BN fusion: fold the block's per-lane channel partials in LDS
-    // (8 lanes per channel: 4 waves x 2 kh halves) and store one plain
-    // slab entry per channel — replaces BN's separate bn_stats pass
+    // conv->BN fusion: fold the block's per-lane channel partials (8
+    // lanes per channel: 4 waves x 2 kh halves) — kh halves via one
+    // cross-lane xor-shuffle, waves via per-wave LDS rows summed by the
+    // storing threads. (LDS float atomicAdd here measured ~3.7 us/block
+    // of serialized conflicts — 50% on the whole patch kernel.)
     constexpr int BNTC = NT * 32;
-    float* lsum = reinterpret_cast<float*>(lds);  // reuse after barrier
-    __syncthreads();
-    for (int t = tid; t < 2 * BNTC; t += 256) lsum[t] = 0.f;
-    __syncthreads();
 #pragma unroll
     for (int tnt = 0; tnt < NT; ++tnt) {
-      atomicAdd(lsum + tnt * 32 + li, ssum[tnt]);
-      atomicAdd(lsum + BNTC + tnt * 32 + li, ssq[tnt]);
+      ssum[tnt] += __shfl_xor(ssum[tnt], 32, 64);
+      ssq[tnt] += __shfl_xor(ssq[tnt], 32, 64);
+    }
+    float* lsum = reinterpret_cast<float*>(lds);  // [4 waves][2*BNTC]
+    __syncthreads();
+    if (kh == 0) {
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        lsum[wave * 2 * BNTC + tnt * 32 + li] = ssum[tnt];
+        lsum[wave * 2 * BNTC + BNTC + tnt * 32 + li] = ssq[tnt];
+      }
     }
     __syncthreads();
     float* slab =
         stats_slab + ((long)blockIdx.y * gridDim.x + blockIdx.x) * 2 * BNTC;
-    for (int t = tid; t < 2 * BNTC; t += 256) slab[t] = lsum[t];
+    for (int t = tid; t < 2 * BNTC; t += 256)
+      slab[t] = lsum[t] + lsum[2 * BNTC + t] + lsum[4 * BNTC + t] +
+                lsum[6 * BNTC + t];
   }
+}
+
+// L1 of the two-level slab fold: plain partial sums over a gx-slice
+// (large-M convs have 10k-65k slab rows; a single 64-block reduce over
+// them measured 139 us latency-bound — this level keeps the chip full)
+__global__ void conv_stats_partial(const float* __restrict__ slab,
+                                   float* __restrict__ out, int gx, int bnt2,
+                                   int bpb) {
+  const int by = blockIdx.y;
+  const int t = threadIdx.x;
+  if (t >= bnt2) return;
+  const int b0 = blockIdx.x * bpb;
+  const int b1 = min(gx, b0 + bpb);
+  float acc = 0.f;
+  for (int bx = b0; bx < b1; ++bx)
+    acc += slab[((long)by * gx + bx) * bnt2 + t];
+  out[((long)by * gridDim.x + blockIdx.x) * bnt2 + t] = acc;
 }
 
 // stats[2][C] = sum over gx of slab[gy][gx][2][BNT]; one block per
@@ -276,6 +302,27 @@ __global__ void conv_stats_reduce(const float* __restrict__ slab,
     stats[h * C + c] = acc;
   else
     atomicAdd(stats + h * C + c, acc);
+}
+
+// fold slab [by][gx][bnt2] -> stats[2,C]; two-level when gx is large
+void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
+                       int C) {
+  const int by = (int)(slab.numel() / ((long)gx * bnt2));
+  if (gx > 4096) {
+    const int bpb1 = (int)cdiv_l(gx, 1024);
+    const int gx2 = (int)cdiv_l(gx, bpb1);
+    auto l2 = at::empty({(long)by * gx2 * bnt2}, slab.options());
+    hipLaunchKernelGGL(conv_stats_partial, dim3(gx2, by), dim3(256), 0,
+                       cur_stream(), slab.data_ptr<float>(),
+                       l2.data_ptr<float>(), gx, bnt2, bpb1);
+    slab = l2;
+    gx = gx2;
+  }
+  const int bpb = (int)cdiv_l(gx, 64);
+  dim3 rgrid((unsigned)cdiv_l(gx, bpb), by);
+  hipLaunchKernelGGL(conv_stats_reduce, rgrid, dim3(256), 0, cur_stream(),
+                     slab.data_ptr<float>(), stats.data_ptr<float>(), gx,
+                     bnt2, C, bpb);
 }
 
 }  // namespace
@@ -443,22 +490,27 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     }
   }
   if (stats_slab) {
-    // conv->BN fusion (same fold as conv_gather_gemm): 8 lanes per
-    // channel (4 waves x 2 kh halves) reduce through LDS, one slab row
-    // per block — replaces BN's separate bn_stats pass
-    float* lsum = reinterpret_cast<float*>(psmem);
-    __syncthreads();
-    for (int t = tid; t < 2 * 64; t += 256) lsum[t] = 0.f;
-    __syncthreads();
+    // conv->BN fusion (same fold as conv_gather_gemm): kh halves by
+    // xor-shuffle, waves by per-wave LDS rows (no LDS atomics)
 #pragma unroll
     for (int t2 = 0; t2 < 2; ++t2) {
-      atomicAdd(lsum + t2 * 32 + li, ssum[t2]);
-      atomicAdd(lsum + 64 + t2 * 32 + li, ssq[t2]);
+      ssum[t2] += __shfl_xor(ssum[t2], 32, 64);
+      ssq[t2] += __shfl_xor(ssq[t2], 32, 64);
+    }
+    float* lsum = reinterpret_cast<float*>(psmem);  // [4 waves][128]
+    __syncthreads();
+    if (kh == 0) {
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2) {
+        lsum[wave * 128 + t2 * 32 + li] = ssum[t2];
+        lsum[wave * 128 + 64 + t2 * 32 + li] = ssq[t2];
+      }
     }
     __syncthreads();
     float* slab =
         stats_slab + ((long)blockIdx.y * gridDim.x + blockIdx.x) * 2 * 64;
-    for (int t = tid; t < 2 * 64; t += 256) slab[t] = lsum[t];
+    for (int t = tid; t < 2 * 64; t += 256)
+      slab[t] = lsum[t] + lsum[128 + t] + lsum[256 + t] + lsum[384 + t];
   }
 }
 
@@ -1605,13 +1657,8 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          (T16*)y.data_ptr(), pslab_ptr, N, Hi, Wi, CI, KO,
                          (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
     });
-    if (pslab_ptr) {
-      const int bx_per_block = (int)cdiv_l(pgrid_.x, 64);
-      dim3 rgrid((unsigned)cdiv_l(pgrid_.x, bx_per_block), pgrid_.y);
-      hipLaunchKernelGGL(conv_stats_reduce, rgrid, dim3(256), 0,
-                         cur_stream(), pslab_ptr, stats.data_ptr<float>(),
-                         pgrid_.x, 128, KO, bx_per_block);
-    }
+    if (pslab_ptr)
+      stats_slab_reduce(pslab, stats, pgrid_.x, 128, KO);
     return;
   }
   // BN=128 halves barriers per MFMA but also halves the grid — only use
@@ -1649,13 +1696,8 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
                          (int)act, has_bias);
   });
-  if (stats_slab_ptr) {
-    const int bx_per_block = (int)cdiv_l(grid.x, 64);
-    dim3 rgrid((unsigned)cdiv_l(grid.x, bx_per_block), grid.y);
-    hipLaunchKernelGGL(conv_stats_reduce, rgrid, dim3(256), 0, cur_stream(),
-                       stats_slab_ptr, stats.data_ptr<float>(), grid.x, bnt2,
-                       KO, bx_per_block);
-  }
+  if (stats_slab_ptr)
+    stats_slab_reduce(slab, stats, grid.x, bnt2, KO);
 }
 
 // dgrad: in = dy[N,P,Q,KO], wflip = [R,S,CI,KO] (w[k,R-1-r,S-1-s,c]),

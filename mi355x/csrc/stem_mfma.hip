@@ -1,0 +1,356 @@
+// MFMA GEMM kernels for the ImageNet stem conv (7x7/stride-2/C=3 —
+// BASELINE config 4's first layer; reference conv surface
+// /root/reference/cifar_example.py:20-29 generalized to ResNet-50).
+//
+// The dot2 stem kernels (conv.hip) are VALU/LDS-bound at ~30-64 TF/s;
+// these route the same im2col GEMM through the matrix cores. The im2col
+// k-axis uses a ROW-PADDED layout: kg = r*24 + (s*3 + c), rows padded
+// 21 -> 24 so every 8-element k-group stays inside one filter row
+// (24 % 8 == 0 — a group is then 8 CONTIGUOUS elements of one input
+// image row, loadable as one 16-byte read away from the edges), total
+// padded K = 192 (7*24 rounded to 3 BK=64 chunks; utilization 147/192).
+// The weight is re-laid [KO, 192] on the host per call (27 KB, cached
+// activations dwarf it).
+//
+//   fwd  : y[m, ko]   = sum_kg A_im2col[m, kg] * w24[ko, kg]
+//          one block = 128 m x 64 ko, full k staged once (one barrier
+//          pair, 24 MFMAs per wave), epilogue = bias + act + store.
+//   wgrad: dw[ko, kg] = sum_m dy[m, ko] * A_im2col[m, kg]
+//          conv_wgrad_mfma_kernel shape: both operands staged TRANSPOSED
+//          ([ko][m] / [kg][m]) per 64-m step, split-M chunk slabs in the
+//          [KO,C,R,S] parameter layout, wgrad_reduce sums chunks.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) _Float16 half8_s;
+typedef __attribute__((ext_vector_type(16))) float f32x16_s;
+
+namespace {
+
+constexpr int SROW = 24;   // padded elements per filter row (21 valid)
+constexpr int SKG = 192;   // padded im2col K (7*24 -> 3 chunks of 64)
+constexpr int SLDK = SKG + 8;
+constexpr int SLDM = 72;   // wgrad m-minor row length
+
+template <typename T16>
+struct SMfma {};
+template <>
+struct SMfma<__hip_bfloat16> {
+  static DEV_INLINE f32x16_s run(short8 a, short8 b, f32x16_s c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+  }
+};
+template <>
+struct SMfma<__half> {
+  static DEV_INLINE f32x16_s run(short8 a, short8 b, f32x16_s c) {
+    return __builtin_amdgcn_mfma_f32_32x32x16_f16((half8_s)a, (half8_s)b, c,
+                                                  0, 0, 0);
+  }
+};
+
+// one 8-element k-group of the im2col row for output (n,p,q):
+// kg0 = r*24 + jj, elements x[n, 2p-pad+r, (2q-pad)*3 + jj + u]
+template <typename T16>
+DEV_INLINE short8 stem_gather8(const T16* __restrict__ x, int n, int p, int q,
+                               int kg0, int H, int W3, int pad, bool m_ok,
+                               long last_safe) {
+  const int r = kg0 / SROW;
+  const int jj = kg0 - r * SROW;
+  const int ih = 2 * p - pad + r;
+  const short8 zero8 = {};
+  if (!m_ok || r >= 7 || (unsigned)ih >= (unsigned)H) return zero8;
+  const int e0 = (2 * q - pad) * 3 + jj;
+  const long base = ((long)n * H + ih) * (long)W3;
+  if (jj + 8 <= 21 && e0 >= 0 && e0 + 8 <= W3)  // interior: one 16B read
+    return *reinterpret_cast<const short8*>(x + base + e0);
+  short8 v = {};
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const int e = e0 + u;
+    if (jj + u < 21 && (unsigned)e < (unsigned)W3)
+      v[u] = *reinterpret_cast<const short*>(x + base + e);
+  }
+  (void)last_safe;
+  return v;
+}
+
+// ---- forward ----------------------------------------------------------
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
+    const T16* __restrict__ x,    // [N, H, W, 3]
+    const T16* __restrict__ w24,  // [KO, 192] row-padded
+    const float* __restrict__ bias, T16* __restrict__ y, const int N,
+    const int H, const int W, const int KO, const int Ho, const int Wo,
+    const int pad, const int act, const int has_bias) {
+  __shared__ T16 lds[128 * SLDK + 64 * SLDK];
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const long bm0 = (long)blockIdx.x * 128;
+  const int k0 = blockIdx.y * 64;
+  const int W3 = 3 * W;
+
+  // A staging: 2 threads per m-row, 96 k-elements (12 groups) each
+  const int sa_m = tid >> 1;
+  const int sa_c = (tid & 1) * 96;
+  const long m_a = bm0 + sa_m;
+  const bool m_ok = m_a < Mtot;
+  int n_ = 0, p_ = 0, q_ = 0;
+  if (m_ok) {
+    n_ = (int)(m_a / ((long)Ho * Wo));
+    const int pq = (int)(m_a % ((long)Ho * Wo));
+    p_ = pq / Wo;
+    q_ = pq % Wo;
+  }
+  {
+    T16* pa = lds + sa_m * SLDK + sa_c;
+#pragma unroll
+    for (int i = 0; i < 12; ++i)
+      *reinterpret_cast<short8*>(reinterpret_cast<short*>(pa) + 8 * i) =
+          stem_gather8(x, n_, p_, q_, sa_c + 8 * i, H, W3, pad, m_ok, 0);
+  }
+  // B staging: flat 8-groups of the [64, 192] weight tile
+  {
+    T16* ldsB = lds + 128 * SLDK;
+    for (int e = tid * 8; e < 64 * SKG; e += 256 * 8) {
+      const int row = e / SKG, col = e - row * SKG;
+      *reinterpret_cast<short8*>(
+          reinterpret_cast<short*>(ldsB + row * SLDK + col)) =
+          *reinterpret_cast<const short8*>(w24 + (long)(k0 + row) * SKG +
+                                           col);
+    }
+  }
+  __syncthreads();
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int wm = wave * 32;
+  f32x16_s acc[2] = {};
+  const T16* ldsA = lds;
+  const T16* ldsB = lds + 128 * SLDK;
+#pragma unroll
+  for (int kk = 0; kk < SKG; kk += 16) {
+    const short8 af = *reinterpret_cast<const short8*>(
+        ldsA + (wm + li) * SLDK + kk + kh * 8);
+    const short8 b0 = *reinterpret_cast<const short8*>(
+        ldsB + li * SLDK + kk + kh * 8);
+    const short8 b1 = *reinterpret_cast<const short8*>(
+        ldsB + (32 + li) * SLDK + kk + kh * 8);
+    acc[0] = SMfma<T16>::run(af, b0, acc[0]);
+    acc[1] = SMfma<T16>::run(af, b1, acc[1]);
+  }
+
+  float bv[2];
+  bv[0] = has_bias ? bias[k0 + li] : 0.f;
+  bv[1] = has_bias ? bias[k0 + 32 + li] : 0.f;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mtot) {
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2) {
+        float v = acc[t2][reg] + bv[t2];
+        if (act == 1) v = fmaxf(v, 0.f);
+        y[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
+      }
+    }
+  }
+}
+
+// ---- wgrad ------------------------------------------------------------
+// dw slab scatter target layout: [KO, 3, 7, 7] (parameter layout), one
+// slab per m-chunk; grid = (KO/64, 3 kg-chunks, m-chunks).
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
+    const T16* __restrict__ x,   // [N, H, W, 3]
+    const T16* __restrict__ dy,  // [M, KO]
+    float* __restrict__ dw,      // chunk slabs of [KO*3*7*7]
+    const int N, const int H, const int W, const int KO, const int Ho,
+    const int Wo, const int pad, const long m_per_chunk) {
+  __shared__ T16 lds[(64 + 64) * SLDM];
+  const int tid = threadIdx.x;
+  const long Mtot = (long)N * Ho * Wo;
+  const int k0 = blockIdx.x * 64;
+  const int c0 = blockIdx.y * 64;  // kg-chunk
+  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_end = min(Mtot, m_begin + m_per_chunk);
+  const int W3 = 3 * W;
+
+  const bool do_dy = tid < 128;
+  const bool do_x = tid >= 128;
+  const int t = do_dy ? tid : 0;
+  const int sm = (t & 15) * 4;
+  const int sk = (t >> 4) * 8;
+  const int tx = tid & 127;
+  const int smx = (tx & 15) * 4;
+  const int skx = (tx >> 4) * 8;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int i0 = (wave & 1) * 32;       // KO sub-tile
+  const int j0 = (wave >> 1) * 32;      // kg sub-tile
+
+  f32x16_s acc = {};
+
+  int dn = 0, dp = 0, dq = 0;
+  {
+    const long m_first = m_begin + smx;
+    dn = (int)(m_first / ((long)Ho * Wo));
+    const int pq = (int)(m_first % ((long)Ho * Wo));
+    dp = pq / Wo;
+    dq = pq % Wo;
+  }
+  auto advance = [&](int by) {
+    dq += by;
+    while (dq >= Wo) {
+      dq -= Wo;
+      if (++dp == Ho) {
+        dp = 0;
+        ++dn;
+      }
+    }
+  };
+
+  short8 vdy[4], vx[4];
+  auto load_m = [&](long m0) {
+    if (do_dy) {
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const long m = m0 + sm + mi;
+        vdy[mi] = (m < m_end) ? *reinterpret_cast<const short8*>(
+                                    dy + m * KO + k0 + sk)
+                              : short8{};
+      }
+    }
+    if (do_x) {
+      int n_ = dn, p_ = dp, q_ = dq;
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const long m = m0 + smx + mi;
+        vx[mi] = stem_gather8<T16>(x, n_, p_, q_, c0 + skx, H, W3, pad,
+                                   m < m_end, 0);
+        if (mi < 3 && ++q_ == Wo) {
+          q_ = 0;
+          if (++p_ == Ho) {
+            p_ = 0;
+            ++n_;
+          }
+        }
+      }
+      advance(64);
+    }
+  };
+  auto stage_m = [&]() {
+    if (do_dy) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(lds + (sk + e) * SLDM + sm)) = pk;
+      }
+    }
+    if (do_x) {
+      T16* ldsT = lds + 64 * SLDM;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        short4v pk = {vx[0][e], vx[1][e], vx[2][e], vx[3][e]};
+        *reinterpret_cast<short4v*>(
+            reinterpret_cast<short*>(ldsT + (skx + e) * SLDM + smx)) = pk;
+      }
+    }
+  };
+
+  load_m(m_begin);
+  for (long m0 = m_begin; m0 < m_end; m0 += 64) {
+    __syncthreads();
+    stage_m();
+    __syncthreads();
+    if (m0 + 64 < m_end) load_m(m0 + 64);
+    const T16* ldsDyT = lds;
+    const T16* ldsXT = lds + 64 * SLDM;
+#pragma unroll
+    for (int kk = 0; kk < 64; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsDyT + (i0 + li) * SLDM + kk + kh * 8);
+      const short8 bf = *reinterpret_cast<const short8*>(
+          ldsXT + (j0 + li) * SLDM + kk + kh * 8);
+      acc = SMfma<T16>::run(af, bf, acc);
+    }
+  }
+
+  // scatter the 32(ko) x 32(kg) fp32 tile into the [KO,3,7,7] slab;
+  // row-pad positions (jj >= 21) and r >= 7 are dropped
+  float* slab = dw + (long)blockIdx.z * ((long)KO * 3 * 7 * 7);
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const int kg = c0 + j0 + li;
+    const int r = kg / SROW;
+    const int jj = kg - r * SROW;
+    if (jj < 21 && r < 7) {
+      const int s = jj / 3, c = jj - s * 3;
+      slab[(((long)(k0 + i0 + i) * 3 + c) * 7 + r) * 7 + s] = acc[reg];
+    }
+  }
+}
+
+}  // namespace
+
+// host launchers --------------------------------------------------------
+
+// w: [KO, KGP] tap-major padded ([kg = (r*7+s)*3+c]) -> [KO, 192]
+// row-padded (kg = r*24 + s*3 + c)
+static at::Tensor stem_w24(const at::Tensor& w) {
+  const long KO = w.size(0);
+  auto valid = w.narrow(1, 0, 147).reshape({KO, 7, 21});
+  auto padded = at::constant_pad_nd(valid, {0, 3}, 0);  // [KO,7,24]
+  return at::constant_pad_nd(padded.reshape({KO, 168}), {0, SKG - 168}, 0)
+      .contiguous();
+}
+
+void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                               at::Tensor y, long pad, long act) {
+  const int N = x.size(0), H = x.size(1), W = x.size(2);
+  const int KO = w.size(0);
+  const int Ho = y.size(1), Wo = y.size(2);
+  TORCH_CHECK(KO % 64 == 0, "stem GEMM expects KO % 64 == 0");
+  auto w24 = stem_w24(w);
+  const long M = (long)N * Ho * Wo;
+  dim3 grid((unsigned)cdiv_l(M, 128), KO / 64);
+  const int has_bias = bias.numel() > 0;
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL((conv_fwd_stem_gemm<T16>), grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)w24.data_ptr(),
+                       has_bias ? bias.data_ptr<float>() : nullptr,
+                       (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo, (int)pad,
+                       (int)act, has_bias);
+  });
+}
+
+void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
+
+// dw [KO,3,7,7] fp32
+void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+                                 long pad) {
+  const int N = x.size(0), H = x.size(1), W = x.size(2);
+  const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
+  TORCH_CHECK(KO % 64 == 0, "stem wgrad GEMM expects KO % 64 == 0");
+  const long M = (long)N * Ho * Wo;
+  const long E = (long)KO * 3 * 7 * 7;
+  long nchunks = std::min<long>(512, cdiv_l(M, 4096));
+  const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
+  nchunks = cdiv_l(M, m_per_chunk);
+  auto part = at::empty({nchunks * E}, x.options().dtype(at::kFloat));
+  dim3 grid(KO / 64, SKG / 64, (unsigned)nchunks);
+  DISPATCH_16(x, T16, {
+    hipLaunchKernelGGL((conv_wgrad_stem_gemm<T16>), grid, dim3(256), 0,
+                       cur_stream(), (const T16*)x.data_ptr(),
+                       (const T16*)dy.data_ptr(), part.data_ptr<float>(), N,
+                       H, W, KO, Ho, Wo, (int)pad, m_per_chunk);
+  });
+  wgrad_reduce_launch(part, dw, E, nchunks);
+}

@@ -135,7 +135,11 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
     (69.6k -> 54.4k img/s) and mildly positive on r50-224 (3.9k -> 4.15k),
     so it defaults OFF pending investigation of the epilogue cost."""
     if (_fuse_stats() and x.is_cuda and bn.training and conv.bias is None
-            and conv.act is None):
+            and conv.act is None
+            # only shapes with a stats-emitting MFMA path: C % 64 == 0
+            # (the C=3 stem would fall off its dedicated kernel and onto
+            # the slow per-element GENC gather — measured 7.3 ms/call)
+            and conv.weight.shape[1] % 64 == 0):
         from mi355x.ops import functional as F_
 
         bn._nbt += 1

@@ -25,6 +25,7 @@ SRC = [
     "mi355x/csrc/pool.hip",
     "mi355x/csrc/gemm.hip",
     "mi355x/csrc/gemm_mfma.hip",
+    "mi355x/csrc/stem_mfma.hip",
     "mi355x/csrc/loss.hip",
     "mi355x/csrc/rccl_comm.cpp",
 ]
