@@ -80,9 +80,9 @@ def main():
         from mi355x import amp
         amp.set_compute_dtype(torch.float16)
 
-    # conv->BN stats fusion (MI355X_FUSE_BN) measured NEGATIVE on both
-    # configs once the stem kernels were fixed (r50-224: 5356 -> 4740
-    # img/s with it on) — leave it opt-in via env only
+    # conv->BN stats fusion (MI355X_FUSE_BN) defaults ON since round 2
+    # (+0.9% r18 / +2.8% r50 after the shuffle stats fold + two-level
+    # slab reduce; MI355X_FUSE_BN=0 disables)
     num_classes = args.classes or (1000 if args.size >= 224 else 10)
     model_name = args.model if args.size < 224 or args.model != "resnet18" \
         else "resnet18_imagenet"

@@ -15,7 +15,7 @@ at::Tensor cast_permute_rsck(at::Tensor w, at::Tensor like);
 at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                       long stride, long pad, long act, long kR, long kS);
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
-                        long pad, long H, long W);
+                        long pad, long H, long W, at::Tensor addin);
 at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                         long stride, long pad);
 std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,

@@ -825,7 +825,8 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                                at::Tensor y, long R, long S, long stride,
                                long pad, long act);
 void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
-                            long R, long S, long stride, long pad);
+                            long R, long S, long stride, long pad,
+                            at::Tensor addin);
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                             long R, long S, long stride, long pad);
 // stem_mfma.hip — MFMA GEMM stem (7x7/s2/C=3), replaces the dot2 kernels
@@ -981,8 +982,11 @@ std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
   return {y, stats};
 }
 
+// addin (optional, dx-shaped): dx = dgrad(dy) + addin — carries the
+// residual-junction gradient of a ResNet block so autograd's separate
+// full-tensor add at the junction disappears (MFMA path only).
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
-                        long pad, long H, long W) {
+                        long pad, long H, long W, at::Tensor addin) {
   CHECK_GPU(dy);
   CHECK_CONTIG(dy);
   CHECK_16BIT(dy);
@@ -990,9 +994,11 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
   const int R = wflip.size(0), S = wflip.size(1), C = wflip.size(2);
   auto dx = at::empty({N, H, W, (long)C}, dy.options());
   if (conv_mfma_supported(K, C)) {
-    conv_dgrad_mfma_launch(dy, wflip, dx, R, S, stride, pad);
+    conv_dgrad_mfma_launch(dy, wflip, dx, R, S, stride, pad, addin);
     return dx;
   }
+  TORCH_CHECK(!(addin.defined() && addin.numel() > 0),
+              "addin requires the MFMA dgrad path (C,K % 64 == 0)");
   const long total = (long)N * H * W * C;
   DISPATCH_16(dy, T16, {
     hipLaunchKernelGGL(conv_dgrad_direct<T16>, conv_grid(total), dim3(256), 0,

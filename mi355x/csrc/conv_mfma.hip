@@ -45,13 +45,17 @@ constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
 // GENC (fwd only): CI % 64 != 0 — the weight is pre-padded to
 // [KO, KGP=ceil(R*S*CI/64)*64] with kg=(r*S+s)*CI+c and zeros beyond, and
 // the A side gathers per-ELEMENT across tap boundaries (stem convs C=3/6).
-template <typename T16, bool TRANS, bool GENC, int NT>
+// ADDIN (compile-time — a runtime addin branch in this shared epilogue
+// measurably slowed the non-addin convs): out += addin elementwise, the
+// residual-junction gradient fused into the dgrad epilogue.
+template <typename T16, bool TRANS, bool GENC, int NT, bool ADDIN = false>
 // min-blocks hint: same effect as on the wgrad kernel (see comment there)
 __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
     const T16* __restrict__ wgt,   // fwd: [KO, R*S*CI]; dgrad: [R*S, CI... ] via strides
     const float* __restrict__ bias,  // [KO] or null
     const T16* __restrict__ zpage,   // >=256 zero elements (OOB gather target)
+    const T16* __restrict__ addin,   // ADDIN only: same layout as out
     T16* __restrict__ out,         // [N*Ho*Wo, KO]
     float* __restrict__ stats_slab,  // null, or [gy][gx][2][BNT] partial
     const int N, const int Hi, const int Wi, const int CI, const int KO,  //  (sum,sumsq) of this block's output tile (conv->BN fusion)
@@ -226,6 +230,8 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
 #pragma unroll
       for (int tnt = 0; tnt < NT; ++tnt) {
         float v = acc[tnt][reg] + bv[tnt];
+        if constexpr (ADDIN)
+          v += F16<T16>::to_f32(addin[m_out * KO + k0 + tnt * 32 + li]);
         if (act == 1) v = fmaxf(v, 0.f);
         if (stats_slab) {
           ssum[tnt] += v;
@@ -348,11 +354,13 @@ void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
 // ---------------------------------------------------------------------------
 constexpr int PCS = 72;  // patch col stride (elements)
 
-template <typename T16, bool TRANS>
+template <typename T16, bool TRANS, bool ADDIN = false>
 __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     const T16* __restrict__ in,   // [N, Hi, Wi, CI] (dgrad: dy, CI=KO)
     const T16* __restrict__ wgt,  // fwd: [KO, 9*CI]; dgrad: wflip strides
-    const float* __restrict__ bias, T16* __restrict__ out,
+    const float* __restrict__ bias,
+    const T16* __restrict__ addin,  // ADDIN only: out += addin
+    T16* __restrict__ out,
     float* __restrict__ stats_slab,  // null, or per-block (sum,sumsq) rows
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const long b_row_stride, const long b_rs_stride, const int act,
@@ -480,6 +488,8 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
 #pragma unroll
       for (int t2 = 0; t2 < 2; ++t2) {
         float v = acc[t2][reg] + bv[t2];
+        if constexpr (ADDIN)
+          v += F16<T16>::to_f32(addin[m_out * KO + k0 + t2 * 32 + li]);
         if (act == 1) v = fmaxf(v, 0.f);
         if (stats_slab) {
           ssum[t2] += v;
@@ -1577,7 +1587,7 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
-                       (const T16*)zp.data_ptr(),
+                       (const T16*)zp.data_ptr(), nullptr,
                        (T16*)y.data_ptr(), nullptr, N, Hi, Wi, CI, KO, Ho,
                        Wo, (int)R, (int)S, (int)stride, (int)pad, KGP, 0,
                        (int)act, has_bias);
@@ -1654,7 +1664,8 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          smem, cur_stream(), (const T16*)x.data_ptr(),
                          (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (T16*)y.data_ptr(), pslab_ptr, N, Hi, Wi, CI, KO,
+                         nullptr, (T16*)y.data_ptr(), pslab_ptr, N, Hi, Wi,
+                         CI, KO,
                          (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
     });
     if (pslab_ptr)
@@ -1680,7 +1691,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(),
+                         (const T16*)zp2.data_ptr(), nullptr,
                          (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
                          KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
@@ -1690,7 +1701,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                          dim3(256), 0, cur_stream(),
                          (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
                          has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(),
+                         (const T16*)zp2.data_ptr(), nullptr,
                          (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
                          KO, Ho, Wo, R, S,
                          (int)stride, (int)pad, (long)R * S * CI, (long)CI,
@@ -1701,12 +1712,16 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
 }
 
 // dgrad: in = dy[N,P,Q,KO], wflip = [R,S,CI,KO] (w[k,R-1-r,S-1-s,c]),
-// out = dx[N,H,W,CI]
+// out = dx[N,H,W,CI]. addin (optional): dx = dgrad(dy) + addin — the
+// residual-junction gradient fused into the epilogue (compile-time
+// instantiation; the no-addin kernels are untouched).
 void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
-                            long R, long S, long stride, long pad) {
+                            long R, long S, long stride, long pad,
+                            at::Tensor addin) {
   const int N = dy.size(0), P = dy.size(1), Q = dy.size(2), KO = dy.size(3);
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
+  const bool has_add = addin.defined() && addin.numel() > 0;
   static const bool patch_on = [] {
     const char* e = getenv("MI355X_CONV_PATCH");
     return !e || e[0] != '0';
@@ -1717,11 +1732,20 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
     const size_t smem = ((size_t)NR * (W + 2) * PCS + PCS + 64 * LDK) * 2;
     dim3 pgrid_((unsigned)cdiv_l(M, 128), CI / 64);
     DISPATCH_16(dy, T16, {
-      hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
-                         smem, cur_stream(), (const T16*)dy.data_ptr(),
-                         (const T16*)wflip.data_ptr(), nullptr,
-                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
-                         (long)KO, (long)CI * KO, 0, 0, NR);
+      if (has_add)
+        hipLaunchKernelGGL((conv_patch_gemm<T16, true, true>), pgrid_,
+                           dim3(256), smem, cur_stream(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr,
+                           (const T16*)addin.data_ptr(),
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
+                           (long)KO, (long)CI * KO, 0, 0, NR);
+      else
+        hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
+                           smem, cur_stream(), (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr, nullptr,
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
+                           (long)KO, (long)CI * KO, 0, 0, NR);
     });
     return;
   }
@@ -1729,23 +1753,45 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   dim3 grid((unsigned)cdiv_l(M, BM), CI / (wide ? 128 : 64));
   at::Tensor zp = conv_zero_page(dy);
   DISPATCH_16(dy, T16, {
-    if (wide)
-      hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 4>), grid,
-                         dim3(256), 0, cur_stream(),
-                         (const T16*)dy.data_ptr(),
-                         (const T16*)wflip.data_ptr(), nullptr,
-                         (const T16*)zp.data_ptr(),
-                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
-                         (int)R, (int)S, (int)stride, (int)pad, (long)KO,
-                         (long)CI * KO, 0, 0);
-    else
-      hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 2>), grid,
-                         dim3(256), 0, cur_stream(),
-                         (const T16*)dy.data_ptr(),
-                         (const T16*)wflip.data_ptr(), nullptr,
-                         (const T16*)zp.data_ptr(),
-                         (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H, W,
-                         (int)R, (int)S, (int)stride, (int)pad, (long)KO,
-                         (long)CI * KO, 0, 0);
+    const T16* add_p = has_add ? (const T16*)addin.data_ptr() : nullptr;
+    if (wide) {
+      if (has_add)
+        hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 4, true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), add_p,
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
+                           W, (int)R, (int)S, (int)stride, (int)pad,
+                           (long)KO, (long)CI * KO, 0, 0);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 4>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), nullptr,
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
+                           W, (int)R, (int)S, (int)stride, (int)pad,
+                           (long)KO, (long)CI * KO, 0, 0);
+    } else {
+      if (has_add)
+        hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 2, true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), add_p,
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
+                           W, (int)R, (int)S, (int)stride, (int)pad,
+                           (long)KO, (long)CI * KO, 0, 0);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<T16, true, false, 2>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)dy.data_ptr(),
+                           (const T16*)wflip.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), nullptr,
+                           (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
+                           W, (int)R, (int)S, (int)stride, (int)pad,
+                           (long)KO, (long)CI * KO, 0, 0);
+    }
   });
 }

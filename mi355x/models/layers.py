@@ -123,9 +123,11 @@ class BatchNorm2d(nn.Module):
 import os
 
 def _fuse_stats():
-    # read per call (bench.py turns it on for 224-px configs where it
-    # measured positive; r18-CIFAR measured negative, stays off)
-    return os.environ.get("MI355X_FUSE_BN", "0") == "1"
+    # default ON since round 2: with the shuffle-based stats fold, the
+    # two-level slab reduce and the patch-kernel stats path it measures
+    # +0.9% r18 / +2.8% r50 (was negative in round 1 when it knocked the
+    # 3x3 convs off the patch kernel and LDS atomics serialized)
+    return os.environ.get("MI355X_FUSE_BN", "1") == "1"
 
 
 def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
@@ -152,6 +154,25 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
                               bn.running_var, bn.training, bn.momentum,
                               residual, bn.act, bn.process_group, stats)
     return bn(conv(x), residual=residual)
+
+
+def _tap_on():
+    return os.environ.get("MI355X_TAP", "1") != "0"
+
+
+def conv_bn_tap(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
+    """conv -> BN returning (out, tap). The block's shortcut consumes
+    `tap` instead of x directly, so the residual-junction gradient fuses
+    into this conv's dgrad epilogue (ops.functional._ConvTapFn) instead
+    of a separate full-tensor add. Falls back to plain conv_bn semantics
+    (tap = x) on CPU / non-MFMA shapes / MI355X_TAP=0."""
+    if not (_tap_on() and x.is_cuda and conv.bias is None
+            and conv.act is None):
+        return conv_bn(conv, bn, x, residual=residual), x
+    from mi355x.ops import functional as F_
+
+    y, tap = F_.conv2d_tap(x, conv.weight, conv.stride, conv.padding)
+    return bn(y, residual=residual), tap
 
 
 class MaxPool2d(nn.Module):

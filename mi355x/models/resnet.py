@@ -13,7 +13,7 @@ import torch
 from torch import nn
 
 from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, conv_bn,
-                     to_model_layout)
+                     conv_bn_tap, to_model_layout)
 from mi355x import ops
 
 
@@ -29,8 +29,11 @@ class BasicBlock(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = conv_bn(self.conv1, self.bn1, x)
+        # the shortcut consumes conv1's TAP: the junction gradient fuses
+        # into conv1's dgrad epilogue (layers.conv_bn_tap) instead of
+        # autograd adding two full tensors at the block input
+        out, tap = conv_bn_tap(self.conv1, self.bn1, x)
+        identity = tap if self.downsample is None else self.downsample(tap)
         out = conv_bn(self.conv2, self.bn2, out, residual=identity)
         return out
 
@@ -49,8 +52,8 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        identity = x if self.downsample is None else self.downsample(x)
-        out = conv_bn(self.conv1, self.bn1, x)
+        out, tap = conv_bn_tap(self.conv1, self.bn1, x)
+        identity = tap if self.downsample is None else self.downsample(tap)
         out = conv_bn(self.conv2, self.bn2, out)
         out = conv_bn(self.conv3, self.bn3, out, residual=identity)
         return out

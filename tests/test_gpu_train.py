@@ -141,3 +141,33 @@ def test_device_dataloader_prefetch_roundtrip():
     torch.cuda.synchronize()
     torch.testing.assert_close(torch.cat(seen_x), ds.data)
     torch.testing.assert_close(torch.cat(seen_y), ds.targets)
+
+
+@pytest.mark.parametrize("model,size", [("resnet18", 32), ("resnet50", 64)])
+def test_residual_tap_grad_parity(model, size):
+    """The residual-tap fusion (junction gradient added in the dgrad
+    epilogue, mi355x/ops/functional._ConvTapFn) produces the same flat
+    gradient as the plain double-use path (MI355X_TAP=0)."""
+    import os
+
+    torch.manual_seed(3)
+    classes = 10 if model == "resnet18" else 1000
+    net = build_model(model, num_classes=classes).cuda()
+    flat = FlatState(net)
+    g = torch.Generator().manual_seed(5)
+    x = torch.randn(8, 3, size, size, generator=g).cuda()
+    yl = torch.randint(0, classes, (8,), generator=g).cuda()
+    try:
+        os.environ["MI355X_TAP"] = "0"
+        flat.zero_grad()
+        cross_entropy(net(x), yl).backward()
+        g0 = flat.flat_grad.clone()
+        os.environ["MI355X_TAP"] = "1"
+        flat.zero_grad()
+        cross_entropy(net(x), yl).backward()
+        g1 = flat.flat_grad.clone()
+    finally:
+        os.environ.pop("MI355X_TAP", None)
+    # only the junction-add arithmetic differs (fp32 epilogue vs bf16
+    # torch add) — grads must agree to bf16-accumulation tolerance
+    torch.testing.assert_close(g1, g0, rtol=3e-2, atol=3e-3)
