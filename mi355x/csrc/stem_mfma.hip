@@ -8,17 +8,20 @@
 // 21 -> 24 so every 8-element k-group stays inside one filter row
 // (24 % 8 == 0 — a group is then 8 CONTIGUOUS elements of one input
 // image row, loadable as one 16-byte read away from the edges), total
-// padded K = 192 (7*24 rounded to 3 BK=64 chunks; utilization 147/192).
-// The weight is re-laid [KO, 192] on the host per call (27 KB, cached
-// activations dwarf it).
+// padded K = 192 = 3 BK=64 chunks (utilization 147/192). The weight is
+// re-laid [KO, 192] on the host per call (27 KB).
+//
+// Load discipline (v2): every k-group issues ONE unconditional 16-byte
+// read from a selected address (real row or a zero page) and fixes up
+// masks/edges afterwards — v1's per-group branches serialized the gather
+// into dependent round trips (same lesson as conv_gather_gemm's
+// "address select, not a branch"). The fwd kernel stages per-64-chunk
+// (27.6 KB LDS, next chunk's loads ride under the MFMA phase) instead of
+// v1's full-K 77 KB tile that capped occupancy at 2 blocks/CU.
 //
 //   fwd  : y[m, ko]   = sum_kg A_im2col[m, kg] * w24[ko, kg]
-//          one block = 128 m x 64 ko, full k staged once (one barrier
-//          pair, 24 MFMAs per wave), epilogue = bias + act + store.
 //   wgrad: dw[ko, kg] = sum_m dy[m, ko] * A_im2col[m, kg]
-//          conv_wgrad_mfma_kernel shape: both operands staged TRANSPOSED
-//          ([ko][m] / [kg][m]) per 64-m step, split-M chunk slabs in the
-//          [KO,C,R,S] parameter layout, wgrad_reduce sums chunks.
+//          (transposed LDS staging, split-M chunk slabs, wgrad_reduce)
 #include "common.h"
 
 typedef __attribute__((ext_vector_type(8))) _Float16 half8_s;
@@ -28,7 +31,7 @@ namespace {
 
 constexpr int SROW = 24;   // padded elements per filter row (21 valid)
 constexpr int SKG = 192;   // padded im2col K (7*24 -> 3 chunks of 64)
-constexpr int SLDK = SKG + 8;
+constexpr int SLDK = 64 + 8;
 constexpr int SLDM = 72;   // wgrad m-minor row length
 
 template <typename T16>
@@ -47,40 +50,64 @@ struct SMfma<__half> {
   }
 };
 
-// one 8-element k-group of the im2col row for output (n,p,q):
-// kg0 = r*24 + jj, elements x[n, 2p-pad+r, (2q-pad)*3 + jj + u]
+// per-group gather state: phase 1 picks an always-loadable address,
+// phase 2 issues the (batched) loads, phase 3 fixes up pads/edges
 template <typename T16>
-DEV_INLINE short8 stem_gather8(const T16* __restrict__ x, int n, int p, int q,
-                               int kg0, int H, int W3, int pad, bool m_ok,
-                               long last_safe) {
+struct G8 {
+  const T16* src;
+  long base;
+  int e0, jj;
+  unsigned char fast, row_ok;
+};
+
+template <typename T16>
+DEV_INLINE void g8_plan(G8<T16>& g, const T16* __restrict__ x,
+                        const T16* __restrict__ zpage, int n, int p, int q,
+                        int kg0, int H, int W3, int pad, bool m_ok) {
   const int r = kg0 / SROW;
-  const int jj = kg0 - r * SROW;
+  g.jj = kg0 - r * SROW;
   const int ih = 2 * p - pad + r;
-  const short8 zero8 = {};
-  if (!m_ok || r >= 7 || (unsigned)ih >= (unsigned)H) return zero8;
-  const int e0 = (2 * q - pad) * 3 + jj;
-  const long base = ((long)n * H + ih) * (long)W3;
-  if (jj + 8 <= 21 && e0 >= 0 && e0 + 8 <= W3)  // interior: one 16B read
-    return *reinterpret_cast<const short8*>(x + base + e0);
-  short8 v = {};
+  g.e0 = (2 * q - pad) * 3 + g.jj;
+  g.row_ok = m_ok && r < 7 && (unsigned)ih < (unsigned)H;
+  g.fast = g.row_ok && g.e0 >= 0 && g.e0 + 8 <= W3;
+  g.base = ((long)n * H + ih) * (long)W3;
+  g.src = g.fast ? x + g.base + g.e0 : zpage;
+}
+
+template <typename T16>
+DEV_INLINE short8 g8_fix(const G8<T16>& g, short8 v,
+                         const T16* __restrict__ x, int W3) {
+  if (g.fast) {
+    if (g.jj > 21 - 8) {
 #pragma unroll
-  for (int u = 0; u < 8; ++u) {
-    const int e = e0 + u;
-    if (jj + u < 21 && (unsigned)e < (unsigned)W3)
-      v[u] = *reinterpret_cast<const short*>(x + base + e);
+      for (int u = 0; u < 8; ++u)
+        if (g.jj + u >= 21) v[u] = 0;
+    }
+    return v;
   }
-  (void)last_safe;
-  return v;
+  short8 w = {};
+  if (g.row_ok) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const int e = g.e0 + u;
+      if (g.jj + u < 21 && (unsigned)e < (unsigned)W3)
+        w[u] = *reinterpret_cast<const short*>(x + g.base + e);
+    }
+  }
+  return w;
 }
 
 // ---- forward ----------------------------------------------------------
+// conv_gather_gemm anatomy: 128m x 64ko block, 3 k-chunks of 64, next
+// chunk's gather rides under the MFMA phase, single-buffered LDS.
 template <typename T16>
 __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
     const T16* __restrict__ x,    // [N, H, W, 3]
     const T16* __restrict__ w24,  // [KO, 192] row-padded
-    const float* __restrict__ bias, T16* __restrict__ y, const int N,
-    const int H, const int W, const int KO, const int Ho, const int Wo,
-    const int pad, const int act, const int has_bias) {
+    const float* __restrict__ bias, const T16* __restrict__ zpage,
+    T16* __restrict__ y, const int N, const int H, const int W,
+    const int KO, const int Ho, const int Wo, const int pad, const int act,
+    const int has_bias) {
   __shared__ T16 lds[128 * SLDK + 64 * SLDK];
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -88,9 +115,9 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
   const int k0 = blockIdx.y * 64;
   const int W3 = 3 * W;
 
-  // A staging: 2 threads per m-row, 96 k-elements (12 groups) each
+  // A staging: 2 threads per m-row, 32 k-elements (4 groups) each
   const int sa_m = tid >> 1;
-  const int sa_c = (tid & 1) * 96;
+  const int sa_c = (tid & 1) * 32;
   const long m_a = bm0 + sa_m;
   const bool m_ok = m_a < Mtot;
   int n_ = 0, p_ = 0, q_ = 0;
@@ -100,25 +127,9 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
     p_ = pq / Wo;
     q_ = pq % Wo;
   }
-  {
-    T16* pa = lds + sa_m * SLDK + sa_c;
-#pragma unroll
-    for (int i = 0; i < 12; ++i)
-      *reinterpret_cast<short8*>(reinterpret_cast<short*>(pa) + 8 * i) =
-          stem_gather8(x, n_, p_, q_, sa_c + 8 * i, H, W3, pad, m_ok, 0);
-  }
-  // B staging: flat 8-groups of the [64, 192] weight tile
-  {
-    T16* ldsB = lds + 128 * SLDK;
-    for (int e = tid * 8; e < 64 * SKG; e += 256 * 8) {
-      const int row = e / SKG, col = e - row * SKG;
-      *reinterpret_cast<short8*>(
-          reinterpret_cast<short*>(ldsB + row * SLDK + col)) =
-          *reinterpret_cast<const short8*>(w24 + (long)(k0 + row) * SKG +
-                                           col);
-    }
-  }
-  __syncthreads();
+  // B staging: 4 threads per ko-row, 16 k-elements each
+  const int sb_n = tid >> 2;
+  const int sb_c = (tid & 3) * 16;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -126,18 +137,52 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
   const int kh = lane >> 5;
   const int wm = wave * 32;
   f32x16_s acc[2] = {};
-  const T16* ldsA = lds;
-  const T16* ldsB = lds + 128 * SLDK;
+
+  short8 sa[4], sb[2];
+  auto load_step = [&](int j) {
+    G8<T16> g[4];
 #pragma unroll
-  for (int kk = 0; kk < SKG; kk += 16) {
-    const short8 af = *reinterpret_cast<const short8*>(
-        ldsA + (wm + li) * SLDK + kk + kh * 8);
-    const short8 b0 = *reinterpret_cast<const short8*>(
-        ldsB + li * SLDK + kk + kh * 8);
-    const short8 b1 = *reinterpret_cast<const short8*>(
-        ldsB + (32 + li) * SLDK + kk + kh * 8);
-    acc[0] = SMfma<T16>::run(af, b0, acc[0]);
-    acc[1] = SMfma<T16>::run(af, b1, acc[1]);
+    for (int i = 0; i < 4; ++i)
+      g8_plan(g[i], x, zpage, n_, p_, q_, j * 64 + sa_c + 8 * i, H, W3, pad,
+              m_ok);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      sa[i] = *reinterpret_cast<const short8*>(g[i].src);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) sa[i] = g8_fix(g[i], sa[i], x, W3);
+    const T16* wp = w24 + (long)(k0 + sb_n) * SKG + j * 64 + sb_c;
+    sb[0] = *reinterpret_cast<const short8*>(wp);
+    sb[1] = *reinterpret_cast<const short8*>(wp + 8);
+  };
+  auto stage = [&]() {
+    short* pa = reinterpret_cast<short*>(lds + sa_m * SLDK + sa_c);
+#pragma unroll
+    for (int i = 0; i < 4; ++i) *reinterpret_cast<short8*>(pa + 8 * i) = sa[i];
+    short* pb =
+        reinterpret_cast<short*>(lds + 128 * SLDK + sb_n * SLDK + sb_c);
+    *reinterpret_cast<short8*>(pb) = sb[0];
+    *reinterpret_cast<short8*>(pb + 8) = sb[1];
+  };
+
+  load_step(0);
+  for (int j = 0; j < SKG / 64; ++j) {
+    __syncthreads();
+    stage();
+    __syncthreads();
+    if (j + 1 < SKG / 64) load_step(j + 1);
+    const T16* ldsA = lds;
+    const T16* ldsB = lds + 128 * SLDK;
+#pragma unroll
+    for (int kk = 0; kk < 64; kk += 16) {
+      const short8 af = *reinterpret_cast<const short8*>(
+          ldsA + (wm + li) * SLDK + kk + kh * 8);
+      const short8 b0 = *reinterpret_cast<const short8*>(
+          ldsB + li * SLDK + kk + kh * 8);
+      const short8 b1 = *reinterpret_cast<const short8*>(
+          ldsB + (32 + li) * SLDK + kk + kh * 8);
+      acc[0] = SMfma<T16>::run(af, b0, acc[0]);
+      acc[1] = SMfma<T16>::run(af, b1, acc[1]);
+    }
   }
 
   float bv[2];
@@ -165,6 +210,7 @@ template <typename T16>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     const T16* __restrict__ x,   // [N, H, W, 3]
     const T16* __restrict__ dy,  // [M, KO]
+    const T16* __restrict__ zpage,
     float* __restrict__ dw,      // chunk slabs of [KO*3*7*7]
     const int N, const int H, const int W, const int KO, const int Ho,
     const int Wo, const int pad, const long m_per_chunk) {
@@ -226,12 +272,13 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
       }
     }
     if (do_x) {
+      G8<T16> g[4];
       int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const long m = m0 + smx + mi;
-        vx[mi] = stem_gather8<T16>(x, n_, p_, q_, c0 + skx, H, W3, pad,
-                                   m < m_end, 0);
+        g8_plan(g[mi], x, zpage, n_, p_, q_, c0 + skx, H, W3, pad,
+                m < m_end);
         if (mi < 3 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -240,6 +287,11 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
           }
         }
       }
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+        vx[mi] = *reinterpret_cast<const short8*>(g[mi].src);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) vx[mi] = g8_fix(g[mi], vx[mi], x, W3);
       advance(64);
     }
   };
@@ -301,6 +353,8 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
 
 // host launchers --------------------------------------------------------
 
+at::Tensor conv_zero_page(const at::Tensor& like);  // conv_mfma.hip
+
 // w: [KO, KGP] tap-major padded ([kg = (r*7+s)*3+c]) -> [KO, 192]
 // row-padded (kg = r*24 + s*3 + c)
 static at::Tensor stem_w24(const at::Tensor& w) {
@@ -318,6 +372,7 @@ void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int Ho = y.size(1), Wo = y.size(2);
   TORCH_CHECK(KO % 64 == 0, "stem GEMM expects KO % 64 == 0");
   auto w24 = stem_w24(w);
+  at::Tensor zp = conv_zero_page(x);
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)cdiv_l(M, 128), KO / 64);
   const int has_bias = bias.numel() > 0;
@@ -326,8 +381,8 @@ void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                        cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)w24.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
-                       (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo, (int)pad,
-                       (int)act, has_bias);
+                       (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N, H,
+                       W, KO, Ho, Wo, (int)pad, (int)act, has_bias);
   });
 }
 
@@ -345,12 +400,14 @@ void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
   nchunks = cdiv_l(M, m_per_chunk);
   auto part = at::empty({nchunks * E}, x.options().dtype(at::kFloat));
+  at::Tensor zp = conv_zero_page(x);
   dim3 grid(KO / 64, SKG / 64, (unsigned)nchunks);
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL((conv_wgrad_stem_gemm<T16>), grid, dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
-                       (const T16*)dy.data_ptr(), part.data_ptr<float>(), N,
-                       H, W, KO, Ho, Wo, (int)pad, m_per_chunk);
+                       (const T16*)dy.data_ptr(), (const T16*)zp.data_ptr(),
+                       part.data_ptr<float>(), N, H, W, KO, Ho, Wo, (int)pad,
+                       m_per_chunk);
   });
   wgrad_reduce_launch(part, dw, E, nchunks);
 }
