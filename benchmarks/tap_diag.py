@@ -1,5 +1,7 @@
 """Diagnose the r18 tap-parity gap: noise floor (two TAP=0 runs) vs the
 TAP=1 delta, worst element and owning parameter."""
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import os
 import torch
 from mi355x.models import build_model

@@ -20,6 +20,13 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                         long stride, long pad);
 std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
                                          long stride, long pad);
+std::vector<at::Tensor> conv2d_fwd_scaled(at::Tensor x, at::Tensor w,
+                                          at::Tensor asc, at::Tensor ash,
+                                          long stride, long pad,
+                                          bool want_stats);
+at::Tensor conv2d_wgrad_scaled(at::Tensor x, at::Tensor asc, at::Tensor ash,
+                               at::Tensor dy, long R, long S, long stride,
+                               long pad);
 
 at::Tensor bn_stats(at::Tensor x);
 at::Tensor bn_finalize(at::Tensor stats, at::Tensor running_mean,
@@ -30,12 +37,13 @@ std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean, at::Tensor invst
                     long act, bool want_mask);
 at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
                          at::Tensor mean, at::Tensor invstd,
-                         at::Tensor mask);
+                         at::Tensor mask, at::Tensor asc, at::Tensor ash);
 std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
                                   at::Tensor mean, at::Tensor invstd,
                                   at::Tensor gamma, at::Tensor dgamma,
                                   at::Tensor dbeta, double m_total,
-                                  bool want_dres, at::Tensor mask);
+                                  bool want_dres, at::Tensor mask,
+                                  at::Tensor asc, at::Tensor ash);
 
 std::vector<at::Tensor> maxpool_fwd(at::Tensor x, long kernel, long stride,
                                     long pad);
@@ -69,6 +77,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_dgrad", &conv2d_dgrad);
   m.def("conv2d_wgrad", &conv2d_wgrad);
   m.def("conv2d_fwd_stats", &conv2d_fwd_stats);
+  m.def("conv2d_fwd_scaled", &conv2d_fwd_scaled);
+  m.def("conv2d_wgrad_scaled", &conv2d_wgrad_scaled);
   m.def("bn_stats", &bn_stats);
   m.def("bn_finalize", &bn_finalize);
   m.def("bn_apply", &bn_apply);

@@ -124,22 +124,29 @@ __global__ void bn_bwd_dx_kernel(const T16* __restrict__ x,
 // accumulator, then one global atomicAdd per channel per block.
 // BWD=false: (sum x, sum x^2). BWD=true: (sum dy*xhat, sum dy), with the
 // fused ReLU mask from y when given.
+// asc/ash (BWD only, nullable): recompute the ReLU mask as
+// (x*asc[c] + ash[c]) > 0 — the lazy-BN path (apply fused into the
+// consuming conv) never materializes y or a mask
 template <typename T16, bool BWD>
 __global__ __launch_bounds__(256) void bn_reduce_fast(
     const T16* __restrict__ x, const T16* __restrict__ dy,
     const unsigned char* __restrict__ msk, const float* __restrict__ mean,
-    const float* __restrict__ invstd, float* __restrict__ out,  // [2,C]
+    const float* __restrict__ invstd,
+    const float* __restrict__ asc, const float* __restrict__ ash,
+    float* __restrict__ out,  // [2,C]
     long E, int C, long e_per_block) {
   extern __shared__ __attribute__((aligned(16))) float lsum[];  // [2*C]
   const int tid = threadIdx.x;
   const int c0 = (tid * 8) % C;
   float s0[8] = {}, s1[8] = {};
-  float mu[8], is[8];
+  float mu[8], is[8], ac[8], ah[8];
   if (BWD) {
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       mu[u] = mean[c0 + u];
       is[u] = invstd[c0 + u];
+      ac[u] = asc ? asc[c0 + u] : 0.f;
+      ah[u] = ash ? ash[c0 + u] : 0.f;
     }
   }
   const long e0 = (long)blockIdx.x * e_per_block;
@@ -163,8 +170,13 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
 #pragma unroll
       for (int u = 0; u < 8; ++u) {
         float d = s16_to_f32<T16>(vd[u]);
-        if (!((mb >> u) & 1)) d = 0.f;
-        const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
+        const float xv = s16_to_f32<T16>(vx[u]);
+        if (asc) {
+          if (xv * ac[u] + ah[u] <= 0.f) d = 0.f;
+        } else if (!((mb >> u) & 1)) {
+          d = 0.f;
+        }
+        const float xh = (xv - mu[u]) * is[u];
         s0[u] += d * xh;
         s1[u] += d;
       }
@@ -260,10 +272,11 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     const unsigned char* __restrict__ msk, const float* __restrict__ mean,
     const float* __restrict__ invstd, const float* __restrict__ gamma,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
+    const float* __restrict__ asc, const float* __restrict__ ash,
     T16* __restrict__ dx, T16* __restrict__ dres, long E, int C,
     float inv_m) {
   const int c0 = ((blockIdx.x * blockDim.x + threadIdx.x) * 8) % C;
-  float mu[8], is[8], g_[8], a_[8], b_[8];
+  float mu[8], is[8], g_[8], a_[8], b_[8], ac[8], ah[8];
 #pragma unroll
   for (int u = 0; u < 8; ++u) {
     mu[u] = mean[c0 + u];
@@ -271,6 +284,8 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     g_[u] = gamma[c0 + u] * is[u];
     a_[u] = dgamma[c0 + u] * inv_m;
     b_[u] = dbeta[c0 + u] * inv_m;
+    ac[u] = asc ? asc[c0 + u] : 0.f;
+    ah[u] = ash ? ash[c0 + u] : 0.f;
   }
   const long stride = (long)gridDim.x * blockDim.x * 8;
   auto body = [&](long e) {
@@ -281,9 +296,14 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       float d = s16_to_f32<T16>(vd[u]);
-      if (!((mb >> u) & 1)) d = 0.f;
+      const float xv = s16_to_f32<T16>(vx[u]);
+      if (asc) {
+        if (xv * ac[u] + ah[u] <= 0.f) d = 0.f;
+      } else if (!((mb >> u) & 1)) {
+        d = 0.f;
+      }
       if (dres) ods[u] = f32_to_s16<T16>(d);
-      const float xh = (s16_to_f32<T16>(vx[u]) - mu[u]) * is[u];
+      const float xh = (xv - mu[u]) * is[u];
       odx[u] = f32_to_s16<T16>(g_[u] * (d - b_[u] - xh * a_[u]));
     }
     *reinterpret_cast<short8*>(dx + e) = odx;
@@ -380,7 +400,8 @@ at::Tensor bn_stats(at::Tensor x) {
       hipLaunchKernelGGL((bn_reduce_fast<T16, false>), grid, dim3(256),
                          2 * C * sizeof(float), cur_stream(),
                          (const T16*)x.data_ptr(), nullptr, nullptr, nullptr,
-                         nullptr, out.data_ptr<float>(), E, C, e_per_block);
+                         nullptr, nullptr, nullptr, out.data_ptr<float>(), E,
+                         C, e_per_block);
     });
     return out;
   }
@@ -442,9 +463,14 @@ std::vector<at::Tensor> bn_apply(at::Tensor x, at::Tensor mean,
   return {y, at::Tensor()};
 }
 
+// asc/ash (optional [C] fp32): lazy-BN path — recompute the ReLU mask
+// as (x*asc+ash) > 0 instead of reading a stored mask or y
 at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
                          at::Tensor mean, at::Tensor invstd,
-                         at::Tensor mask) {
+                         at::Tensor mask, at::Tensor asc, at::Tensor ash) {
+  TORCH_CHECK(!(asc.defined() && asc.numel() > 0) ||
+                  bn_fast_ok(x.numel() / x.size(-1), x.size(-1)),
+              "mask recompute (asc/ash) needs the fast BN path");
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
@@ -462,6 +488,8 @@ at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
                          mask.numel() ? mask.data_ptr<unsigned char>()
                                       : nullptr,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                         asc.numel() ? asc.data_ptr<float>() : nullptr,
+                         ash.numel() ? ash.data_ptr<float>() : nullptr,
                          out.data_ptr<float>(), E, C, e_per_block);
     });
     return out;
@@ -486,7 +514,11 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
                                   at::Tensor mean, at::Tensor invstd,
                                   at::Tensor gamma, at::Tensor dgamma,
                                   at::Tensor dbeta, double m_total,
-                                  bool want_dres, at::Tensor mask) {
+                                  bool want_dres, at::Tensor mask,
+                                  at::Tensor asc, at::Tensor ash) {
+  TORCH_CHECK(!(asc.defined() && asc.numel() > 0) ||
+                  bn_fast_ok(x.numel() / x.size(-1), x.size(-1)),
+              "mask recompute (asc/ash) needs the fast BN path");
   CHECK_GPU(x);
   CHECK_CONTIG(dy);
   const int C = x.size(-1);
@@ -504,7 +536,10 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
                                       : nullptr,
                          mean.data_ptr<float>(), invstd.data_ptr<float>(),
                          gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
-                         dbeta.data_ptr<float>(), (T16*)dx.data_ptr(),
+                         dbeta.data_ptr<float>(),
+                         asc.numel() ? asc.data_ptr<float>() : nullptr,
+                         ash.numel() ? ash.data_ptr<float>() : nullptr,
+                         (T16*)dx.data_ptr(),
                          want_dres ? (T16*)dres.data_ptr() : nullptr, E, C,
                          (float)(1.0 / m_total));
     });

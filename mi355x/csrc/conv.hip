@@ -820,7 +820,7 @@ at::Tensor conv_zero_page(const at::Tensor& like);
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act,
-                          at::Tensor stats);
+                          at::Tensor stats, at::Tensor asc, at::Tensor ash);
 void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                                at::Tensor y, long R, long S, long stride,
                                long pad, long act);
@@ -828,7 +828,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                             long R, long S, long stride, long pad,
                             at::Tensor addin);
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
-                            long R, long S, long stride, long pad);
+                            long R, long S, long stride, long pad,
+                            at::Tensor asc, at::Tensor ash);
 // stem_mfma.hip — MFMA GEMM stem (7x7/s2/C=3), replaces the dot2 kernels
 void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                                at::Tensor y, long pad, long act);
@@ -939,7 +940,8 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   }
   TORCH_CHECK(w.size(3) == C, "conv weight/input channel mismatch");
   if (conv_mfma_supported(C, K)) {
-    conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act, at::Tensor());
+    conv_fwd_mfma_launch(x, w, bias, y, stride, pad, act, at::Tensor(),
+                         at::Tensor(), at::Tensor());
     return y;
   }
   const long total = (long)N * P * Q * K;
@@ -978,13 +980,58 @@ std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
   auto y = at::empty({N, P, Q, K}, x.options());
   auto stats = at::zeros({2, K}, x.options().dtype(at::kFloat));
   auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
-  conv_fwd_mfma_launch(x, w, empty_bias, y, stride, pad, 0, stats);
+  conv_fwd_mfma_launch(x, w, empty_bias, y, stride, pad, 0, stats,
+                       at::Tensor(), at::Tensor());
   return {y, stats};
 }
 
 // addin (optional, dx-shaped): dx = dgrad(dy) + addin — carries the
 // residual-junction gradient of a ResNet block so autograd's separate
 // full-tensor add at the junction disappears (MFMA path only).
+// lazy-BN fused entry: z = relu(x*asc + ash) applied to the x operand on
+// load (the BN apply pass and its output tensor never materialize);
+// returns {y, stats[2,KO] or empty} (stats when want_stats, from the conv
+// epilogue — feeds the NEXT BN).
+std::vector<at::Tensor> conv2d_fwd_scaled(at::Tensor x, at::Tensor w,
+                                          at::Tensor asc, at::Tensor ash,
+                                          long stride, long pad,
+                                          bool want_stats) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_16BIT(x);
+  CHECK_CONTIG(w);
+  const int N = x.size(0), H = x.size(1), W_ = x.size(2), C = x.size(3);
+  const int K = w.size(0);
+  TORCH_CHECK(w.dim() == 4 && conv_mfma_supported(C, K),
+              "conv2d_fwd_scaled needs the MFMA path (C,K % 64 == 0)");
+  const int R = w.size(1), S = w.size(2);
+  const int P = out_dim(H, R, stride, pad), Q = out_dim(W_, S, stride, pad);
+  auto y = at::empty({N, P, Q, K}, x.options());
+  auto stats = want_stats
+                   ? at::zeros({2, K}, x.options().dtype(at::kFloat))
+                   : at::Tensor();
+  auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
+  conv_fwd_mfma_launch(x, w, empty_bias, y, stride, pad, 0, stats, asc, ash);
+  return {y, want_stats ? stats : at::Tensor()};
+}
+
+// lazy-BN wgrad: the x operand transformed on load like conv2d_fwd_scaled
+at::Tensor conv2d_wgrad_scaled(at::Tensor x, at::Tensor asc, at::Tensor ash,
+                               at::Tensor dy, long R, long S, long stride,
+                               long pad) {
+  CHECK_GPU(x);
+  CHECK_CONTIG(x);
+  CHECK_CONTIG(dy);
+  const int C = x.size(3);
+  const int K = dy.size(3);
+  TORCH_CHECK(conv_mfma_supported(C, K),
+              "conv2d_wgrad_scaled needs the MFMA path (C,K % 64 == 0)");
+  auto dw = at::empty({(long)K, (long)C, R, S},
+                      x.options().dtype(at::kFloat));
+  conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad, asc, ash);
+  return dw;
+}
+
 at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
                         long pad, long H, long W, at::Tensor addin) {
   CHECK_GPU(dy);
@@ -1023,7 +1070,8 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     // empty, not zeros: the MFMA wgrad tiles cover every [K,C,R,S] element
     // with plain stores (nchunks==1) or via wgrad_reduce_chunks (nchunks>1)
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
-    conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad);
+    conv_wgrad_mfma_launch(x, dy, dw, R, S, stride, pad, at::Tensor(),
+                           at::Tensor());
     return dw;
   }
   if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&

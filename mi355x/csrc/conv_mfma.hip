@@ -35,6 +35,21 @@ struct Mfma32<__half> {
   }
 };
 
+// lazy-BN A-side transform: z = relu(x*asc[c] + ash[c]) applied to an
+// 8-channel group in registers at load/stage time (the normalized
+// activation is never materialized; VALU rides under the MFMA phase)
+template <typename T16>
+DEV_INLINE short8 scale8(short8 v, const float* __restrict__ asc,
+                         const float* __restrict__ ash, int c0) {
+  short8 o;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const float f = s16_to_f32<T16>(v[u]) * asc[c0 + u] + ash[c0 + u];
+    o[u] = f32_to_s16<T16>(fmaxf(f, 0.f));
+  }
+  return o;
+}
+
 constexpr int BM = 128, BN = 64, BK = 64;
 constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
 // NT = 32-wide N(output-channel) tiles per block: 2 (BN=64) or 4 (BN=128);
@@ -48,7 +63,8 @@ constexpr int LDK = BK + 8;  // +16B pad: spreads fragment reads over banks
 // ADDIN (compile-time — a runtime addin branch in this shared epilogue
 // measurably slowed the non-addin convs): out += addin elementwise, the
 // residual-junction gradient fused into the dgrad epilogue.
-template <typename T16, bool TRANS, bool GENC, int NT, bool ADDIN = false>
+template <typename T16, bool TRANS, bool GENC, int NT, bool ADDIN = false,
+          bool SCALED = false>
 // min-blocks hint: same effect as on the wgrad kernel (see comment there)
 __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const T16* __restrict__ in,    // [N, Hi, Wi, CI]
@@ -56,6 +72,8 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const float* __restrict__ bias,  // [KO] or null
     const T16* __restrict__ zpage,   // >=256 zero elements (OOB gather target)
     const T16* __restrict__ addin,   // ADDIN only: same layout as out
+    const float* __restrict__ asc,   // SCALED only: [CI] scale
+    const float* __restrict__ ash,   // SCALED only: [CI] shift
     T16* __restrict__ out,         // [N*Ho*Wo, KO]
     float* __restrict__ stats_slab,  // null, or [gy][gx][2][BNT] partial
     const int N, const int Hi, const int Wi, const int CI, const int KO,  //  (sum,sumsq) of this block's output tile (conv->BN fusion)
@@ -176,6 +194,13 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
 #pragma unroll
     for (int i = 0; i < 4; ++i)
       sa[i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
+    if constexpr (SCALED) {
+      if (va) {
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          sa[i] = scale8<T16>(sa[i], asc, ash, c0 + sa_c + 8 * i);
+      }
+    }
     const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
@@ -354,12 +379,14 @@ void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
 // ---------------------------------------------------------------------------
 constexpr int PCS = 72;  // patch col stride (elements)
 
-template <typename T16, bool TRANS, bool ADDIN = false>
+template <typename T16, bool TRANS, bool ADDIN = false, bool SCALED = false>
 __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
     const T16* __restrict__ in,   // [N, Hi, Wi, CI] (dgrad: dy, CI=KO)
     const T16* __restrict__ wgt,  // fwd: [KO, 9*CI]; dgrad: wflip strides
     const float* __restrict__ bias,
     const T16* __restrict__ addin,  // ADDIN only: out += addin
+    const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
+    const float* __restrict__ ash,  // SCALED only: [CI] lazy-BN shift
     T16* __restrict__ out,
     float* __restrict__ stats_slab,  // null, or per-block (sum,sumsq) rows
     const int N, const int Hi, const int Wi, const int CI, const int KO,
@@ -434,6 +461,8 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
         const long nn = gr / Ho, pp = gr % Ho;
         v = *reinterpret_cast<const short8*>(
             in + ((nn * Hi + pp) * Wi + iw) * CI + cc * BK + g * 8);
+        if constexpr (SCALED)
+          v = scale8<T16>(v, asc, ash, cc * BK + g * 8);
       }
       *reinterpret_cast<short8*>(patch + ((long)row * (Wi + 2) + col) * PCS +
                                  g * 8) = v;
@@ -541,7 +570,7 @@ constexpr int LDM = 72;   // m-minor row length (+16B: alignment + banks)
 
 // KT = KO-tile width (64 or 128): wider k-tiles amortize the x staging
 // and double the MFMA work per m-step for the K>=128 layers.
-template <typename T16, int KT>
+template <typename T16, int KT, bool SCALED = false>
 // the second launch-bounds arg (min 2 blocks/CU) is load-bearing: without
 // it the compiler allocates 188-256 VGPRs (occupancy 1-2) for no speedup;
 // with it 104-124 VGPRs, zero spills, occupancy 4 - the extra waves are
@@ -549,6 +578,8 @@ template <typename T16, int KT>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
+    const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
+    const float* __restrict__ ash,
     float* __restrict__ dw,       // [KO, R*S*CI]
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
@@ -634,6 +665,9 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
                           x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 +
                           skx)
                     : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[mi] = scale8<T16>(vx[mi], asc, ash, c0 + skx);
+        }
         if (mi < 3 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -712,9 +746,11 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
 // dy+x stage feeds 16 MFMAs per wave per m-step instead of 8, and each
 // fragment read is reused twice. Same transposed-staging/split-M/slab
 // machinery as conv_wgrad_mfma_kernel.
-template <typename T16>
+template <typename T16, bool SCALED = false>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
     const T16* __restrict__ x, const T16* __restrict__ dy,
+    const float* __restrict__ asc,  // SCALED only
+    const float* __restrict__ ash,
     float* __restrict__ dw,  // chunk slabs of [KO, R*S*CI]
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const int R, const int S, const int stride,
@@ -779,6 +815,9 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
                       (unsigned)iw < (unsigned)Wi;
       const T16* xp = x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
       vx0[mi] = ok ? *reinterpret_cast<const short8*>(xp) : short8{};
+      if constexpr (SCALED) {
+        if (ok) vx0[mi] = scale8<T16>(vx0[mi], asc, ash, c0 + sk);
+      }
       if (mi < 3 && ++q_ == Wo) {
         q_ = 0;
         if (++p_ == Ho) {
@@ -853,10 +892,12 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
 // slots that cross a q-row boundary (q==0 for s=0, q==Wo-1 for s=2).
 // Triples the MFMA work per staged dy/x byte — the plain kernel was
 // staging-bound at ~190 TF while the fwd gather-GEMM reaches 470-670.
-template <typename T16, int KT>
+template <typename T16, int KT, bool SCALED = false>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
+    const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
+    const float* __restrict__ ash,
     float* __restrict__ dw,       // chunk slabs of [KO, 3*3*CI]
     const int N, const int Hi, const int Wi, const int CI, const int KO,
     const int Ho, const int Wo, const long m_per_chunk, const int nchunks) {
@@ -938,6 +979,9 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
                          x + (((long)n_ * Hi + ih) * Wi + (q_ - 1)) * CI +
                          c0 + skx)
                    : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[0] = scale8<T16>(vx[0], asc, ash, c0 + skx);
+        }
       }
 #pragma unroll
       for (int i = 1; i < 6; ++i) {
@@ -949,6 +993,9 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
                          x + (((long)n_ * Hi + ih) * Wi + q_) * CI + c0 +
                          skx)
                    : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[i] = scale8<T16>(vx[i], asc, ash, c0 + skx);
+        }
         if (i < 5 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -1425,8 +1472,14 @@ void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz) {
                      dw.data_ptr<float>(), E, (int)nz);
 }
 
+// asc/ash (optional): lazy-BN transform on the x operand (see fwd launcher)
 void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
-                            long R, long S, long stride, long pad) {
+                            long R, long S, long stride, long pad,
+                            at::Tensor asc, at::Tensor ash) {
+  const float* asc_p = asc.defined() && asc.numel() ? asc.data_ptr<float>()
+                                                    : nullptr;
+  const float* ash_p = ash.defined() && ash.numel() ? ash.data_ptr<float>()
+                                                    : nullptr;
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   const long M = (long)N * Ho * Wo;
@@ -1438,7 +1491,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     const char* e = getenv("MI355X_WGRAD_HALO");
     return e && e[0] == '1';
   }();
-  const bool halo = halo_on && R == 3 && S == 3 && stride == 1 && pad == 1 &&
+  const bool halo = halo_on && !asc_p && R == 3 && S == 3 && stride == 1 && pad == 1 &&
                     Hi == Ho && Wi == Wo && Wo >= 8 && Wo <= 64 &&
                     (Wo & (Wo - 1)) == 0 && ((long)Ho * Wo) % 64 == 0;
   if (halo) {
@@ -1485,7 +1538,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     }();
     // KT=64 default even when KO%128==0: occupancy 3 vs 2 outweighs the
     // wider tile's staging amortization (measured +3.4% whole-bench)
-    const int KT = ktf == 128 ? 128 : 64;
+    const int KT = (ktf == 128 && !asc_p) ? 128 : 64;  // scaled: KT=64 only
     dim3 grid(KO / KT, 3 * (CI / 64), nchunks);
     const long E = (long)KO * 9 * CI;
     at::Tensor part = nchunks > 1
@@ -1495,12 +1548,21 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
       if (KT == 128)
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 128>), grid, dim3(256),
                            0, cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
+      else if (asc_p)
+        hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
       else
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64>), grid, dim3(256),
                            0, cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), part.data_ptr<float>(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
     });
     if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
@@ -1524,11 +1586,21 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                           ? at::empty({nchunks, E}, dw.options())
                           : dw;
     DISPATCH_16(x, T16, {
-      hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16>), grid, dim3(256), 0,
-                         cur_stream(), (const T16*)x.data_ptr(),
-                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
-                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
-                         (int)stride, (int)pad, m_per_chunk, nchunks);
+      if (asc_p)
+        hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else
+        hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16>), grid, dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
     });
     if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
     return;
@@ -1554,18 +1626,41 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                         ? at::empty({nchunks, E}, dw.options())
                         : dw;
   DISPATCH_16(x, T16, {
-    if (KT == 128)
-      hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128>), grid, dim3(256),
-                         0, cur_stream(), (const T16*)x.data_ptr(),
-                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
-                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
-                         (int)stride, (int)pad, m_per_chunk, nchunks);
-    else
-      hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64>), grid, dim3(256),
-                         0, cur_stream(), (const T16*)x.data_ptr(),
-                         (const T16*)dy.data_ptr(), part.data_ptr<float>(),
-                         N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
-                         (int)stride, (int)pad, m_per_chunk, nchunks);
+    if (KT == 128) {
+      if (asc_p)
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+    } else {
+      if (asc_p)
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+    }
   });
   if (nchunks > 1) wgrad_reduce_launch(part, dw, E, nchunks);
 }
@@ -1587,7 +1682,8 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
                        dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)wpad.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
-                       (const T16*)zp.data_ptr(), nullptr,
+                       (const T16*)zp.data_ptr(), nullptr, nullptr,
+                       nullptr,
                        (T16*)y.data_ptr(), nullptr, N, Hi, Wi, CI, KO, Ho,
                        Wo, (int)R, (int)S, (int)stride, (int)pad, KGP, 0,
                        (int)act, has_bias);
@@ -1598,9 +1694,16 @@ void conv_fwd_mfma_genc_launch(at::Tensor x, at::Tensor wpad, at::Tensor bias,
 // stats (optional, zeroed [2,C]): conv->BN fusion — the epilogue emits
 // per-block (sum,sumsq) partials to a slab and conv_stats_reduce folds
 // them, replacing BN's separate full-tensor bn_stats pass.
+// asc/ash (optional [CI] fp32): lazy-BN A-side transform — the x operand
+// is normalized+ReLU'd on load (z = relu(x*asc+ash)); the BN apply pass
+// and its output tensor disappear.
 void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                           at::Tensor y, long stride, long pad, long act,
-                          at::Tensor stats) {
+                          at::Tensor stats, at::Tensor asc, at::Tensor ash) {
+  const float* asc_p = asc.defined() && asc.numel() ? asc.data_ptr<float>()
+                                                    : nullptr;
+  const float* ash_p = ash.defined() && ash.numel() ? ash.data_ptr<float>()
+                                                    : nullptr;
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int KO = w.size(0), R = w.size(1), S = w.size(2);
   const int Ho = y.size(1), Wo = y.size(2);
@@ -1614,7 +1717,7 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     const char* e = getenv("MI355X_CONV_GLDS");
     return e ? atoi(e) : 0;
   }();
-  if (glds_mode && !(stats.defined() && stats.numel() > 0)) {
+  if (glds_mode && !asc_p && !(stats.defined() && stats.numel() > 0)) {
     at::Tensor zp = conv_zero_page(x);
     const bool wide = KO % 128 == 0 && cdiv_l(M, BM) * (KO / 128) >= 1024;
     dim3 grid((unsigned)cdiv_l(M, BM), KO / (wide ? 128 : 64));
@@ -1660,13 +1763,22 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
       pslab_ptr = pslab.data_ptr<float>();
     }
     DISPATCH_16(x, T16, {
-      hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
-                         smem, cur_stream(), (const T16*)x.data_ptr(),
-                         (const T16*)w.data_ptr(),
-                         has_bias ? bias.data_ptr<float>() : nullptr,
-                         nullptr, (T16*)y.data_ptr(), pslab_ptr, N, Hi, Wi,
-                         CI, KO,
-                         (long)R * S * CI, (long)CI, (int)act, has_bias, NR);
+      if (asc_p)
+        hipLaunchKernelGGL((conv_patch_gemm<T16, false, false, true>),
+                           pgrid_, dim3(256), smem, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           nullptr, asc_p, ash_p, (T16*)y.data_ptr(),
+                           pslab_ptr, N, Hi, Wi, CI, KO, (long)R * S * CI,
+                           (long)CI, (int)act, has_bias, NR);
+      else
+        hipLaunchKernelGGL((conv_patch_gemm<T16, false>), pgrid_, dim3(256),
+                           smem, cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           nullptr, nullptr, nullptr, (T16*)y.data_ptr(),
+                           pslab_ptr, N, Hi, Wi, CI, KO, (long)R * S * CI,
+                           (long)CI, (int)act, has_bias, NR);
     });
     if (pslab_ptr)
       stats_slab_reduce(pslab, stats, pgrid_.x, 128, KO);
@@ -1686,26 +1798,49 @@ void conv_fwd_mfma_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     stats_slab_ptr = slab.data_ptr<float>();
   }
   DISPATCH_16(x, T16, {
-    if (wide)
-      hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4>), grid,
-                         dim3(256), 0, cur_stream(),
-                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
-                         has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(), nullptr,
-                         (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
-                         KO, Ho, Wo, R, S,
-                         (int)stride, (int)pad, (long)R * S * CI, (long)CI,
-                         (int)act, has_bias);
-    else
-      hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 2>), grid,
-                         dim3(256), 0, cur_stream(),
-                         (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
-                         has_bias ? bias.data_ptr<float>() : nullptr,
-                         (const T16*)zp2.data_ptr(), nullptr,
-                         (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
-                         KO, Ho, Wo, R, S,
-                         (int)stride, (int)pad, (long)R * S * CI, (long)CI,
-                         (int)act, has_bias);
+    if (wide) {
+      if (asc_p)
+        hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4, false,
+                                             true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp2.data_ptr(), nullptr, asc_p, ash_p,
+                           (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                           KO, Ho, Wo, R, S, (int)stride, (int)pad,
+                           (long)R * S * CI, (long)CI, (int)act, has_bias);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 4>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp2.data_ptr(), nullptr, nullptr,
+                           nullptr,
+                           (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                           KO, Ho, Wo, R, S, (int)stride, (int)pad,
+                           (long)R * S * CI, (long)CI, (int)act, has_bias);
+    } else {
+      if (asc_p)
+        hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 2, false,
+                                             true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp2.data_ptr(), nullptr, asc_p, ash_p,
+                           (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                           KO, Ho, Wo, R, S, (int)stride, (int)pad,
+                           (long)R * S * CI, (long)CI, (int)act, has_bias);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<T16, false, false, 2>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(), (const T16*)w.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (const T16*)zp2.data_ptr(), nullptr, nullptr,
+                           nullptr,
+                           (T16*)y.data_ptr(), stats_slab_ptr, N, Hi, Wi, CI,
+                           KO, Ho, Wo, R, S, (int)stride, (int)pad,
+                           (long)R * S * CI, (long)CI, (int)act, has_bias);
+    }
   });
   if (stats_slab_ptr)
     stats_slab_reduce(slab, stats, grid.x, bnt2, KO);
@@ -1737,13 +1872,14 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                            dim3(256), smem, cur_stream(),
                            (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr,
-                           (const T16*)addin.data_ptr(),
+                           (const T16*)addin.data_ptr(), nullptr, nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
                            (long)KO, (long)CI * KO, 0, 0, NR);
       else
         hipLaunchKernelGGL((conv_patch_gemm<T16, true>), pgrid_, dim3(256),
                            smem, cur_stream(), (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr, nullptr,
+                           nullptr, nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI,
                            (long)KO, (long)CI * KO, 0, 0, NR);
     });
@@ -1760,7 +1896,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                            grid, dim3(256), 0, cur_stream(),
                            (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr,
-                           (const T16*)zp.data_ptr(), add_p,
+                           (const T16*)zp.data_ptr(), add_p, nullptr,
+                           nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
                            W, (int)R, (int)S, (int)stride, (int)pad,
                            (long)KO, (long)CI * KO, 0, 0);
@@ -1769,7 +1906,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                            dim3(256), 0, cur_stream(),
                            (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr,
-                           (const T16*)zp.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), nullptr, nullptr,
+                           nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
                            W, (int)R, (int)S, (int)stride, (int)pad,
                            (long)KO, (long)CI * KO, 0, 0);
@@ -1779,7 +1917,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                            grid, dim3(256), 0, cur_stream(),
                            (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr,
-                           (const T16*)zp.data_ptr(), add_p,
+                           (const T16*)zp.data_ptr(), add_p, nullptr,
+                           nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
                            W, (int)R, (int)S, (int)stride, (int)pad,
                            (long)KO, (long)CI * KO, 0, 0);
@@ -1788,7 +1927,8 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
                            dim3(256), 0, cur_stream(),
                            (const T16*)dy.data_ptr(),
                            (const T16*)wflip.data_ptr(), nullptr,
-                           (const T16*)zp.data_ptr(), nullptr,
+                           (const T16*)zp.data_ptr(), nullptr, nullptr,
+                           nullptr,
                            (T16*)dx.data_ptr(), nullptr, N, P, Q, KO, CI, H,
                            W, (int)R, (int)S, (int)stride, (int)pad,
                            (long)KO, (long)CI * KO, 0, 0);

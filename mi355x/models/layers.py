@@ -112,12 +112,14 @@ class BatchNorm2d(nn.Module):
         if key in unexpected_keys:
             unexpected_keys.remove(key)
 
-    def forward(self, x, residual=None):
+    def forward(self, x, residual=None, stats=None):
         if self.training:
             self._nbt += 1
         return ops.batch_norm(x, self.weight, self.bias, self.running_mean,
                               self.running_var, self.training, self.momentum,
-                              residual, self.act, self.process_group if self.training else None)
+                              residual, self.act,
+                              self.process_group if self.training else None,
+                              stats)
 
 
 import os
@@ -158,6 +160,42 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
 
 def _tap_on():
     return os.environ.get("MI355X_TAP", "1") != "0"
+
+
+def _lazy_on():
+    return os.environ.get("MI355X_LAZY_BN", "1") != "0"
+
+
+def _lazy_ok(bn: "BatchNorm2d", conv: "Conv2d", x) -> bool:
+    """Lazy-BN (apply fused into the consuming conv) eligibility: GPU
+    training, relu BN with no residual, MFMA conv shapes, fast-BN channel
+    layout (C % 8 == 0 and 2048 % C == 0 for the mask-recompute backward)."""
+    C = conv.weight.shape[1]
+    return (_lazy_on() and x.is_cuda and bn.training and bn.act == "relu"
+            and conv.bias is None and conv.act is None
+            and C % 64 == 0 and conv.weight.shape[0] % 64 == 0
+            and 2048 % C == 0)
+
+
+def bn_conv_lazy(bn: "BatchNorm2d", conv: "Conv2d", x, stats=None,
+                 want_stats=False):
+    """relu(BN(x)) -> conv, with the BN apply fused into the conv's
+    A-side loads (ops.functional.bn_conv): the normalized activation is
+    never materialized. stats: precomputed (sum,sumsq) from the conv that
+    PRODUCED x; want_stats: emit epilogue stats of this conv's output for
+    the next BN. Returns (y, stats2_or_None)."""
+    from mi355x.ops import functional as F_
+
+    if bn.training:
+        bn._nbt += 1
+    out = F_.bn_conv(x, bn.weight, bn.bias, conv.weight, bn.running_mean,
+                     bn.running_var, bn.momentum, stats,
+                     bn.process_group if bn.training else None,
+                     conv.stride, conv.padding, want_stats)
+    if want_stats:
+        y, s2 = out
+        return y, (s2 if s2 is not None and s2.numel() else None)
+    return out, None
 
 
 def conv_bn_tap(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):

@@ -12,8 +12,8 @@ from __future__ import annotations
 import torch
 from torch import nn
 
-from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, conv_bn,
-                     conv_bn_tap, to_model_layout)
+from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, _lazy_ok,
+                     bn_conv_lazy, conv_bn, conv_bn_tap, to_model_layout)
 from mi355x import ops
 
 
@@ -29,6 +29,19 @@ class BasicBlock(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
+        # lazy-BN: bn1's apply fuses into conv2's A-loads (the normalized
+        # activation never materializes); conv1 carries the junction-grad
+        # TAP and emits bn1's stats from its epilogue
+        if _lazy_ok(self.bn1, self.conv2, x) and x.requires_grad:
+            from mi355x.ops import functional as F_
+
+            y1, tap, stats1 = F_.conv2d_tap_stats(
+                x, self.conv1.weight, self.conv1.stride, self.conv1.padding)
+            identity = (tap if self.downsample is None
+                        else self.downsample(tap))
+            y2, stats2 = bn_conv_lazy(self.bn1, self.conv2, y1, stats1,
+                                      want_stats=True)
+            return self.bn2(y2, residual=identity, stats=stats2)
         # the shortcut consumes conv1's TAP: the junction gradient fuses
         # into conv1's dgrad epilogue (layers.conv_bn_tap) instead of
         # autograd adding two full tensors at the block input
@@ -52,6 +65,20 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
+        if (_lazy_ok(self.bn1, self.conv2, x)
+                and _lazy_ok(self.bn2, self.conv3, x)
+                and x.requires_grad):
+            from mi355x.ops import functional as F_
+
+            y1, tap, stats1 = F_.conv2d_tap_stats(
+                x, self.conv1.weight, self.conv1.stride, self.conv1.padding)
+            identity = (tap if self.downsample is None
+                        else self.downsample(tap))
+            y2, stats2 = bn_conv_lazy(self.bn1, self.conv2, y1, stats1,
+                                      want_stats=True)
+            y3, stats3 = bn_conv_lazy(self.bn2, self.conv3, y2, stats2,
+                                      want_stats=True)
+            return self.bn3(y3, residual=identity, stats=stats3)
         out, tap = conv_bn_tap(self.conv1, self.bn1, x)
         identity = tap if self.downsample is None else self.downsample(tap)
         out = conv_bn(self.conv2, self.bn2, out)

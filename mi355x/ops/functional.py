@@ -223,6 +223,135 @@ def conv2d_tap(x, weight, stride=1, padding=0):
     return conv2d(x, weight, None, stride, padding, None), x
 
 
+class _ConvTapStatsFn(torch.autograd.Function):
+    """_ConvTapFn + epilogue BN stats: (y, tap, stats[2,K]) — the lazy-BN
+    block entry (conv1 with junction-grad tap and stats for bn1)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, stride, padding):
+        w16 = _w16_conv(weight, x.dtype)
+        y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding)
+        if stats is None:
+            stats = torch.empty(0, device=x.device, dtype=torch.float32)
+        ctx.save_for_backward(x, w16)
+        ctx.weight_ref = weight
+        ctx.conf = (stride, padding, weight.shape)
+        ctx.mark_non_differentiable(stats)
+        return y, x.view_as(x), stats
+
+    @staticmethod
+    def backward(ctx, dy, dtap, _unused):
+        x, w16 = ctx.saved_tensors
+        stride, padding, wshape = ctx.conf
+        dy = dy.contiguous()
+        dx = None
+        if ctx.needs_input_grad[0]:
+            wflip = _w16_conv_flip(ctx.weight_ref, dy.dtype)
+            add = (dtap.contiguous() if dtap is not None
+                   else torch.empty(0, device=x.device, dtype=x.dtype))
+            dx = ext().conv2d_dgrad(dy, wflip, stride, padding,
+                                    x.shape[1], x.shape[2], add)
+        elif dtap is not None:
+            dx = dtap
+        dw = None
+        if ctx.needs_input_grad[1]:
+            dw = ext().conv2d_wgrad(x, dy, wshape[2], wshape[3], stride,
+                                    padding)
+        return dx, dw, None, None
+
+
+def conv2d_tap_stats(x, weight, stride=1, padding=0):
+    """(y, tap, stats_or_None) — see _ConvTapStatsFn."""
+    y, tap, stats = _ConvTapStatsFn.apply(x, weight, stride, padding)
+    return y, tap, (stats if stats.numel() else None)
+
+
+class _BNConvFn(torch.autograd.Function):
+    """Lazy BN: the BatchNorm apply (normalize + affine + ReLU) is fused
+    into the CONSUMING conv's A-side load — the normalized activation z
+    is never materialized (saves one full read + write of the tensor per
+    internal BN; the bn_apply pass was 6-10% of the r18/r50 step).
+
+    forward(x=conv1 output, bn params, conv2 weight) -> conv2(relu(bn(x)))
+    [+ epilogue stats for the NEXT BN]. backward recomputes the ReLU mask
+    from x (z > 0  <=>  x*asc+ash > 0), runs conv2's dgrad/wgrad (wgrad
+    re-applies the transform on its x loads) and the standard fused BN
+    backward."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, conv_weight, running_mean, running_var,
+                momentum, stats, process_group, stride, padding, want_stats):
+        N, H, W, C = x.shape
+        m_local = N * H * W
+        s = stats if stats is not None else ext().bn_stats(x)
+        m_total = m_local
+        if process_group is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(s, group=process_group)
+            m_total = m_local * dist.get_world_size(process_group)
+        empty = torch.empty(0, device=x.device)
+        mi = ext().bn_finalize(
+            s, running_mean if running_mean is not None else empty,
+            running_var if running_var is not None else empty,
+            float(m_total), momentum, BN_EPS)
+        mean, invstd = mi[0], mi[1]
+        g32 = gamma.detach().float()
+        asc = g32 * invstd
+        ash = beta.detach().float() - mean * asc
+        w16 = _w16_conv(conv_weight, x.dtype)
+        y, stats2 = ext().conv2d_fwd_scaled(x, w16, asc, ash, stride,
+                                            padding, want_stats)
+        ctx.save_for_backward(x, w16, mean, invstd, gamma, asc, ash)
+        ctx.weight_ref = conv_weight
+        ctx.conf = (stride, padding, conv_weight.shape, m_total,
+                    process_group)
+        if want_stats:
+            ctx.mark_non_differentiable(stats2)
+            return y, stats2
+        return y
+
+    @staticmethod
+    def backward(ctx, dy2, *unused_stats_grad):
+        x, w16, mean, invstd, gamma, asc, ash = ctx.saved_tensors
+        stride, padding, wshape, m_total, pg = ctx.conf
+        dy2 = dy2.contiguous()
+        ef = torch.empty(0, device=x.device)
+        efx = torch.empty(0, device=x.device, dtype=x.dtype)
+        # conv2 backward w.r.t. the virtual z
+        dz = None
+        dw = None
+        if ctx.needs_input_grad[3]:
+            dw = ext().conv2d_wgrad_scaled(x, asc, ash, dy2, wshape[2],
+                                           wshape[3], stride, padding)
+        wflip = _w16_conv_flip(ctx.weight_ref, dy2.dtype)
+        dz = ext().conv2d_dgrad(dy2, wflip, stride, padding, x.shape[1],
+                                x.shape[2], efx)
+        # fused BN backward with the ReLU mask recomputed from x
+        emask = torch.empty(0, device=x.device, dtype=torch.uint8)
+        r = ext().bn_bwd_reduce(x, dz, efx, mean, invstd, emask, asc, ash)
+        if pg is not None:
+            import torch.distributed as dist
+
+            dist.all_reduce(r, group=pg)
+        dgamma, dbeta = r[0], r[1]
+        dx, _ = ext().bn_bwd_dx(x, dz, efx, mean, invstd,
+                                gamma.detach().float(), dgamma, dbeta,
+                                float(m_total), False, emask, asc, ash)
+        return (dx, dgamma, dbeta, dw, None, None, None, None, None, None,
+                None, None)
+
+
+def bn_conv(x, gamma, beta, conv_weight, running_mean, running_var,
+            momentum=0.1, stats=None, process_group=None, stride=1,
+            padding=0, want_stats=False):
+    """Fused BN(relu) -> conv2d (lazy-BN consumer fusion). Returns y, or
+    (y, stats2) when want_stats. GPU/MFMA shapes only; callers gate."""
+    return _BNConvFn.apply(x, gamma, beta, conv_weight, running_mean,
+                           running_var, momentum, stats, process_group,
+                           stride, padding, want_stats)
+
+
 def conv2d_with_stats(x, weight, stride=1, padding=0):
     """GPU conv returning (y, bn_stats[2,C] or None) — the fused
     conv->BN entry used by the ResNet blocks. stats is None when the
@@ -339,7 +468,8 @@ class _BNFn(torch.autograd.Function):
         # ReLU mask fused into both backward kernels (no standalone pass)
         y_arg = y if act == _ACT_RELU else torch.empty(0, device=x.device,
                                                        dtype=x.dtype)
-        r = ext().bn_bwd_reduce(x, dy, y_arg, mean, invstd, mask)  # [2,C]
+        ef = torch.empty(0, device=x.device)
+        r = ext().bn_bwd_reduce(x, dy, y_arg, mean, invstd, mask, ef, ef)
         if pg is not None:
             import torch.distributed as dist
 
@@ -348,7 +478,7 @@ class _BNFn(torch.autograd.Function):
         want_dres = has_res and act == _ACT_RELU
         dx, dres = ext().bn_bwd_dx(x, dy, y_arg, mean, invstd,
                                    gamma.detach().float(), dgamma, dbeta,
-                                   float(m_total), want_dres, mask)
+                                   float(m_total), want_dres, mask, ef, ef)
         if has_res and not want_dres:
             dres = dy  # no activation: residual grad is dy itself
         elif not has_res:

@@ -171,3 +171,35 @@ def test_residual_tap_grad_parity(model, size):
     # only the junction-add arithmetic differs (fp32 epilogue vs bf16
     # torch add) — grads must agree to bf16-accumulation tolerance
     torch.testing.assert_close(g1, g0, rtol=3e-2, atol=3e-3)
+
+
+@pytest.mark.parametrize("model,size", [("resnet18", 32), ("resnet50", 64)])
+def test_lazy_bn_grad_parity(model, size):
+    """Lazy BN (apply fused into the consuming conv's A-loads with
+    mask-recompute backward, ops.functional._BNConvFn) matches the
+    materialized-apply path within bf16 tolerance."""
+    import os
+
+    torch.manual_seed(4)
+    classes = 10 if model == "resnet18" else 1000
+    net = build_model(model, num_classes=classes).cuda()
+    flat = FlatState(net)
+    g = torch.Generator().manual_seed(6)
+    x = torch.randn(8, 3, size, size, generator=g).cuda()
+    yl = torch.randint(0, classes, (8,), generator=g).cuda()
+
+    def run(lazy):
+        os.environ["MI355X_LAZY_BN"] = lazy
+        flat.zero_grad()
+        loss = cross_entropy(net(x), yl)
+        loss.backward()
+        torch.cuda.synchronize()
+        return float(loss), flat.flat_grad.clone()
+
+    try:
+        l0, g0 = run("0")
+        l1, g1 = run("1")
+    finally:
+        os.environ.pop("MI355X_LAZY_BN", None)
+    assert abs(l1 - l0) < 5e-2 * max(abs(l0), 1.0), (l0, l1)
+    torch.testing.assert_close(g1, g0, rtol=5e-2, atol=5e-3)
