@@ -35,17 +35,38 @@ struct Mfma32<__half> {
   }
 };
 
-// lazy-BN A-side transform: z = relu(x*asc[c] + ash[c]) applied to an
-// 8-channel group in registers at load/stage time (the normalized
-// activation is never materialized; VALU rides under the MFMA phase)
+// lazy-BN A-side transform: z = relu(x*sc + sh) applied to an 8-channel
+// group in registers at load/stage time (the normalized activation is
+// never materialized; the VALU rides under the MFMA phase). Scale/shift
+// come in as four float4 REGISTERS — v1 indexed asc[c0+u] per element
+// (64 scalar L1 loads per thread per k-step) and measured -10% end to
+// end from load-issue bloat alone.
+struct Sc8 {
+  float4v s0, s1, h0, h1;
+};
+
+DEV_INLINE Sc8 sc8_load(const float* __restrict__ asc,
+                        const float* __restrict__ ash, int c) {
+  Sc8 r;
+  r.s0 = *reinterpret_cast<const float4v*>(asc + c);
+  r.s1 = *reinterpret_cast<const float4v*>(asc + c + 4);
+  r.h0 = *reinterpret_cast<const float4v*>(ash + c);
+  r.h1 = *reinterpret_cast<const float4v*>(ash + c + 4);
+  return r;
+}
+
 template <typename T16>
-DEV_INLINE short8 scale8(short8 v, const float* __restrict__ asc,
-                         const float* __restrict__ ash, int c0) {
+DEV_INLINE short8 scale8(short8 v, const Sc8& sc) {
   short8 o;
 #pragma unroll
-  for (int u = 0; u < 8; ++u) {
-    const float f = s16_to_f32<T16>(v[u]) * asc[c0 + u] + ash[c0 + u];
+  for (int u = 0; u < 4; ++u) {
+    const float f = s16_to_f32<T16>(v[u]) * sc.s0[u] + sc.h0[u];
     o[u] = f32_to_s16<T16>(fmaxf(f, 0.f));
+  }
+#pragma unroll
+  for (int u = 0; u < 4; ++u) {
+    const float f = s16_to_f32<T16>(v[4 + u]) * sc.s1[u] + sc.h1[u];
+    o[4 + u] = f32_to_s16<T16>(fmaxf(f, 0.f));
   }
   return o;
 }
@@ -198,7 +219,8 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
       if (va) {
 #pragma unroll
         for (int i = 0; i < 4; ++i)
-          sa[i] = scale8<T16>(sa[i], asc, ash, c0 + sa_c + 8 * i);
+          sa[i] = scale8<T16>(sa[i],
+                              sc8_load(asc, ash, c0 + sa_c + 8 * i));
       }
     }
     const T16* wp = wrow + (long)(r_ * S + s_) * b_rs_stride + c0;
@@ -462,7 +484,7 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
         v = *reinterpret_cast<const short8*>(
             in + ((nn * Hi + pp) * Wi + iw) * CI + cc * BK + g * 8);
         if constexpr (SCALED)
-          v = scale8<T16>(v, asc, ash, cc * BK + g * 8);
+          v = scale8<T16>(v, sc8_load(asc, ash, cc * BK + g * 8));
       }
       *reinterpret_cast<short8*>(patch + ((long)row * (Wi + 2) + col) * PCS +
                                  g * 8) = v;
@@ -617,6 +639,10 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
   const int j0base = KT == 64 ? (wave >> 1) * 32 : 0;
 
   f32x16 acc[NJ] = {};
+  Sc8 wsc{};  // SCALED: this thread's x channels are fixed (c0+skx)
+  if constexpr (SCALED) {
+    if (do_x) wsc = sc8_load(asc, ash, c0 + skx);
+  }
 
   // incremental (n,p,q) decode for the x-gather: one div/mod at entry,
   // add-with-carry as m advances by WGM per step (divisions in the inner
@@ -666,7 +692,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
                           skx)
                     : short8{};
         if constexpr (SCALED) {
-          if (ok) vx[mi] = scale8<T16>(vx[mi], asc, ash, c0 + skx);
+          if (ok) vx[mi] = scale8<T16>(vx[mi], wsc);
         }
         if (mi < 3 && ++q_ == Wo) {
           q_ = 0;
@@ -780,6 +806,8 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
   const int j0 = (wave >> 1) * 32;  // CI sub-offset
 
   f32x16 acc[2][2] = {};
+  Sc8 wsc{};
+  if constexpr (SCALED) wsc = sc8_load(asc, ash, c0 + sk);
 
   int dn = 0, dp = 0, dq = 0;
   {
@@ -816,7 +844,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
       const T16* xp = x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
       vx0[mi] = ok ? *reinterpret_cast<const short8*>(xp) : short8{};
       if constexpr (SCALED) {
-        if (ok) vx0[mi] = scale8<T16>(vx0[mi], asc, ash, c0 + sk);
+        if (ok) vx0[mi] = scale8<T16>(vx0[mi], wsc);
       }
       if (mi < 3 && ++q_ == Wo) {
         q_ = 0;
@@ -930,6 +958,10 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
   const int j0base = KT == 64 ? (wave >> 1) * 32 : 0;
 
   f32x16 acc[3][NJ] = {};
+  Sc8 wsc{};
+  if constexpr (SCALED) {
+    if (do_x) wsc = sc8_load(asc, ash, c0 + skx);
+  }
 
   // incremental decode for the CENTER gather position m = m0 + smx (the
   // i=0 halo position is derived from it: clamping a -1 start desyncs the
@@ -980,7 +1012,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
                          c0 + skx)
                    : short8{};
         if constexpr (SCALED) {
-          if (ok) vx[0] = scale8<T16>(vx[0], asc, ash, c0 + skx);
+          if (ok) vx[0] = scale8<T16>(vx[0], wsc);
         }
       }
 #pragma unroll
@@ -994,7 +1026,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
                          skx)
                    : short8{};
         if constexpr (SCALED) {
-          if (ok) vx[i] = scale8<T16>(vx[i], asc, ash, c0 + skx);
+          if (ok) vx[i] = scale8<T16>(vx[i], wsc);
         }
         if (i < 5 && ++q_ == Wo) {
           q_ = 0;

@@ -162,19 +162,49 @@ def _tap_on():
     return os.environ.get("MI355X_TAP", "1") != "0"
 
 
-def _lazy_on():
-    return os.environ.get("MI355X_LAZY_BN", "1") != "0"
+def _lazy_mode():
+    # "1": all eligible consumers; "g": only gather-path consumers (1x1 or
+    # strided convs — the HBM-bound shapes where removing the apply pass
+    # pays; the 3x3/s1 patch kernel is compute-bound and the fill-time
+    # transform sits on its critical path); "0": off
+    return os.environ.get("MI355X_LAZY_BN", "g")
 
 
 def _lazy_ok(bn: "BatchNorm2d", conv: "Conv2d", x) -> bool:
     """Lazy-BN (apply fused into the consuming conv) eligibility: GPU
     training, relu BN with no residual, MFMA conv shapes, fast-BN channel
     layout (C % 8 == 0 and 2048 % C == 0 for the mask-recompute backward)."""
+    mode = _lazy_mode()
+    if mode == "0":
+        return False
+    if mode == "g" and conv.weight.shape[2] != 1 and conv.stride == 1:
+        return False  # 3x3/s1 consumer = patch kernel: keep materialized
     C = conv.weight.shape[1]
-    return (_lazy_on() and x.is_cuda and bn.training and bn.act == "relu"
+    return (x.is_cuda and bn.training and bn.act == "relu"
             and conv.bias is None and conv.act is None
             and C % 64 == 0 and conv.weight.shape[0] % 64 == 0
             and 2048 % C == 0)
+
+
+def bn_then_conv(bn: "BatchNorm2d", conv: "Conv2d", x, stats=None,
+                 want_stats=False):
+    """bn(relu) -> conv with per-pair dispatch: lazy fusion when eligible,
+    else materialized apply followed by a (possibly stats-emitting) conv.
+    Returns (y, stats_out_or_None)."""
+    if _lazy_ok(bn, conv, x) and x.requires_grad:
+        return bn_conv_lazy(bn, conv, x, stats, want_stats)
+    z = bn(x, stats=stats)
+    if (want_stats and _fuse_stats() and z.is_cuda and bn.training
+            and conv.bias is None and conv.act is None
+            and conv.weight.shape[1] % 64 == 0):
+        from mi355x.ops import functional as F_
+
+        y, s2 = F_.conv2d_with_stats(z, conv.weight, conv.stride,
+                                     conv.padding)
+        if s2 is not None and s2.numel() == 0:
+            s2 = None
+        return y, s2
+    return conv(z), None
 
 
 def bn_conv_lazy(bn: "BatchNorm2d", conv: "Conv2d", x, stats=None,

@@ -13,7 +13,8 @@ import torch
 from torch import nn
 
 from .layers import (BatchNorm2d, Conv2d, Linear, MaxPool2d, _lazy_ok,
-                     bn_conv_lazy, conv_bn, conv_bn_tap, to_model_layout)
+                     bn_conv_lazy, bn_then_conv, conv_bn, conv_bn_tap,
+                     to_model_layout)
 from mi355x import ops
 
 
@@ -65,18 +66,21 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x):
-        if (_lazy_ok(self.bn1, self.conv2, x)
-                and _lazy_ok(self.bn2, self.conv3, x)
-                and x.requires_grad):
+        # per-pair lazy dispatch: each internal BN fuses into its consumer
+        # conv when eligible (bn_then_conv), else applies materialized —
+        # one fused-stats chain either way
+        if x.is_cuda and x.requires_grad and self.bn1.training and (
+                _lazy_ok(self.bn1, self.conv2, x)
+                or _lazy_ok(self.bn2, self.conv3, x)):
             from mi355x.ops import functional as F_
 
             y1, tap, stats1 = F_.conv2d_tap_stats(
                 x, self.conv1.weight, self.conv1.stride, self.conv1.padding)
             identity = (tap if self.downsample is None
                         else self.downsample(tap))
-            y2, stats2 = bn_conv_lazy(self.bn1, self.conv2, y1, stats1,
+            y2, stats2 = bn_then_conv(self.bn1, self.conv2, y1, stats1,
                                       want_stats=True)
-            y3, stats3 = bn_conv_lazy(self.bn2, self.conv3, y2, stats2,
+            y3, stats3 = bn_then_conv(self.bn2, self.conv3, y2, stats2,
                                       want_stats=True)
             return self.bn3(y3, residual=identity, stats=stats3)
         out, tap = conv_bn_tap(self.conv1, self.bn1, x)
