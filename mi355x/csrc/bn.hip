@@ -127,7 +127,7 @@ __global__ void bn_bwd_dx_kernel(const T16* __restrict__ x,
 // asc/ash (BWD only, nullable): recompute the ReLU mask as
 // (x*asc[c] + ash[c]) > 0 — the lazy-BN path (apply fused into the
 // consuming conv) never materializes y or a mask
-template <typename T16, bool BWD>
+template <typename T16, bool BWD, bool RC = false>
 __global__ __launch_bounds__(256) void bn_reduce_fast(
     const T16* __restrict__ x, const T16* __restrict__ dy,
     const unsigned char* __restrict__ msk, const float* __restrict__ mean,
@@ -145,8 +145,10 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
     for (int u = 0; u < 8; ++u) {
       mu[u] = mean[c0 + u];
       is[u] = invstd[c0 + u];
-      ac[u] = asc ? asc[c0 + u] : 0.f;
-      ah[u] = ash ? ash[c0 + u] : 0.f;
+      if (RC) {
+        ac[u] = asc[c0 + u];
+        ah[u] = ash[c0 + u];
+      }
     }
   }
   const long e0 = (long)blockIdx.x * e_per_block;
@@ -171,7 +173,7 @@ __global__ __launch_bounds__(256) void bn_reduce_fast(
       for (int u = 0; u < 8; ++u) {
         float d = s16_to_f32<T16>(vd[u]);
         const float xv = s16_to_f32<T16>(vx[u]);
-        if (asc) {
+        if (RC) {
           if (xv * ac[u] + ah[u] <= 0.f) d = 0.f;
         } else if (!((mb >> u) & 1)) {
           d = 0.f;
@@ -266,7 +268,7 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
   for (; e < E; e += stride) body(e);
 }
 
-template <typename T16>
+template <typename T16, bool RC = false>
 __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     const T16* __restrict__ x, const T16* __restrict__ dy,
     const unsigned char* __restrict__ msk, const float* __restrict__ mean,
@@ -284,8 +286,10 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     g_[u] = gamma[c0 + u] * is[u];
     a_[u] = dgamma[c0 + u] * inv_m;
     b_[u] = dbeta[c0 + u] * inv_m;
-    ac[u] = asc ? asc[c0 + u] : 0.f;
-    ah[u] = ash ? ash[c0 + u] : 0.f;
+    if (RC) {
+      ac[u] = asc[c0 + u];
+      ah[u] = ash[c0 + u];
+    }
   }
   const long stride = (long)gridDim.x * blockDim.x * 8;
   auto body = [&](long e) {
@@ -297,7 +301,7 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     for (int u = 0; u < 8; ++u) {
       float d = s16_to_f32<T16>(vd[u]);
       const float xv = s16_to_f32<T16>(vx[u]);
-      if (asc) {
+      if (RC) {
         if (xv * ac[u] + ah[u] <= 0.f) d = 0.f;
       } else if (!((mb >> u) & 1)) {
         d = 0.f;
@@ -481,16 +485,26 @@ at::Tensor bn_bwd_reduce(at::Tensor x, at::Tensor dy, at::Tensor y,
     dim3 grid = bn_fast_grid(E, e_per_block);
     auto out = grid.x == 1 ? at::empty({2, C}, x.options().dtype(at::kFloat))
                            : at::zeros({2, C}, x.options().dtype(at::kFloat));
+    const bool rc = asc.defined() && asc.numel() > 0;
     DISPATCH_16(x, T16, {
-      hipLaunchKernelGGL((bn_reduce_fast<T16, true>), grid, dim3(256),
-                         2 * C * sizeof(float), cur_stream(),
-                         (const T16*)x.data_ptr(), (const T16*)dy.data_ptr(),
-                         mask.numel() ? mask.data_ptr<unsigned char>()
-                                      : nullptr,
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         asc.numel() ? asc.data_ptr<float>() : nullptr,
-                         ash.numel() ? ash.data_ptr<float>() : nullptr,
-                         out.data_ptr<float>(), E, C, e_per_block);
+      if (rc)
+        hipLaunchKernelGGL((bn_reduce_fast<T16, true, true>), grid,
+                           dim3(256), 2 * C * sizeof(float), cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           asc.data_ptr<float>(), ash.data_ptr<float>(),
+                           out.data_ptr<float>(), E, C, e_per_block);
+      else
+        hipLaunchKernelGGL((bn_reduce_fast<T16, true>), grid, dim3(256),
+                           2 * C * sizeof(float), cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(),
+                           mask.numel() ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           nullptr, nullptr,
+                           out.data_ptr<float>(), E, C, e_per_block);
     });
     return out;
   }
@@ -528,20 +542,31 @@ std::vector<at::Tensor> bn_bwd_dx(at::Tensor x, at::Tensor dy, at::Tensor y,
   if (bn_fast_ok(M, C)) {
     const long E = M * (long)C;
     const int grid = (int)std::min<long>(cdiv_l(E, 256 * 8), 2048);
+    const bool rc = asc.defined() && asc.numel() > 0;
     DISPATCH_16(x, T16, {
-      hipLaunchKernelGGL(bn_bwd_dx_fast<T16>, dim3(grid), dim3(256), 0,
-                         cur_stream(), (const T16*)x.data_ptr(),
-                         (const T16*)dy.data_ptr(),
-                         mask.numel() ? mask.data_ptr<unsigned char>()
-                                      : nullptr,
-                         mean.data_ptr<float>(), invstd.data_ptr<float>(),
-                         gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
-                         dbeta.data_ptr<float>(),
-                         asc.numel() ? asc.data_ptr<float>() : nullptr,
-                         ash.numel() ? ash.data_ptr<float>() : nullptr,
-                         (T16*)dx.data_ptr(),
-                         want_dres ? (T16*)dres.data_ptr() : nullptr, E, C,
-                         (float)(1.0 / m_total));
+      if (rc)
+        hipLaunchKernelGGL((bn_bwd_dx_fast<T16, true>), dim3(grid),
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+                           dbeta.data_ptr<float>(), asc.data_ptr<float>(),
+                           ash.data_ptr<float>(), (T16*)dx.data_ptr(),
+                           want_dres ? (T16*)dres.data_ptr() : nullptr, E, C,
+                           (float)(1.0 / m_total));
+      else
+        hipLaunchKernelGGL((bn_bwd_dx_fast<T16>), dim3(grid), dim3(256), 0,
+                           cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(),
+                           mask.numel() ? mask.data_ptr<unsigned char>()
+                                        : nullptr,
+                           mean.data_ptr<float>(), invstd.data_ptr<float>(),
+                           gamma.data_ptr<float>(), dgamma.data_ptr<float>(),
+                           dbeta.data_ptr<float>(), nullptr, nullptr,
+                           (T16*)dx.data_ptr(),
+                           want_dres ? (T16*)dres.data_ptr() : nullptr, E, C,
+                           (float)(1.0 / m_total));
     });
     return {dx, want_dres ? dres : at::Tensor()};
   }

@@ -596,8 +596,9 @@ template <typename T16, int KT, bool SCALED = false>
 // the second launch-bounds arg (min 2 blocks/CU) is load-bearing: without
 // it the compiler allocates 188-256 VGPRs (occupancy 1-2) for no speedup;
 // with it 104-124 VGPRs, zero spills, occupancy 4 - the extra waves are
-// what covers the gather's HBM latency (PMC: 66% WAIT_ANY at occ 2)
-__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
+// what covers the gather's HBM latency (PMC: 66% WAIT_ANY at occ 2).
+// SCALED variants pin the unscaled occupancy (the Sc8 regs cost a tier).
+__global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
     const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
@@ -772,8 +773,10 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_kernel(
 // dy+x stage feeds 16 MFMAs per wave per m-step instead of 8, and each
 // fragment read is reused twice. Same transposed-staging/split-M/slab
 // machinery as conv_wgrad_mfma_kernel.
+// SCALED: min 3 waves/SIMD — the Sc8 registers pushed the allocation 4
+// VGPRs over the 168 boundary (occ 3 -> 2, measured 1.77x slower)
 template <typename T16, bool SCALED = false>
-__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
+__global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
     const T16* __restrict__ x, const T16* __restrict__ dy,
     const float* __restrict__ asc,  // SCALED only
     const float* __restrict__ ash,
@@ -921,7 +924,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_t128(
 // Triples the MFMA work per staged dy/x byte — the plain kernel was
 // staging-bound at ~190 TF while the fwd gather-GEMM reaches 470-670.
 template <typename T16, int KT, bool SCALED = false>
-__global__ __launch_bounds__(256, 2) void conv_wgrad_mfma_s3(
+__global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
     const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale

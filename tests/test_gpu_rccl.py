@@ -37,3 +37,37 @@ def test_native_comm_bf16():
     c.wait()
     torch.cuda.synchronize()
     torch.testing.assert_close(t, ref)
+
+
+def test_native_comm_inside_hip_graph():
+    """RCCL collectives capture into a hipGraph and replay (world 1): the
+    mechanics behind MI355X_GRAPH_DIST — DDP bucket all-reduces inside a
+    captured train step. The comm runs on its own
+    stream with event edges, so capture must thread through it."""
+    from mi355x.parallel.rccl import native_comm
+
+    c = native_comm()
+    t = torch.zeros(1024, device="cuda")
+    # warmup on a side stream (allocator + comm init outside capture)
+    s = torch.cuda.Stream()
+    s.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(s):
+        t += 1.0
+        c.all_reduce(t)
+        c.wait()
+    torch.cuda.current_stream().wait_stream(s)
+    torch.cuda.synchronize()
+    t.zero_()
+    graph = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(graph):
+        t += 1.0
+        c.all_reduce(t)  # world-1 sum: identity, but exercises capture
+        c.wait()
+        t *= 2.0
+    torch.cuda.synchronize()
+    t.zero_()
+    for _ in range(3):
+        graph.replay()
+    torch.cuda.synchronize()
+    # 3 replays of (t += 1; allreduce(identity); t *= 2): 2, 6, 14
+    torch.testing.assert_close(t.cpu(), torch.full((1024,), 14.0))
