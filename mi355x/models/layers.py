@@ -163,11 +163,13 @@ def _tap_on():
 
 
 def _lazy_mode():
-    # "1": all eligible consumers; "g": only gather-path consumers (1x1 or
-    # strided convs — the HBM-bound shapes where removing the apply pass
-    # pays; the 3x3/s1 patch kernel is compute-bound and the fill-time
-    # transform sits on its critical path); "0": off
-    return os.environ.get("MI355X_LAZY_BN", "g")
+    # "1": all eligible consumers; "g": only gather-path consumers (1x1 /
+    # strided convs); "0": off. DEFAULT OFF — a measured negative result:
+    # even with vectorized Sc8 loads and pinned occupancy the transform
+    # work on the conv staging paths costs more than the removed apply
+    # pass saves (r50-224: off 9507, g 9198, all 8704 img/s; r18: off
+    # 111.7k, all 98.7k). Kept as a tested opt-in for future work.
+    return os.environ.get("MI355X_LAZY_BN", "0")
 
 
 def _lazy_ok(bn: "BatchNorm2d", conv: "Conv2d", x) -> bool:
