@@ -357,6 +357,144 @@ __global__ void conv_stats_reduce(const float* __restrict__ slab,
     atomicAdd(stats + h * C + c, acc);
 }
 
+// stride-2 dgrad, parity-specialized: dx positions partition by
+// (ih%2, iw%2) and each class only has contributions from taps of
+// matching parity ((ih+pad-r) % 2 == 0), so the k-loop runs
+// |Ra|*|Sa|*cchunks steps instead of R*S*cchunks. The plain TRANS gather
+// wastes ~75% of its staging+MFMA on parity-invalid (all-zero) slices
+// for 3x3/s2, and ALL of it for 3 of the 4 classes of a 1x1/s2
+// downsample (those classes here just store zeros). ADDIN as in
+// conv_gather_gemm (residual-tap fusion).
+template <typename T16, int NT, bool ADDIN = false>
+__global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
+    const T16* __restrict__ dy,     // [N, P, Q, KO]
+    const T16* __restrict__ wflip,  // [R, S, CI, KO]
+    const T16* __restrict__ zpage,
+    const T16* __restrict__ addin,  // ADDIN only: dx-shaped
+    T16* __restrict__ dx,           // [N, H, W, CI]
+    const int N, const int P, const int Q, const int KO, const int CI,
+    const int H, const int W, const int R, const int S, const int pad) {
+  __shared__ T16 lds[BM * LDK + NT * 32 * LDK];
+  const int tid = threadIdx.x;
+  const int pa = blockIdx.z >> 1;  // ih % 2
+  const int pb = blockIdx.z & 1;   // iw % 2
+  const int Ha = (H - pa + 1) >> 1;
+  const int Wa = (W - pb + 1) >> 1;
+  const long Mc = (long)N * Ha * Wa;
+  const long bm0 = (long)blockIdx.x * BM;
+  constexpr int BNT = NT * 32;
+  const int k0 = blockIdx.y * BNT;  // dx channel tile
+
+  const int r0 = (pa + pad) & 1;  // r parity for this class
+  const int s0 = (pb + pad) & 1;
+  const int nr = r0 < R ? ((R - r0 + 1) >> 1) : 0;
+  const int ns = s0 < S ? ((S - s0 + 1) >> 1) : 0;
+  const int cchunks = KO / BK;
+  const int ksteps = nr * ns * cchunks;
+
+  const int sa_m = tid >> 1;
+  const int sa_c = (tid & 1) * (BK / 2);
+  const long m_a = bm0 + sa_m;
+  const bool m_ok = m_a < Mc;
+  int n_ = 0, ih_ = 0, iw_ = 0;
+  if (m_ok) {
+    n_ = (int)(m_a / ((long)Ha * Wa));
+    const int rem = (int)(m_a % ((long)Ha * Wa));
+    ih_ = 2 * (rem / Wa) + pa;
+    iw_ = 2 * (rem % Wa) + pb;
+  }
+
+  constexpr int TPR = 256 / BNT;
+  constexpr int EPT = BK / TPR;
+  const int sb_n = tid / TPR;
+  const int sb_c = (tid % TPR) * EPT;
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int wm = wave * 32;
+  f32x16 acc[NT] = {};
+  short8 sa[4], sb[EPT / 8];
+
+  auto load_step = [&](int j) {
+    const int c0 = (j % cchunks) * BK;
+    const int t = j / cchunks;
+    const int s_ = s0 + 2 * (t % ns);
+    const int r_ = r0 + 2 * (t / ns);
+    const int dh = ih_ + pad - r_;  // even by construction
+    const int dw_ = iw_ + pad - s_;
+    const int pp = dh >> 1, qq = dw_ >> 1;
+    const bool va = m_ok && dh >= 0 && dw_ >= 0 && pp < P && qq < Q;
+    const T16* xp =
+        va ? dy + (((long)n_ * P + pp) * Q + qq) * KO + c0 + sa_c : zpage;
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      sa[i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
+    const T16* wp = wflip + (((long)r_ * S + s_) * CI + (k0 + sb_n)) * KO +
+                    c0 + sb_c;
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
+  };
+  auto stage = [&]() {
+    T16* ldsA = lds;
+    T16* ldsB = lds + BM * LDK;
+    short* pa_ = reinterpret_cast<short*>(ldsA + sa_m * LDK + sa_c);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+      *reinterpret_cast<short8*>(pa_ + 8 * i) = sa[i];
+    short* pb_ = reinterpret_cast<short*>(ldsB + sb_n * LDK + sb_c);
+#pragma unroll
+    for (int i = 0; i < EPT / 8; ++i)
+      *reinterpret_cast<short8*>(pb_ + 8 * i) = sb[i];
+  };
+
+  if (ksteps > 0) {
+    load_step(0);
+    for (int j = 0; j < ksteps; ++j) {
+      __syncthreads();
+      stage();
+      __syncthreads();
+      if (j + 1 < ksteps) load_step(j + 1);
+      const T16* ldsA = lds;
+      const T16* ldsB = lds + BM * LDK;
+#pragma unroll
+      for (int kk = 0; kk < BK; kk += 16) {
+        const short8 af = *reinterpret_cast<const short8*>(
+            ldsA + (wm + li) * LDK + kk + kh * 8);
+#pragma unroll
+        for (int tnt = 0; tnt < NT; ++tnt) {
+          const short8 bf = *reinterpret_cast<const short8*>(
+              ldsB + (tnt * 32 + li) * LDK + kk + kh * 8);
+          acc[tnt] = Mfma32<T16>::run(af, bf, acc[tnt]);
+        }
+      }
+    }
+  }
+
+  // epilogue: scatter into the strided class positions of dx
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mc) {
+      const int nn = (int)(m_out / ((long)Ha * Wa));
+      const int rem = (int)(m_out % ((long)Ha * Wa));
+      const int ih = 2 * (rem / Wa) + pa;
+      const int iw = 2 * (rem % Wa) + pb;
+      const long off = (((long)nn * H + ih) * W + iw) * CI;
+#pragma unroll
+      for (int tnt = 0; tnt < NT; ++tnt) {
+        float v = acc[tnt][reg];
+        if constexpr (ADDIN)
+          v += F16<T16>::to_f32(addin[off + k0 + tnt * 32 + li]);
+        dx[off + k0 + tnt * 32 + li] = F16<T16>::from_f32(v);
+      }
+    }
+  }
+}
+
 // fold slab [by][gx][bnt2] -> stats[2,C]; two-level when gx is large
 void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
                        int C) {
@@ -1892,6 +2030,57 @@ void conv_dgrad_mfma_launch(at::Tensor dy, at::Tensor wflip, at::Tensor dx,
   const int H = dx.size(1), W = dx.size(2), CI = dx.size(3);
   const long M = (long)N * H * W;
   const bool has_add = addin.defined() && addin.numel() > 0;
+  // stride-2: parity-specialized kernel (4 classes in blockIdx.z), ~2.6x
+  // fewer k-steps for 3x3/s2 and 4x for the 1x1 downsamples
+  static const bool s2_on = [] {
+    const char* e = getenv("MI355X_DGRAD_S2");
+    return !e || e[0] != '0';
+  }();
+  if (s2_on && stride == 2 && pad <= 8) {
+    const long Mmax = (long)N * ((H + 1) / 2) * ((W + 1) / 2);
+    const bool wide =
+        CI % 128 == 0 && cdiv_l(Mmax, BM) * (CI / 128) * 4 >= 1024;
+    const int BNTs = wide ? 128 : 64;
+    dim3 grid((unsigned)cdiv_l(Mmax, BM), CI / BNTs, 4);
+    at::Tensor zp = conv_zero_page(dy);
+    DISPATCH_16(dy, T16, {
+      const T16* add_p = has_add ? (const T16*)addin.data_ptr() : nullptr;
+      if (wide) {
+        if (has_add)
+          hipLaunchKernelGGL((conv_dgrad_s2_gemm<T16, 4, true>), grid,
+                             dim3(256), 0, cur_stream(),
+                             (const T16*)dy.data_ptr(),
+                             (const T16*)wflip.data_ptr(),
+                             (const T16*)zp.data_ptr(), add_p,
+                             (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W,
+                             (int)R, (int)S, (int)pad);
+        else
+          hipLaunchKernelGGL((conv_dgrad_s2_gemm<T16, 4>), grid, dim3(256),
+                             0, cur_stream(), (const T16*)dy.data_ptr(),
+                             (const T16*)wflip.data_ptr(),
+                             (const T16*)zp.data_ptr(), nullptr,
+                             (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W,
+                             (int)R, (int)S, (int)pad);
+      } else {
+        if (has_add)
+          hipLaunchKernelGGL((conv_dgrad_s2_gemm<T16, 2, true>), grid,
+                             dim3(256), 0, cur_stream(),
+                             (const T16*)dy.data_ptr(),
+                             (const T16*)wflip.data_ptr(),
+                             (const T16*)zp.data_ptr(), add_p,
+                             (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W,
+                             (int)R, (int)S, (int)pad);
+        else
+          hipLaunchKernelGGL((conv_dgrad_s2_gemm<T16, 2>), grid, dim3(256),
+                             0, cur_stream(), (const T16*)dy.data_ptr(),
+                             (const T16*)wflip.data_ptr(),
+                             (const T16*)zp.data_ptr(), nullptr,
+                             (T16*)dx.data_ptr(), N, P, Q, KO, CI, H, W,
+                             (int)R, (int)S, (int)pad);
+      }
+    });
+    return;
+  }
   static const bool patch_on = [] {
     const char* e = getenv("MI355X_CONV_PATCH");
     return !e || e[0] != '0';
