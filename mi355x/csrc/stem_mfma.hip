@@ -203,6 +203,138 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
   }
 }
 
+// ---- forward v3: strip-staged -----------------------------------------
+// One block = 128 consecutive outputs of ONE image (requires Ho*Wo % 128
+// == 0 so blocks never straddle images). The 7-11 input rows the block's
+// receptive fields span are staged ONCE into an LDS strip (coalesced row
+// copies with zero pads); the MFMA A-fragments then read 8-element tap
+// spans DIRECTLY from the strip (strip idx 6q + jj + u — contiguous,
+// 4B-aligned) instead of v2's per-m scattered 16-byte global gathers
+// (~45 KB of uncoalesced reads per block). Weights [64,192] live in LDS.
+template <typename T16>
+__global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
+    const T16* __restrict__ x,    // [N, H, W, 3]
+    const T16* __restrict__ w24,  // [KO, 192] row-padded
+    const float* __restrict__ bias, T16* __restrict__ y, const int N,
+    const int H, const int W, const int KO, const int Ho, const int Wo,
+    const int pad, const int act, const int has_bias, const int nrows,
+    const int sstride) {
+  extern __shared__ __attribute__((aligned(16))) char ssmem[];
+  T16* strip = reinterpret_cast<T16*>(ssmem);  // [nrows+1][sstride]
+  T16* ldsB = strip + (long)(nrows + 1) * sstride;  // [64][200]
+  const int tid = threadIdx.x;
+  const int W3 = 3 * W;
+  const long Mimg = (long)Ho * Wo;
+  const long bm0 = (long)blockIdx.x * 128;
+  const int n = (int)(bm0 / Mimg);
+  const long bmi = bm0 - (long)n * Mimg;
+  const int p0 = (int)(bmi / Wo);
+  const int ih0 = 2 * p0 - pad;  // strip row 0 = input row ih0
+  const int k0 = blockIdx.y * 64;
+
+  // ---- strip fill: rows ih0 .. ih0+nrows-1, each 3*pad zeros | row |
+  // zeros; row nrows is the all-zero stub for r >= 7 k-pad reads ----
+  const int off3 = 3 * pad;
+  const int nv8 = (W3 + 7) / 8;
+  for (int t = tid; t < nrows * nv8; t += 256) {
+    const int rr = t / nv8;
+    const int j8 = (t - rr * nv8) * 8;
+    const int ih = ih0 + rr;
+    T16* dst = strip + (long)rr * sstride + off3 + j8;
+    if ((unsigned)ih < (unsigned)H && j8 + 8 <= W3) {
+      *reinterpret_cast<short8*>(dst) =
+          *reinterpret_cast<const short8*>(x + ((long)n * H + ih) * W3 + j8);
+    } else {
+      const T16 z{};
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        dst[u] = (j8 + u < W3 && (unsigned)ih < (unsigned)H)
+                     ? x[((long)n * H + ih) * W3 + j8 + u]
+                     : z;
+    }
+  }
+  {  // pad columns of every row + the zero stub row
+    const int nz = sstride - W3;
+    const T16 z{};
+    for (int t = tid; t < nrows * nz; t += 256) {
+      const int rr = t / nz;
+      const int e = t - rr * nz;
+      strip[(long)rr * sstride + (e < off3 ? e : W3 + e)] = z;
+    }
+    for (int t = tid; t < sstride; t += 256)
+      strip[(long)nrows * sstride + t] = z;
+  }
+  // ---- B fill: [64][SKG] at row stride SKG+8 ----
+  for (int e = tid * 8; e < 64 * SKG; e += 256 * 8) {
+    const int row = e / SKG, col = e - row * SKG;
+    *reinterpret_cast<short8*>(
+        reinterpret_cast<short*>(ldsB + row * (SKG + 8) + col)) =
+        *reinterpret_cast<const short8*>(w24 + (long)(k0 + row) * SKG + col);
+  }
+  __syncthreads();
+
+  const int wave = tid >> 6;
+  const int lane = tid & 63;
+  const int li = lane & 31;
+  const int kh = lane >> 5;
+  const int wm = wave * 32;
+  // per-lane output coords (fixed): m = bm0 + wm + li
+  const long mi = bmi + wm + li;
+  const int p_lane = (int)(mi / Wo);
+  const int q_lane = (int)(mi - (long)p_lane * Wo);
+  const int rr_base = 2 * (p_lane - p0);  // strip row of tap r=0
+  const int ebase = 6 * q_lane;           // strip elem of (jj=0) tap col
+  f32x16_s acc[2] = {};
+
+#pragma unroll
+  for (int slice = 0; slice < SKG / 16; ++slice) {
+    const int kg0 = slice * 16 + kh * 8;  // wave-uniform per kh
+    const int r = kg0 / SROW;
+    const int jj = kg0 - r * SROW;
+    const int rr = r < 7 ? rr_base + r : nrows;  // stub row for k-pad
+    const short* ap = reinterpret_cast<const short*>(
+        strip + (long)rr * sstride + ebase + jj);
+    short8 af;
+#pragma unroll
+    for (int u2 = 0; u2 < 4; ++u2) {
+      int v32;
+      __builtin_memcpy(&v32, ap + 2 * u2, 4);
+      __builtin_memcpy(reinterpret_cast<char*>(&af) + 4 * u2, &v32, 4);
+    }
+    // zero the row-pad lanes (jj+u >= 21) — weights there are zero too,
+    // but the strip span can run past the zero pads near the row end
+    if (jj > 21 - 8) {
+#pragma unroll
+      for (int u = 0; u < 8; ++u)
+        if (jj + u >= 21) af[u] = 0;
+    }
+    const short8 b0 = *reinterpret_cast<const short8*>(
+        ldsB + li * (SKG + 8) + slice * 16 + kh * 8);
+    const short8 b1 = *reinterpret_cast<const short8*>(
+        ldsB + (32 + li) * (SKG + 8) + slice * 16 + kh * 8);
+    acc[0] = SMfma<T16>::run(af, b0, acc[0]);
+    acc[1] = SMfma<T16>::run(af, b1, acc[1]);
+  }
+
+  float bv[2];
+  bv[0] = has_bias ? bias[k0 + li] : 0.f;
+  bv[1] = has_bias ? bias[k0 + 32 + li] : 0.f;
+  const long Mtot = (long)N * Mimg;
+#pragma unroll
+  for (int reg = 0; reg < 16; ++reg) {
+    const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+    const long m_out = bm0 + wm + row;
+    if (m_out < Mtot) {
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2) {
+        float v = acc[t2][reg] + bv[t2];
+        if (act == 1) v = fmaxf(v, 0.f);
+        y[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
+      }
+    }
+  }
+}
+
 // ---- wgrad ------------------------------------------------------------
 // dw slab scatter target layout: [KO, 3, 7, 7] (parameter layout), one
 // slab per m-chunk; grid = (KO/64, 3 kg-chunks, m-chunks).
@@ -372,10 +504,39 @@ void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const int Ho = y.size(1), Wo = y.size(2);
   TORCH_CHECK(KO % 64 == 0, "stem GEMM expects KO % 64 == 0");
   auto w24 = stem_w24(w);
-  at::Tensor zp = conv_zero_page(x);
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)cdiv_l(M, 128), KO / 64);
   const int has_bias = bias.numel() > 0;
+  // strip variant (v3) when blocks can't straddle images: the A operand
+  // comes from a once-staged LDS strip instead of per-m global gathers
+  static const bool strip_on = [] {
+    const char* e = getenv("MI355X_STEM_STRIP");
+    return !e || e[0] != '0';
+  }();
+  const long Mimg = (long)Ho * Wo;
+  if (strip_on && Mimg % 128 == 0) {
+    // rows spanned: 128 outputs cover <= ceil(127/Wo)+1 p-rows; input
+    // rows 2*p0-pad .. 2*p_last+6-pad
+    const int pspan = (int)((127 / Wo) + 1);
+    const int nrows = 2 * (pspan - 1) + 7;
+    // row length: left pads (3*pad) + payload (3W) + right slack for the
+    // widest tap span (max strip idx = 3W + 6*pad + 2), rounded to 4
+    const int sstride = (3 * W + 6 * (int)pad + 8 + 3) & ~3;
+    const size_t smem =
+        ((size_t)(nrows + 1) * sstride + 64 * (SKG + 8)) * x.element_size();
+    if (smem <= 150 * 1024) {
+      DISPATCH_16(x, T16, {
+        hipLaunchKernelGGL((conv_fwd_stem_strip<T16>), grid, dim3(256),
+                           smem, cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)w24.data_ptr(),
+                           has_bias ? bias.data_ptr<float>() : nullptr,
+                           (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo,
+                           (int)pad, (int)act, has_bias, nrows, sstride);
+      });
+      return;
+    }
+  }
+  at::Tensor zp = conv_zero_page(x);
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL((conv_fwd_stem_gemm<T16>), grid, dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
