@@ -337,7 +337,10 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
 
 // ---- wgrad ------------------------------------------------------------
 // dw slab scatter target layout: [KO, 3, 7, 7] (parameter layout), one
-// slab per m-chunk; grid = (KO/64, 3 kg-chunks, m-chunks).
+// slab per m-chunk; grid = (KO/64, m-chunks). ALL 192 kg in one block:
+// dy is staged once for 3x the MFMA work (the kg-chunked version staged
+// the same dy per chunk and its 4-MFMA inner loop couldn't cover the
+// scattered x-gather latency).
 template <typename T16>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     const T16* __restrict__ x,   // [N, H, W, 3]
@@ -346,12 +349,11 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     float* __restrict__ dw,      // chunk slabs of [KO*3*7*7]
     const int N, const int H, const int W, const int KO, const int Ho,
     const int Wo, const int pad, const long m_per_chunk) {
-  __shared__ T16 lds[(64 + 64) * SLDM];
+  __shared__ T16 lds[(64 + SKG) * SLDM];
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
   const int k0 = blockIdx.x * 64;
-  const int c0 = blockIdx.y * 64;  // kg-chunk
-  const long m_begin = (long)blockIdx.z * m_per_chunk;
+  const long m_begin = (long)blockIdx.y * m_per_chunk;
   const long m_end = min(Mtot, m_begin + m_per_chunk);
   const int W3 = 3 * W;
 
@@ -362,16 +364,16 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
   const int sk = (t >> 4) * 8;
   const int tx = tid & 127;
   const int smx = (tx & 15) * 4;
-  const int skx = (tx >> 4) * 8;
+  const int gx0 = (tx >> 4) * 3;  // this thread's 3 kg-groups
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
   const int li = lane & 31;
   const int kh = lane >> 5;
   const int i0 = (wave & 1) * 32;       // KO sub-tile
-  const int j0 = (wave >> 1) * 32;      // kg sub-tile
+  const int j0 = (wave >> 1) * 32;      // kg sub-tile base (stride 64)
 
-  f32x16_s acc = {};
+  f32x16_s acc[3] = {};
 
   int dn = 0, dp = 0, dq = 0;
   {
@@ -392,7 +394,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     }
   };
 
-  short8 vdy[4], vx[4];
+  short8 vdy[4], vx[3][4];
   auto load_m = [&](long m0) {
     if (do_dy) {
 #pragma unroll
@@ -404,26 +406,30 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
       }
     }
     if (do_x) {
-      G8<T16> g[4];
-      int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        const long m = m0 + smx + mi;
-        g8_plan(g[mi], x, zpage, n_, p_, q_, c0 + skx, H, W3, pad,
-                m < m_end);
-        if (mi < 3 && ++q_ == Wo) {
-          q_ = 0;
-          if (++p_ == Ho) {
-            p_ = 0;
-            ++n_;
+      for (int ii = 0; ii < 3; ++ii) {
+        G8<T16> g[4];
+        int n_ = dn, p_ = dp, q_ = dq;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          const long m = m0 + smx + mi;
+          g8_plan(g[mi], x, zpage, n_, p_, q_, (gx0 + ii) * 8, H, W3, pad,
+                  m < m_end);
+          if (mi < 3 && ++q_ == Wo) {
+            q_ = 0;
+            if (++p_ == Ho) {
+              p_ = 0;
+              ++n_;
+            }
           }
         }
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+          vx[ii][mi] = *reinterpret_cast<const short8*>(g[mi].src);
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+          vx[ii][mi] = g8_fix(g[mi], vx[ii][mi], x, W3);
       }
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        vx[mi] = *reinterpret_cast<const short8*>(g[mi].src);
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi) vx[mi] = g8_fix(g[mi], vx[mi], x, W3);
       advance(64);
     }
   };
@@ -439,10 +445,15 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     if (do_x) {
       T16* ldsT = lds + 64 * SLDM;
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        short4v pk = {vx[0][e], vx[1][e], vx[2][e], vx[3][e]};
-        *reinterpret_cast<short4v*>(
-            reinterpret_cast<short*>(ldsT + (skx + e) * SLDM + smx)) = pk;
+      for (int ii = 0; ii < 3; ++ii) {
+        const int krow = (gx0 + ii) * 8;
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          short4v pk = {vx[ii][0][e], vx[ii][1][e], vx[ii][2][e],
+                        vx[ii][3][e]};
+          *reinterpret_cast<short4v*>(
+              reinterpret_cast<short*>(ldsT + (krow + e) * SLDM + smx)) = pk;
+        }
       }
     }
   };
@@ -459,24 +470,31 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     for (int kk = 0; kk < 64; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
           ldsDyT + (i0 + li) * SLDM + kk + kh * 8);
-      const short8 bf = *reinterpret_cast<const short8*>(
-          ldsXT + (j0 + li) * SLDM + kk + kh * 8);
-      acc = SMfma<T16>::run(af, bf, acc);
+#pragma unroll
+      for (int jj = 0; jj < 3; ++jj) {
+        const short8 bf = *reinterpret_cast<const short8*>(
+            ldsXT + (j0 + jj * 64 + li) * SLDM + kk + kh * 8);
+        acc[jj] = SMfma<T16>::run(af, bf, acc[jj]);
+      }
     }
   }
 
-  // scatter the 32(ko) x 32(kg) fp32 tile into the [KO,3,7,7] slab;
+  // scatter the 32(ko) x 3x32(kg) fp32 tiles into the [KO,3,7,7] slab;
   // row-pad positions (jj >= 21) and r >= 7 are dropped
-  float* slab = dw + (long)blockIdx.z * ((long)KO * 3 * 7 * 7);
+  float* slab = dw + (long)blockIdx.y * ((long)KO * 3 * 7 * 7);
 #pragma unroll
-  for (int reg = 0; reg < 16; ++reg) {
-    const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
-    const int kg = c0 + j0 + li;
+  for (int jt = 0; jt < 3; ++jt) {
+    const int kg = j0 + jt * 64 + li;
     const int r = kg / SROW;
     const int jj = kg - r * SROW;
     if (jj < 21 && r < 7) {
       const int s = jj / 3, c = jj - s * 3;
-      slab[(((long)(k0 + i0 + i) * 3 + c) * 7 + r) * 7 + s] = acc[reg];
+#pragma unroll
+      for (int reg = 0; reg < 16; ++reg) {
+        const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
+        slab[(((long)(k0 + i0 + i) * 3 + c) * 7 + r) * 7 + s] =
+            acc[jt][reg];
+      }
     }
   }
 }
@@ -562,7 +580,7 @@ void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   nchunks = cdiv_l(M, m_per_chunk);
   auto part = at::empty({nchunks * E}, x.options().dtype(at::kFloat));
   at::Tensor zp = conv_zero_page(x);
-  dim3 grid(KO / 64, SKG / 64, (unsigned)nchunks);
+  dim3 grid(KO / 64, (unsigned)nchunks);
   DISPATCH_16(x, T16, {
     hipLaunchKernelGGL((conv_wgrad_stem_gemm<T16>), grid, dim3(256), 0,
                        cur_stream(), (const T16*)x.data_ptr(),
