@@ -240,7 +240,7 @@ def pmcprobe():
         O.ext().conv2d_wgrad(x7, dy7, 7, 7, 2, 3)
         O.ext().maxpool_fwd(xp, 3, 2, 1)
         O.ext().maxpool_bwd(yp.clone(), ip, 112, 112, 3, 2, 1)
-        O.ext().bn_apply(xb, mean, inv, g1, torch.zeros_like(g1), e, 1)
+        O.ext().bn_apply(xb, mean, inv, g1, torch.zeros_like(g1), e, 1, True)
         O.ext().conv2d_wgrad(xw, dyw, 3, 3, 1, 1)
         O.ext().conv2d_fwd(xg, wg, e, 1, 1, 0, 3, 3)
     torch.cuda.synchronize()
@@ -257,6 +257,8 @@ def bnperf():
     s = O.ext().bn_stats(x)
     mean = s[0] / x.numel() * 64
     invstd = torch.rsqrt(torch.ones(64, device="cuda"))
+    eb = torch.empty(0, device="cuda", dtype=torch.uint8)
+    ef = torch.empty(0, device="cuda")
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(50):
@@ -264,7 +266,7 @@ def bnperf():
     torch.cuda.synchronize()
     t1 = time.perf_counter()
     for _ in range(50):
-        O.ext().bn_bwd_reduce(x, dy, y, mean, invstd)
+        O.ext().bn_bwd_reduce(x, dy, y, mean, invstd, eb, ef, ef)
     torch.cuda.synchronize()
     t2 = time.perf_counter()
     mb = x.numel() * 2 / 1e6
@@ -308,9 +310,33 @@ if __name__ == "__main__":
         stemperf()
     if what == "membw":
         membw()
+    if what == "pmcs3":
+        pmcs3()
     if what == "pmcprobe":
         pmcprobe()
     if what == "r50fwd":
         r50_fwd_steps()
     if what in ("r50", "all"):
         r50()
+
+
+def pmcs3():
+    """Drive the r18 hot kernels (s3 wgrad + patch fwd/dgrad, b8192 layer
+    shapes) repeatedly for a rocprofv3 --pmc pass: is s3 LDS-array-bound
+    as the issue-cycle model predicts?"""
+    import mi355x.ops as O
+    e = torch.empty(0, device="cuda")
+    shapes = [(8192, 32, 64), (8192, 16, 128)]
+    for b, hw, c in shapes:
+        x = torch.randn(b, hw, hw, c, device="cuda").to(torch.bfloat16)
+        w = (torch.randn(c, 3, 3, c, device="cuda") * 0.1).to(torch.bfloat16)
+        wf = w.permute(1, 2, 3, 0).contiguous()
+        dy = torch.randn_like(x)
+        for _ in range(6):
+            O.ext().conv2d_wgrad(x, dy, 3, 3, 1, 1)            # s3
+            O.ext().conv2d_fwd(x, w, e, 1, 1, 0, 3, 3)         # patch fwd
+            O.ext().conv2d_dgrad(dy, wf, 1, 1, hw, hw,
+                                 torch.empty(0, device="cuda",
+                                             dtype=x.dtype))   # patch dgrad
+    torch.cuda.synchronize()
+    print("pmcs3 done")
