@@ -294,6 +294,28 @@ def wgradperf():
     print(f"wgrad layer1: {dt:.1f} us/call  ({gf / dt * 1e3:.0f} TF)")
 
 
+def pmcs3():
+    """Drive the r18 hot kernels (s3 wgrad + patch fwd/dgrad, b8192 layer
+    shapes) repeatedly for a rocprofv3 --pmc pass: is s3 LDS-array-bound
+    as the issue-cycle model predicts?"""
+    import mi355x.ops as O
+    e = torch.empty(0, device="cuda")
+    shapes = [(8192, 32, 64), (8192, 16, 128)]
+    for b, hw, c in shapes:
+        x = torch.randn(b, hw, hw, c, device="cuda").to(torch.bfloat16)
+        w = (torch.randn(c, 3, 3, c, device="cuda") * 0.1).to(torch.bfloat16)
+        wf = w.permute(1, 2, 3, 0).contiguous()
+        dy = torch.randn_like(x)
+        for _ in range(6):
+            O.ext().conv2d_wgrad(x, dy, 3, 3, 1, 1)            # s3
+            O.ext().conv2d_fwd(x, w, e, 1, 1, 0, 3, 3)         # patch fwd
+            O.ext().conv2d_dgrad(dy, wf, 1, 1, hw, hw,
+                                 torch.empty(0, device="cuda",
+                                             dtype=x.dtype))   # patch dgrad
+    torch.cuda.synchronize()
+    print("pmcs3 done")
+
+
 if __name__ == "__main__":
     what = sys.argv[1] if len(sys.argv) > 1 else "all"
     if what in ("probe", "all"):
@@ -318,25 +340,3 @@ if __name__ == "__main__":
         r50_fwd_steps()
     if what in ("r50", "all"):
         r50()
-
-
-def pmcs3():
-    """Drive the r18 hot kernels (s3 wgrad + patch fwd/dgrad, b8192 layer
-    shapes) repeatedly for a rocprofv3 --pmc pass: is s3 LDS-array-bound
-    as the issue-cycle model predicts?"""
-    import mi355x.ops as O
-    e = torch.empty(0, device="cuda")
-    shapes = [(8192, 32, 64), (8192, 16, 128)]
-    for b, hw, c in shapes:
-        x = torch.randn(b, hw, hw, c, device="cuda").to(torch.bfloat16)
-        w = (torch.randn(c, 3, 3, c, device="cuda") * 0.1).to(torch.bfloat16)
-        wf = w.permute(1, 2, 3, 0).contiguous()
-        dy = torch.randn_like(x)
-        for _ in range(6):
-            O.ext().conv2d_wgrad(x, dy, 3, 3, 1, 1)            # s3
-            O.ext().conv2d_fwd(x, w, e, 1, 1, 0, 3, 3)         # patch fwd
-            O.ext().conv2d_dgrad(dy, wf, 1, 1, hw, hw,
-                                 torch.empty(0, device="cuda",
-                                             dtype=x.dtype))   # patch dgrad
-    torch.cuda.synchronize()
-    print("pmcs3 done")
