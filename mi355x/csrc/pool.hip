@@ -206,7 +206,10 @@ __global__ void gap_kernel(const T16* __restrict__ x, T16* __restrict__ y,
   }
 }
 
-inline int pgrid(long n) { return (int)std::min<long>(cdiv_l(n, 256), 4096); }
+// 16k blocks (4M threads): the bwd gather is latency-bound (PMC: 33%
+// issue-stall, 49% parked at the 4096 cap) — more resident waves per
+// element-iteration hide more of the 2-4 window round trips
+inline int pgrid(long n) { return (int)std::min<long>(cdiv_l(n, 256), 16384); }
 
 }  // namespace
 
