@@ -1061,7 +1061,12 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
 // slots that cross a q-row boundary (q==0 for s=0, q==Wo-1 for s=2).
 // Triples the MFMA work per staged dy/x byte — the plain kernel was
 // staging-bound at ~190 TF while the fwd gather-GEMM reaches 470-670.
-template <typename T16, int KT, bool SCALED = false>
+// KMIN (KT=64 only): k-minor staging-thread map + rotate-swizzled LDS
+// columns — one wave's gather covers 8 full m-lines instead of 16
+// quarter-lines (PMC: 53% parked on the gather latency), and the rotation
+// keeps the transposed b64 writes bank-conflict-free where a plain
+// k-minor map would collide 8-way (8*LDM elements = 0 mod 32 banks).
+template <typename T16, int KT, bool SCALED = false, bool KMIN = false>
 __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
@@ -1084,11 +1089,11 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
   const bool do_dy = tid < 2 * KT;
   const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
   const int t = do_dy ? tid : 0;
-  const int sm = (t & 15) * 4;
-  const int sk = (t >> 4) * 8;
+  const int sm = KMIN ? ((t >> 3) & 15) * 4 : (t & 15) * 4;
+  const int sk = KMIN ? (t & 7) * 8 : (t >> 4) * 8;
   const int tx = tid & 127;
-  const int smx = (tx & 15) * 4;
-  const int skx = (tx >> 4) * 8;
+  const int smx = KMIN ? (tx >> 3) * 4 : (tx & 15) * 4;
+  const int skx = KMIN ? (tx & 7) * 8 : (tx >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -1185,8 +1190,9 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+        const int col = KMIN ? ((sm + ((sk + e) & 56)) & 63) : sm;
         *reinterpret_cast<short4v*>(
-            reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+            reinterpret_cast<short*>(lds + (sk + e) * LDM + col)) = pk;
       }
     }
     if (do_x) {
@@ -1207,8 +1213,9 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
                         (short)(vx[s + 1][e] & keep[1]),
                         (short)(vx[s + 2][e] & keep[2]),
                         (short)(vx[s + 3][e] & keep[3])};
+          const int col = KMIN ? ((smx + ((skx + e) & 56)) & 63) : smx;
           *reinterpret_cast<short4v*>(
-              reinterpret_cast<short*>(ldsS + (skx + e) * LDM + smx)) = pk;
+              reinterpret_cast<short*>(ldsS + (skx + e) * LDM + col)) = pk;
         }
       }
     }
@@ -1221,17 +1228,21 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
     __syncthreads();
     if (m0 + WGM < m_end) load_m(m0 + WGM);
     const T16* ldsDyT = lds;
+    const int arow = i0 + li;
+    const int arot = KMIN ? (arow & 56) : 0;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
-          ldsDyT + (i0 + li) * LDM + kk + kh * 8);
+          ldsDyT + arow * LDM + ((kk + kh * 8 + arot) & 63));
 #pragma unroll
       for (int s = 0; s < 3; ++s) {
         const T16* ldsXT = lds + (KT + s * 64) * LDM;
 #pragma unroll
         for (int jj = 0; jj < NJ; ++jj) {
+          const int brow = j0base + jj * 32 + li;
           const short8 bf = *reinterpret_cast<const short8*>(
-              ldsXT + (j0base + jj * 32 + li) * LDM + kk + kh * 8);
+              ldsXT + brow * LDM +
+              ((kk + kh * 8 + (KMIN ? (brow & 56) : 0)) & 63));
           acc[s][jj] = Mfma32<T16>::run(af, bf, acc[s][jj]);
         }
       }
@@ -1717,10 +1728,21 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     at::Tensor part = nchunks > 1
                           ? at::empty({nchunks, E}, dw.options())
                           : dw;
+    static const bool kminor = [] {  // A/B knob (see kernel comment)
+      const char* e = getenv("MI355X_S3_KMINOR");
+      return e && e[0] == '1';
+    }();
     DISPATCH_16(x, T16, {
       if (KT == 128)
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 128>), grid, dim3(256),
                            0, cur_stream(), (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
+      else if (kminor && !asc_p)
+        hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64, false, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
