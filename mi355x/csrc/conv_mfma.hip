@@ -1728,9 +1728,9 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     at::Tensor part = nchunks > 1
                           ? at::empty({nchunks, E}, dw.options())
                           : dw;
-    static const bool kminor = [] {  // A/B knob (see kernel comment)
-      const char* e = getenv("MI355X_S3_KMINOR");
-      return e && e[0] == '1';
+    static const bool kminor = [] {  // default ON (+0.5% r18, interleaved
+      const char* e = getenv("MI355X_S3_KMINOR");  // A/B x2); =0 disables
+      return !e || e[0] != '0';
     }();
     DISPATCH_16(x, T16, {
       if (KT == 128)
