@@ -50,7 +50,8 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 15, 15, 128), 128, 3, 2, 1, False, None), # t128 3x3/s2/p1 (r50 conv2)
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
     ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
-    ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # stem7 7x7/2 (dot2 kernel)
+    ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # stem7 GEMM (strip: Ho*Wo%128==0)
+    ((2, 30, 30, 3), 64, 7, 2, 3, False, None),    # stem7 GEMM v2 fallback (15x15 out)
     ((2, 29, 29, 3), 32, 7, 2, 3, False, None),    # stem7 odd W, K<64
     ((2, 10, 10, 24), 64, 3, 1, 1, False, None),   # GENC C=24
 ])
