@@ -913,7 +913,7 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
 // machinery as conv_wgrad_mfma_kernel.
 // SCALED: min 3 waves/SIMD — the Sc8 registers pushed the allocation 4
 // VGPRs over the 168 boundary (occ 3 -> 2, measured 1.77x slower)
-template <typename T16, bool SCALED = false>
+template <typename T16, bool SCALED = false, bool KMIN = false>
 __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
     const T16* __restrict__ x, const T16* __restrict__ dy,
     const float* __restrict__ asc,  // SCALED only
@@ -935,9 +935,10 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
   const long m_end = min(Mtot, m_begin + m_per_chunk);
 
   // staging: all 256 threads stage dy (4m x 8k over 128 k-rows) and x
-  // (4m x 8c over 128 c-rows) with the same (m-group, row-group) map
-  const int sm = (tid & 15) * 4;
-  const int sk = (tid >> 4) * 8;
+  // (4m x 8c over 128 c-rows) with the same (m-group, row-group) map.
+  // KMIN: k-minor map + rotate-swizzled columns (see conv_wgrad_mfma_s3)
+  const int sm = KMIN ? (tid >> 4) * 4 : (tid & 15) * 4;
+  const int sk = KMIN ? (tid & 15) * 8 : (tid >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -1000,12 +1001,13 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
   auto stage_m = [&]() {
 #pragma unroll
     for (int e = 0; e < 8; ++e) {
+      const int col = KMIN ? ((sm + ((sk + e) & 56)) & 63) : sm;
       short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
       *reinterpret_cast<short4v*>(
-          reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+          reinterpret_cast<short*>(lds + (sk + e) * LDM + col)) = pk;
       short4v px0 = {vx0[0][e], vx0[1][e], vx0[2][e], vx0[3][e]};
       *reinterpret_cast<short4v*>(reinterpret_cast<short*>(
-          lds + (128 + sk + e) * LDM + sm)) = px0;
+          lds + (128 + sk + e) * LDM + col)) = px0;
     }
   };
 
@@ -1021,13 +1023,19 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
     for (int kk = 0; kk < WGM; kk += 16) {
       short8 af[2], bf[2];
 #pragma unroll
-      for (int a = 0; a < 2; ++a)
+      for (int a = 0; a < 2; ++a) {
+        const int row = i0 + a * 64 + li;
         af[a] = *reinterpret_cast<const short8*>(
-            ldsDyT + (i0 + a * 64 + li) * LDM + kk + kh * 8);
+            ldsDyT + row * LDM +
+            ((kk + kh * 8 + (KMIN ? (row & 56) : 0)) & 63));
+      }
 #pragma unroll
-      for (int b = 0; b < 2; ++b)
+      for (int b = 0; b < 2; ++b) {
+        const int row = j0 + b * 64 + li;
         bf[b] = *reinterpret_cast<const short8*>(
-            ldsXT + (j0 + b * 64 + li) * LDM + kk + kh * 8);
+            ldsXT + row * LDM +
+            ((kk + kh * 8 + (KMIN ? (row & 56) : 0)) & 63));
+      }
 #pragma unroll
       for (int a = 0; a < 2; ++a)
 #pragma unroll
@@ -1780,12 +1788,24 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     at::Tensor part = nchunks > 1
                           ? at::empty({nchunks, E}, dw.options())
                           : dw;
+    static const bool kmin128 = [] {
+      const char* e = getenv("MI355X_T128_KMINOR");
+      return e && e[0] == '1';
+    }();
     DISPATCH_16(x, T16, {
       if (asc_p)
         hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else if (kmin128)
+        hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16, false, true>), grid,
+                           dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
