@@ -1788,9 +1788,9 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
     at::Tensor part = nchunks > 1
                           ? at::empty({nchunks, E}, dw.options())
                           : dw;
-    static const bool kmin128 = [] {
-      const char* e = getenv("MI355X_T128_KMINOR");
-      return e && e[0] == '1';
+    static const bool kmin128 = [] {  // default ON (+0.8% r50,
+      const char* e = getenv("MI355X_T128_KMINOR");  // interleaved A/B x2)
+      return !e || e[0] != '0';
     }();
     DISPATCH_16(x, T16, {
       if (asc_p)
