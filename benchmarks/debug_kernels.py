@@ -312,6 +312,19 @@ def pmcs3():
             O.ext().conv2d_dgrad(dy, wf, 1, 1, hw, hw,
                                  torch.empty(0, device="cuda",
                                              dtype=x.dtype))   # patch dgrad
+    # round-2 kernels: stem strip fwd / merged wgrad, parity-s2 dgrad
+    xs = torch.randn(256, 224, 224, 3, device="cuda").to(torch.bfloat16)
+    ws = (torch.randn(64, 7, 7, 3, device="cuda") * 0.1).to(torch.bfloat16)
+    dys = torch.randn(256, 112, 112, 64, device="cuda").to(torch.bfloat16)
+    xd = torch.randn(512, 28, 28, 128, device="cuda").to(torch.bfloat16)
+    wd = (torch.randn(256, 3, 3, 128, device="cuda") * 0.1).to(torch.bfloat16)
+    wdf = wd.permute(1, 2, 3, 0).contiguous()
+    dyd = torch.randn(512, 14, 14, 256, device="cuda").to(torch.bfloat16)
+    ea = torch.empty(0, device="cuda", dtype=torch.bfloat16)
+    for _ in range(6):
+        O.ext().conv2d_fwd(xs, ws, e, 2, 3, 0, 7, 7)      # stem strip
+        O.ext().conv2d_wgrad(xs, dys, 7, 7, 2, 3)          # stem wgrad
+        O.ext().conv2d_dgrad(dyd, wdf, 2, 1, 28, 28, ea)   # parity-s2
     torch.cuda.synchronize()
     print("pmcs3 done")
 
