@@ -417,11 +417,21 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
   f32x16 acc[NT] = {};
   short8 sa[4], sb[EPT / 8];
 
+  // incremental (c-chunk, s, r) decode: runtime divisions per k-step
+  // measured as 38% issue-stall (PMC) in this short-k kernel
+  int cc_i = 0, s_i = 0, r_i = 0;
   auto load_step = [&](int j) {
-    const int c0 = (j % cchunks) * BK;
-    const int t = j / cchunks;
-    const int s_ = s0 + 2 * (t % ns);
-    const int r_ = r0 + 2 * (t / ns);
+    (void)j;
+    const int c0 = cc_i * BK;
+    const int s_ = s0 + 2 * s_i;
+    const int r_ = r0 + 2 * r_i;
+    if (++cc_i == cchunks) {
+      cc_i = 0;
+      if (++s_i == ns) {
+        s_i = 0;
+        ++r_i;
+      }
+    }
     const int dh = ih_ + pad - r_;  // even by construction
     const int dw_ = iw_ + pad - s_;
     const int pp = dh >> 1, qq = dw_ >> 1;
@@ -473,16 +483,28 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
     }
   }
 
-  // epilogue: scatter into the strided class positions of dx
+  // epilogue: scatter into the strided class positions of dx. One
+  // image-decode per lane; per register only a rem-walk + one div by Wa
+  // (the per-reg /(Ha*Wa) pair was the other half of the issue stalls)
+  const long m_base = bm0 + wm;
+  const int HaWa = Ha * Wa;
+  int nn0 = 0, rem0 = 0;
+  if (m_base < Mc) {
+    nn0 = (int)(m_base / HaWa);
+    rem0 = (int)(m_base % HaWa);
+  }
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
     const int row = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
-    const long m_out = bm0 + wm + row;
+    const long m_out = m_base + row;
     if (m_out < Mc) {
-      const int nn = (int)(m_out / ((long)Ha * Wa));
-      const int rem = (int)(m_out % ((long)Ha * Wa));
+      int nn = nn0, rem = rem0 + row;
+      while (rem >= HaWa) {
+        rem -= HaWa;
+        ++nn;
+      }
       const int ih = 2 * (rem / Wa) + pa;
-      const int iw = 2 * (rem % Wa) + pb;
+      const int iw = 2 * (rem - (rem / Wa) * Wa) + pb;
       const long off = (((long)nn * H + ih) * W + iw) * CI;
 #pragma unroll
       for (int tnt = 0; tnt < NT; ++tnt) {
