@@ -188,14 +188,26 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
       sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
   };
 
+  // incremental (c-chunk, s, r) decode — load_step is called exactly once
+  // per j in order, so three runtime divisions per k-step reduce to carry
+  // counters (same lever as the wgrad kernels' m-decode)
+  int cc_i = 0, s_i = 0, r_i = 0;
   auto load_step = [&](int j) {
     if constexpr (GENC) {
       load_step_genc(j);
       return;
     }
-    const int c0 = (j % cchunks) * BK;
-    const int s_ = (j / cchunks) % S;
-    const int r_ = j / (cchunks * S);
+    (void)j;
+    const int c0 = cc_i * BK;
+    const int s_ = s_i;
+    const int r_ = r_i;
+    if (++cc_i == cchunks) {
+      cc_i = 0;
+      if (++s_i == S) {
+        s_i = 0;
+        ++r_i;
+      }
+    }
     bool va = m_ok;
     long ioff = 0;
     if constexpr (TRANS) {
