@@ -641,16 +641,13 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
 
   for (int cc = 0; cc < cchunks; ++cc) {
     __syncthreads();  // previous c-chunk's patch reads done
-    // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi.
-    // (col,row) tracked incrementally (constant stride 32 per iteration)
-    // — the per-iteration runtime div/mod pair by (Wi+2) was measured
-    // issue-stall time on this 27%-of-step kernel ----
+    // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi ----
     const int ngroups = NR * (Wi + 2) * (BK / 8);
-    const int W2 = Wi + 2;
-    const int g = tid & 7;             // BK/8 == 8
-    int col = (tid >> 3) % W2;
-    int row = (tid >> 3) / W2;
     for (int i = tid; i < ngroups; i += 256) {
+      const int g = i % (BK / 8);
+      const int ce = i / (BK / 8);
+      const int col = ce % (Wi + 2);
+      const int row = ce / (Wi + 2);
       const long gr = gr0 + row;
       const int iw = col - 1;
       short8 v = {};
@@ -661,13 +658,8 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
         if constexpr (SCALED)
           v = scale8<T16>(v, sc8_load(asc, ash, cc * BK + g * 8));
       }
-      *reinterpret_cast<short8*>(patch + ((long)row * W2 + col) * PCS +
+      *reinterpret_cast<short8*>(patch + ((long)row * (Wi + 2) + col) * PCS +
                                  g * 8) = v;
-      col += 32;  // 256 threads / 8 groups per (row,col) cell
-      while (col >= W2) {
-        col -= W2;
-        ++row;
-      }
     }
     __syncthreads();
 
