@@ -261,11 +261,9 @@ __global__ __launch_bounds__(256) void bn_apply_fast(
   // two batched sub-streams per iteration (same in-flight-bytes rationale
   // as bn_reduce_fast; the channel group is stride-invariant)
   long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  for (; e + 3 * stride < E; e += 4 * stride) {
+  for (; e + stride < E; e += 2 * stride) {
     body(e);
     body(e + stride);
-    body(e + 2 * stride);
-    body(e + 3 * stride);
   }
   for (; e < E; e += stride) body(e);
 }
@@ -316,11 +314,9 @@ __global__ __launch_bounds__(256) void bn_bwd_dx_fast(
     if (dres) *reinterpret_cast<short8*>(dres + e) = ods;
   };
   long e = (long)(blockIdx.x * blockDim.x + threadIdx.x) * 8;
-  for (; e + 3 * stride < E; e += 4 * stride) {
+  for (; e + stride < E; e += 2 * stride) {
     body(e);
     body(e + stride);
-    body(e + 2 * stride);
-    body(e + 3 * stride);
   }
   for (; e < E; e += stride) body(e);
 }
