@@ -8,6 +8,7 @@ mi355x kernels on GPU (bf16/fp16) and the plain-torch fp32 path on CPU.
 from __future__ import annotations
 
 import math
+import os
 
 import torch
 from torch import nn
@@ -121,8 +122,6 @@ class BatchNorm2d(nn.Module):
                               self.process_group if self.training else None,
                               stats)
 
-
-import os
 
 def _fuse_stats():
     # default ON since round 2: with the shuffle-based stats fold, the
