@@ -132,11 +132,10 @@ def _fuse_stats():
 
 
 def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
-    """conv -> BN(+residual+ReLU). With MI355X_FUSE_BN=1 the BN batch
-    statistics come from the conv epilogue (skipping the separate
-    full-tensor bn_stats pass) — measured NEUTRAL-to-NEGATIVE on r18-CIFAR
-    (69.6k -> 54.4k img/s) and mildly positive on r50-224 (3.9k -> 4.15k),
-    so it defaults OFF pending investigation of the epilogue cost."""
+    """conv -> BN(+residual+ReLU). With MI355X_FUSE_BN=1 (default) the BN
+    batch statistics come from the conv epilogue, skipping the separate
+    full-tensor bn_stats pass (+0.9% r18 / +2.8% r50 after the round-2
+    shuffle fold + two-level slab reduce; see _fuse_stats)."""
     if (_fuse_stats() and x.is_cuda and bn.training and conv.bias is None
             and conv.act is None
             # only shapes with a stats-emitting MFMA path: C % 64 == 0
