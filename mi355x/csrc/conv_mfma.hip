@@ -529,7 +529,10 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
   }
 }
 
+}  // namespace
+
 // fold slab [by][gx][bnt2] -> stats[2,C]; two-level when gx is large
+// (external linkage: the stem kernels reuse it)
 void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
                        int C) {
   const int by = (int)(slab.numel() / ((long)gx * bnt2));
@@ -549,8 +552,6 @@ void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
                      slab.data_ptr<float>(), stats.data_ptr<float>(), gx,
                      bnt2, C, bpb);
 }
-
-}  // namespace
 
 // ---------------------------------------------------------------------------
 // ---------------------------------------------------------------------------

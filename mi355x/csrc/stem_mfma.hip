@@ -215,7 +215,9 @@ template <typename T16>
 __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
     const T16* __restrict__ x,    // [N, H, W, 3]
     const T16* __restrict__ w24,  // [KO, 192] row-padded
-    const float* __restrict__ bias, T16* __restrict__ y, const int N,
+    const float* __restrict__ bias,
+    float* __restrict__ stats_slab,  // null, or per-block (sum,sumsq) rows
+    T16* __restrict__ y, const int N,
     const int H, const int W, const int KO, const int Ho, const int Wo,
     const int pad, const int act, const int has_bias, const int nrows,
     const int sstride) {
@@ -319,6 +321,7 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
   float bv[2];
   bv[0] = has_bias ? bias[k0 + li] : 0.f;
   bv[1] = has_bias ? bias[k0 + 32 + li] : 0.f;
+  float ssum[2] = {}, ssq[2] = {};
   const long Mtot = (long)N * Mimg;
 #pragma unroll
   for (int reg = 0; reg < 16; ++reg) {
@@ -329,9 +332,36 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
       for (int t2 = 0; t2 < 2; ++t2) {
         float v = acc[t2][reg] + bv[t2];
         if (act == 1) v = fmaxf(v, 0.f);
+        if (stats_slab) {
+          ssum[t2] += v;
+          ssq[t2] += v * v;
+        }
         y[m_out * KO + k0 + t2 * 32 + li] = F16<T16>::from_f32(v);
       }
     }
+  }
+  if (stats_slab) {
+    // same fold as conv_gather_gemm: kh halves by xor-shuffle, waves by
+    // per-wave LDS rows (strip LDS reused after a barrier)
+#pragma unroll
+    for (int t2 = 0; t2 < 2; ++t2) {
+      ssum[t2] += __shfl_xor(ssum[t2], 32, 64);
+      ssq[t2] += __shfl_xor(ssq[t2], 32, 64);
+    }
+    float* lsum = reinterpret_cast<float*>(ssmem);  // [4 waves][128]
+    __syncthreads();
+    if (kh == 0) {
+#pragma unroll
+      for (int t2 = 0; t2 < 2; ++t2) {
+        lsum[wave * 128 + t2 * 32 + li] = ssum[t2];
+        lsum[wave * 128 + 64 + t2 * 32 + li] = ssq[t2];
+      }
+    }
+    __syncthreads();
+    float* slab =
+        stats_slab + ((long)blockIdx.y * gridDim.x + blockIdx.x) * 2 * 64;
+    for (int t = tid; t < 2 * 64; t += 256)
+      slab[t] = lsum[t] + lsum[128 + t] + lsum[256 + t] + lsum[384 + t];
   }
 }
 
@@ -515,8 +545,14 @@ static at::Tensor stem_w24(const at::Tensor& w) {
       .contiguous();
 }
 
-void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
-                               at::Tensor y, long pad, long act) {
+void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
+                       int C);  // conv_mfma.hip
+
+// returns true when `stats` ([2,KO], zeroed) was filled from the conv
+// epilogue (strip path only; the v2 fallback leaves it to the caller)
+bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                               at::Tensor y, long pad, long act,
+                               at::Tensor stats) {
   const int N = x.size(0), H = x.size(1), W = x.size(2);
   const int KO = w.size(0);
   const int Ho = y.size(1), Wo = y.size(2);
@@ -543,15 +579,25 @@ void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
     const size_t smem =
         ((size_t)(nrows + 1) * sstride + 64 * (SKG + 8)) * x.element_size();
     if (smem <= 150 * 1024) {
+      const bool want_stats = stats.defined() && stats.numel() > 0;
+      at::Tensor slab;
+      float* slab_p = nullptr;
+      if (want_stats) {
+        slab = at::empty({(long)grid.y * grid.x * 128},
+                         x.options().dtype(at::kFloat));
+        slab_p = slab.data_ptr<float>();
+      }
       DISPATCH_16(x, T16, {
         hipLaunchKernelGGL((conv_fwd_stem_strip<T16>), grid, dim3(256),
                            smem, cur_stream(), (const T16*)x.data_ptr(),
                            (const T16*)w24.data_ptr(),
                            has_bias ? bias.data_ptr<float>() : nullptr,
-                           (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo,
+                           slab_p, (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo,
                            (int)pad, (int)act, has_bias, nrows, sstride);
       });
-      return;
+      if (slab_p)
+        stats_slab_reduce(slab, stats, grid.x, 128, KO);
+      return want_stats;
     }
   }
   at::Tensor zp = conv_zero_page(x);
@@ -563,6 +609,7 @@ void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                        (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N, H,
                        W, KO, Ho, Wo, (int)pad, (int)act, has_bias);
   });
+  return false;
 }
 
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);

@@ -232,13 +232,18 @@ def conv_bn_tap(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
     """conv -> BN returning (out, tap). The block's shortcut consumes
     `tap` instead of x directly, so the residual-junction gradient fuses
     into this conv's dgrad epilogue (ops.functional._ConvTapFn) instead
-    of a separate full-tensor add. Falls back to plain conv_bn semantics
-    (tap = x) on CPU / non-MFMA shapes / MI355X_TAP=0."""
+    of a separate full-tensor add; the conv epilogue also emits the BN
+    statistics (no separate bn_stats pass). Falls back to plain conv_bn
+    semantics (tap = x) on CPU / non-MFMA shapes / MI355X_TAP=0."""
     if not (_tap_on() and x.is_cuda and conv.bias is None
             and conv.act is None):
         return conv_bn(conv, bn, x, residual=residual), x
     from mi355x.ops import functional as F_
 
+    if _fuse_stats() and bn.training and conv.weight.shape[1] % 64 == 0:
+        y, tap, stats = F_.conv2d_tap_stats(x, conv.weight, conv.stride,
+                                            conv.padding)
+        return bn(y, residual=residual, stats=stats), tap
     y, tap = F_.conv2d_tap(x, conv.weight, conv.stride, conv.padding)
     return bn(y, residual=residual), tap
 
