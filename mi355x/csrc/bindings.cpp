@@ -19,7 +19,8 @@ at::Tensor conv2d_dgrad(at::Tensor dy, at::Tensor wflip, long stride,
 at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                         long stride, long pad);
 std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
-                                         long stride, long pad);
+                                         long stride, long pad, long kR,
+                                         long kS);
 std::vector<at::Tensor> conv2d_fwd_scaled(at::Tensor x, at::Tensor w,
                                           at::Tensor asc, at::Tensor ash,
                                           long stride, long pad,

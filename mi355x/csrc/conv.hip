@@ -831,8 +831,9 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                             long R, long S, long stride, long pad,
                             at::Tensor asc, at::Tensor ash);
 // stem_mfma.hip — MFMA GEMM stem (7x7/s2/C=3), replaces the dot2 kernels
-void conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
-                               at::Tensor y, long pad, long act);
+bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                               at::Tensor y, long pad, long act,
+                               at::Tensor stats);
 void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                                  long pad);
 
@@ -860,7 +861,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   const bool stem7 = K <= 64 && C == 3 && R == 7 && S == 7 && stride == 2 &&
                      pad <= 3;
   if (stem7 && w.dim() == 2 && K % 64 == 0 && stem_gemm_on()) {
-    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act);
+    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act, at::Tensor());
     return y;
   }
   if (stem7) {
@@ -959,20 +960,34 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
 
 // conv->BN fusion entry: returns {y, stats[2,C]} with stats computed in
 // the conv epilogue (empty stats if this shape has no fusable path — the
-// caller then runs the separate bn_stats pass)
+// caller then runs the separate bn_stats pass). kR/kS: filter dims for
+// dim-2 (pre-padded) weights, so the fallback dispatch stays correct.
 std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
-                                         long stride, long pad) {
+                                         long stride, long pad, long kR,
+                                         long kS) {
   CHECK_GPU(x);
   CHECK_CONTIG(x);
   CHECK_16BIT(x);
   CHECK_CONTIG(w);
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   const int K = w.size(0);
+  // ImageNet stem (7x7/s2/C=3, dim-2 padded weight): the strip kernel's
+  // epilogue can emit stats too
+  if (w.dim() == 2 && C == 3 && kR == 7 && kS == 7 && stride == 2 &&
+      pad <= 3 && K % 64 == 0 && stem_gemm_on()) {
+    const int P = out_dim(H, 7, stride, pad), Q = out_dim(W, 7, stride, pad);
+    auto y = at::empty({N, P, Q, K}, x.options());
+    auto stats = at::zeros({2, K}, x.options().dtype(at::kFloat));
+    auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
+    const bool got =
+        conv_fwd_stem_gemm_launch(x, w, empty_bias, y, pad, 0, stats);
+    return {y, got ? stats : at::Tensor()};
+  }
   if (w.dim() != 4 || !conv_mfma_supported(C, K)) {
     auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
     auto y = conv2d_fwd(x, w, empty_bias, stride, pad, 0,
-                        w.dim() == 4 ? w.size(1) : 0,
-                        w.dim() == 4 ? w.size(2) : 0);
+                        w.dim() == 4 ? w.size(1) : kR,
+                        w.dim() == 4 ? w.size(2) : kS);
     return {y, at::Tensor()};
   }
   const int R = w.size(1), S = w.size(2);

@@ -138,10 +138,11 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
     shuffle fold + two-level slab reduce; see _fuse_stats)."""
     if (_fuse_stats() and x.is_cuda and bn.training and conv.bias is None
             and conv.act is None
-            # only shapes with a stats-emitting MFMA path: C % 64 == 0
-            # (the C=3 stem would fall off its dedicated kernel and onto
-            # the slow per-element GENC gather — measured 7.3 ms/call)
-            and conv.weight.shape[1] % 64 == 0):
+            # shapes with a stats-emitting MFMA path: C % 64 == 0 convs
+            # and the 7x7/s2 ImageNet stem (strip-kernel epilogue stats)
+            and (conv.weight.shape[1] % 64 == 0
+                 or (conv.weight.shape[1] == 3
+                     and conv.weight.shape[2] == 7 and conv.stride == 2))):
         from mi355x.ops import functional as F_
 
         bn._nbt += 1

@@ -131,7 +131,9 @@ class _ConvFn(torch.autograd.Function):
         w16 = _w16_conv(weight, dtype)
         stats = None
         if want_stats:
-            y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding)
+            y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding,
+                                              weight.shape[2],
+                                              weight.shape[3])
         else:
             b32 = bias.detach().float() if bias is not None else torch.empty(0, device=x.device)
             y = ext().conv2d_fwd(x, w16, b32, stride, padding, act,
@@ -230,7 +232,8 @@ class _ConvTapStatsFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, stride, padding):
         w16 = _w16_conv(weight, x.dtype)
-        y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding)
+        y, stats = ext().conv2d_fwd_stats(x, w16, stride, padding,
+                                          weight.shape[2], weight.shape[3])
         if stats is None:
             stats = torch.empty(0, device=x.device, dtype=torch.float32)
         ctx.save_for_backward(x, w16)

@@ -310,3 +310,15 @@ def test_resnet_block_fused_vs_unfused_training_step():
     y_ref = blk2.bn2(blk2.conv2(out), residual=x)
     torch.testing.assert_close(y_fused.float(), y_ref.float(), rtol=5e-2,
                                atol=5e-2)
+
+
+def test_stem_epilogue_stats_match_bn_stats():
+    """The strip stem kernel's epilogue BN statistics equal a separate
+    bn_stats pass over its output (conv->BN fusion for the 7x7/s2 stem)."""
+    g = torch.Generator().manual_seed(11)
+    x = torch.randn(4, 64, 64, 3, generator=g).cuda().to(torch.bfloat16)
+    w = (torch.randn(64, 3, 7, 7, generator=g) * 0.1).cuda()
+    y, stats = fn.conv2d_with_stats(x, w, stride=2, padding=3)
+    assert stats is not None and stats.numel() == 128, "stem stats missing"
+    ref = ops.ext().bn_stats(y.detach().contiguous())
+    torch.testing.assert_close(stats, ref, rtol=1e-3, atol=1e-2)
