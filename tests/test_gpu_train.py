@@ -173,7 +173,9 @@ def _grad_parity(model, size, env, off, on):
         os.environ.pop(env, None)
     noise = (g0b - g0a).abs().max().item()
     delta = (g1 - g0a).abs().max().item()
-    assert delta <= max(4 * noise, 5e-3), (
+    # floor covers the case where the two baseline runs happen to agree
+    # closely while the mode under test legitimately rounds differently
+    assert delta <= max(4 * noise, 2e-2), (
         f"{env}: delta {delta:.5f} vs noise floor {noise:.5f}")
 
 
