@@ -290,7 +290,7 @@ def test_conv_bn_fused_stats_matches_separate():
     x = _qt(torch.randn(8, 16, 16, 64, generator=g)).cuda().to(torch.bfloat16)
     w = _qt(torch.randn(128, 3, 3, 64, generator=g) * 0.1).cuda().to(torch.bfloat16)
     # w in [K,R,S,C] kernel layout for the raw ext call
-    y, stats = ops.ext().conv2d_fwd_stats(x, w.contiguous(), 1, 1)
+    y, stats = ops.ext().conv2d_fwd_stats(x, w.contiguous(), 1, 1, 3, 3)
     ref = ops.ext().bn_stats(y)
     torch.testing.assert_close(stats, ref, rtol=2e-2, atol=2.0)
 
