@@ -765,12 +765,13 @@ constexpr int LDM = 72;   // m-minor row length (+16B: alignment + banks)
 
 // KT = KO-tile width (64 or 128): wider k-tiles amortize the x staging
 // and double the MFMA work per m-step for the K>=128 layers.
-template <typename T16, int KT, bool SCALED = false>
+template <typename T16, int KT, bool SCALED = false, bool KMIN = false>
 // the second launch-bounds arg (min 2 blocks/CU) is load-bearing: without
 // it the compiler allocates 188-256 VGPRs (occupancy 1-2) for no speedup;
 // with it 104-124 VGPRs, zero spills, occupancy 4 - the extra waves are
 // what covers the gather's HBM latency (PMC: 66% WAIT_ANY at occ 2).
 // SCALED variants pin the unscaled occupancy (the Sc8 regs cost a tier).
+// KMIN: k-minor staging + rotate-swizzled columns (conv_wgrad_mfma_s3).
 __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
@@ -798,11 +799,12 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
   const bool do_dy = tid < 2 * KT;
   const bool do_x = KT == 64 ? tid >= 128 : tid < 128;
   const int t = do_dy ? tid : 0;
-  const int sm = (t & 15) * 4;           // m offset (4 rows)
-  const int sk = (t >> 4) * 8;           // k-element offset (dyT rows)
+  constexpr int KG = KT / 8;             // k-groups per operand row set
+  const int sm = KMIN ? (t / KG) * 4 : (t & 15) * 4;   // m offset (4 rows)
+  const int sk = KMIN ? (t % KG) * 8 : (t >> 4) * 8;   // k offset (dyT rows)
   const int tx = tid & 127;
-  const int smx = (tx & 15) * 4;
-  const int skx = (tx >> 4) * 8;
+  const int smx = KMIN ? (tx >> 3) * 4 : (tx & 15) * 4;
+  const int skx = KMIN ? (tx & 7) * 8 : (tx >> 4) * 8;
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -885,8 +887,9 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         short4v pk = {vdy[0][e], vdy[1][e], vdy[2][e], vdy[3][e]};
+        const int col = KMIN ? ((sm + ((sk + e) & 56)) & 63) : sm;
         *reinterpret_cast<short4v*>(
-            reinterpret_cast<short*>(lds + (sk + e) * LDM + sm)) = pk;
+            reinterpret_cast<short*>(lds + (sk + e) * LDM + col)) = pk;
       }
     }
     if (do_x) {
@@ -894,8 +897,9 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
 #pragma unroll
       for (int e = 0; e < 8; ++e) {
         short4v pk = {vx[0][e], vx[1][e], vx[2][e], vx[3][e]};
+        const int col = KMIN ? ((smx + ((skx + e) & 56)) & 63) : smx;
         *reinterpret_cast<short4v*>(
-            reinterpret_cast<short*>(ldsT + (skx + e) * LDM + smx)) = pk;
+            reinterpret_cast<short*>(ldsT + (skx + e) * LDM + col)) = pk;
       }
     }
   };
@@ -908,14 +912,18 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
     if (m0 + WGM < m_end) load_m(m0 + WGM);
     const T16* ldsDyT = lds;
     const T16* ldsXT = lds + KT * LDM;
+    const int arow = i0 + li;
+    const int arot = KMIN ? (arow & 56) : 0;
 #pragma unroll
     for (int kk = 0; kk < WGM; kk += 16) {
       const short8 af = *reinterpret_cast<const short8*>(
-          ldsDyT + (i0 + li) * LDM + kk + kh * 8);
+          ldsDyT + arow * LDM + ((kk + kh * 8 + arot) & 63));
 #pragma unroll
       for (int jj = 0; jj < NJ; ++jj) {
+        const int brow = j0base + jj * 32 + li;
         const short8 bf = *reinterpret_cast<const short8*>(
-            ldsXT + (j0base + jj * 32 + li) * LDM + kk + kh * 8);
+            ldsXT + brow * LDM +
+            ((kk + kh * 8 + (KMIN ? (brow & 56) : 0)) & 63));
         acc[jj] = Mfma32<T16>::run(af, bf, acc[jj]);
       }
     }
@@ -1875,6 +1883,10 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   at::Tensor part = nchunks > 1
                         ? at::empty({nchunks, E}, dw.options())
                         : dw;
+  static const bool kming = [] {  // shares the s3 knob family
+    const char* e = getenv("MI355X_WGRAD_KMINOR");
+    return !e || e[0] != '0';
+  }();
   DISPATCH_16(x, T16, {
     if (KT == 128) {
       if (asc_p)
@@ -1882,6 +1894,14 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else if (kming)
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128, false, true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1899,6 +1919,14 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
                            (const T16*)dy.data_ptr(), asc_p, ash_p,
+                           part.data_ptr<float>(),
+                           N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
+                           (int)stride, (int)pad, m_per_chunk, nchunks);
+      else if (kming)
+        hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64, false, true>),
+                           grid, dim3(256), 0, cur_stream(),
+                           (const T16*)x.data_ptr(),
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
