@@ -36,7 +36,20 @@ def _run_worker(rank, fn, port, q):
             dist.destroy_process_group()
 
 
-def run_distributed(fn, port):
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def run_distributed(fn, port=None):
+    # an OS-assigned free port per call: fixed ports collided with
+    # TIME_WAIT sockets left by earlier subprocess tests (observed as a
+    # transient ConnectionError flake in full-suite runs)
+    port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
     procs = [ctx.Process(target=_run_worker, args=(r, fn, port, q))
