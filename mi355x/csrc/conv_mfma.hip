@@ -190,8 +190,13 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
 
   // incremental (c-chunk, s, r) decode — load_step is called exactly once
   // per j in order, so three runtime divisions per k-step reduce to carry
-  // counters (same lever as the wgrad kernels' m-decode)
+  // counters (same lever as the wgrad kernels' m-decode). The per-tap
+  // position (validity + image offset, incl. the TRANS stride div/mod
+  // pair) is CACHED and recomputed only when the tap advances — for a
+  // 1x1 dgrad the whole k-loop walks c-chunks of one tap.
   int cc_i = 0, s_i = 0, r_i = 0;
+  bool va_tap = m_ok;
+  long ioff_tap = 0;
   auto load_step = [&](int j) {
     if constexpr (GENC) {
       load_step_genc(j);
@@ -201,6 +206,25 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
     const int c0 = cc_i * BK;
     const int s_ = s_i;
     const int r_ = r_i;
+    if (cc_i == 0) {  // first chunk of this tap: refresh the position
+      bool va = m_ok;
+      long ioff = 0;
+      if constexpr (TRANS) {
+        const int ph = ih0 + pad - r_;
+        const int qw = iw0 + pad - s_;
+        const int pp = ph / stride, qq = qw / stride;
+        va = va && ph >= 0 && qw >= 0 && (ph % stride) == 0 &&
+             (qw % stride) == 0 && pp < Hi && qq < Wi;
+        if (va) ioff = (((long)n_ * Hi + pp) * Wi + qq) * CI;
+      } else {
+        const int ih = ih0 + r_;
+        const int iw = iw0 + s_;
+        va = va && (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
+        if (va) ioff = (((long)n_ * Hi + ih) * Wi + iw) * CI;
+      }
+      va_tap = va;
+      ioff_tap = ioff;
+    }
     if (++cc_i == cchunks) {
       cc_i = 0;
       if (++s_i == S) {
@@ -208,21 +232,8 @@ __global__ __launch_bounds__(256, 2) void conv_gather_gemm(
         ++r_i;
       }
     }
-    bool va = m_ok;
-    long ioff = 0;
-    if constexpr (TRANS) {
-      const int ph = ih0 + pad - r_;
-      const int qw = iw0 + pad - s_;
-      const int pp = ph / stride, qq = qw / stride;
-      va = va && ph >= 0 && qw >= 0 && (ph % stride) == 0 &&
-           (qw % stride) == 0 && pp < Hi && qq < Wi;
-      if (va) ioff = (((long)n_ * Hi + pp) * Wi + qq) * CI;
-    } else {
-      const int ih = ih0 + r_;
-      const int iw = iw0 + s_;
-      va = va && (unsigned)ih < (unsigned)Hi && (unsigned)iw < (unsigned)Wi;
-      if (va) ioff = (((long)n_ * Hi + ih) * Wi + iw) * CI;
-    }
+    const bool va = va_tap;
+    const long ioff = ioff_tap;
     const T16* xp = va ? in + ioff + c0 + sa_c : zpage;
 #pragma unroll
     for (int i = 0; i < 4; ++i)
