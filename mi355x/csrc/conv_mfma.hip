@@ -443,11 +443,21 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
   // incremental (c-chunk, s, r) decode: runtime divisions per k-step
   // measured as 38% issue-stall (PMC) in this short-k kernel
   int cc_i = 0, s_i = 0, r_i = 0;
+  bool va_tap = false;
+  long ioff_tap = 0, woff_tap = 0;
   auto load_step = [&](int j) {
     (void)j;
     const int c0 = cc_i * BK;
-    const int s_ = s0 + 2 * s_i;
-    const int r_ = r0 + 2 * r_i;
+    if (cc_i == 0) {  // first chunk of this tap: refresh the position
+      const int s_ = s0 + 2 * s_i;
+      const int r_ = r0 + 2 * r_i;
+      const int dh = ih_ + pad - r_;  // even by construction
+      const int dw_ = iw_ + pad - s_;
+      const int pp = dh >> 1, qq = dw_ >> 1;
+      va_tap = m_ok && dh >= 0 && dw_ >= 0 && pp < P && qq < Q;
+      ioff_tap = (((long)n_ * P + pp) * Q + qq) * KO;
+      woff_tap = ((long)r_ * S + s_) * CI * (long)KO;
+    }
     if (++cc_i == cchunks) {
       cc_i = 0;
       if (++s_i == ns) {
@@ -455,17 +465,12 @@ __global__ __launch_bounds__(256, 2) void conv_dgrad_s2_gemm(
         ++r_i;
       }
     }
-    const int dh = ih_ + pad - r_;  // even by construction
-    const int dw_ = iw_ + pad - s_;
-    const int pp = dh >> 1, qq = dw_ >> 1;
-    const bool va = m_ok && dh >= 0 && dw_ >= 0 && pp < P && qq < Q;
-    const T16* xp =
-        va ? dy + (((long)n_ * P + pp) * Q + qq) * KO + c0 + sa_c : zpage;
+    const bool va = va_tap;
+    const T16* xp = va ? dy + ioff_tap + c0 + sa_c : zpage;
 #pragma unroll
     for (int i = 0; i < 4; ++i)
       sa[i] = *reinterpret_cast<const short8*>(xp + (va ? 8 * i : 0));
-    const T16* wp = wflip + (((long)r_ * S + s_) * CI + (k0 + sb_n)) * KO +
-                    c0 + sb_c;
+    const T16* wp = wflip + woff_tap + (long)(k0 + sb_n) * KO + c0 + sb_c;
 #pragma unroll
     for (int i = 0; i < EPT / 8; ++i)
       sb[i] = *reinterpret_cast<const short8*>(wp + 8 * i);
