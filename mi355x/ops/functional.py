@@ -319,10 +319,8 @@ class _BNConvFn(torch.autograd.Function):
         x, w16, mean, invstd, gamma, asc, ash = ctx.saved_tensors
         stride, padding, wshape, m_total, pg = ctx.conf
         dy2 = dy2.contiguous()
-        ef = torch.empty(0, device=x.device)
         efx = torch.empty(0, device=x.device, dtype=x.dtype)
         # conv2 backward w.r.t. the virtual z
-        dz = None
         dw = None
         if ctx.needs_input_grad[3]:
             dw = ext().conv2d_wgrad_scaled(x, asc, ash, dy2, wshape[2],
