@@ -658,25 +658,49 @@ __global__ __launch_bounds__(256, 2) void conv_patch_gemm(
 
   for (int cc = 0; cc < cchunks; ++cc) {
     __syncthreads();  // previous c-chunk's patch reads done
-    // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi ----
+    // ---- cooperative patch stage: rows gr0..gr0+NR-1, cols -1..Wi.
+    // 4-deep load batches: the load->store pairs of the plain loop expose
+    // one L2/HBM round trip per group (the fill is this kernel's dominant
+    // serial phase — PMC: 67% parked) ----
     const int ngroups = NR * (Wi + 2) * (BK / 8);
-    for (int i = tid; i < ngroups; i += 256) {
-      const int g = i % (BK / 8);
-      const int ce = i / (BK / 8);
-      const int col = ce % (Wi + 2);
-      const int row = ce / (Wi + 2);
-      const long gr = gr0 + row;
-      const int iw = col - 1;
-      short8 v = {};
-      if (gr >= 0 && gr < (long)N * Ho && (unsigned)iw < (unsigned)Wi) {
-        const long nn = gr / Ho, pp = gr % Ho;
-        v = *reinterpret_cast<const short8*>(
-            in + ((nn * Hi + pp) * Wi + iw) * CI + cc * BK + g * 8);
-        if constexpr (SCALED)
-          v = scale8<T16>(v, sc8_load(asc, ash, cc * BK + g * 8));
+    for (int i0f = tid; i0f < ngroups; i0f += 4 * 256) {
+      const T16* srcs[4];
+      short8 vals[4];
+      int dsts[4], gs[4];
+      bool ok[4];
+#pragma unroll
+      for (int b = 0; b < 4; ++b) {
+        const int i = i0f + b * 256;
+        const int g = i % (BK / 8);
+        const int ce = i / (BK / 8);
+        const int col = ce % (Wi + 2);
+        const int row = ce / (Wi + 2);
+        const long gr = gr0 + row;
+        const int iw = col - 1;
+        gs[b] = g;
+        dsts[b] = i < ngroups ? (row * (Wi + 2) + col) * PCS + g * 8 : -1;
+        ok[b] = i < ngroups && gr >= 0 && gr < (long)N * Ho &&
+                (unsigned)iw < (unsigned)Wi;
+        const long nn = ok[b] ? gr / Ho : 0;
+        const long pp = ok[b] ? gr % Ho : 0;
+        srcs[b] = ok[b]
+                      ? in + ((nn * Hi + pp) * Wi + iw) * CI + cc * BK + g * 8
+                      : zstub;  // always-loadable: batches the 4 round trips
       }
-      *reinterpret_cast<short8*>(patch + ((long)row * (Wi + 2) + col) * PCS +
-                                 g * 8) = v;
+#pragma unroll
+      for (int b = 0; b < 4; ++b)
+        vals[b] = *reinterpret_cast<const short8*>(srcs[b]);
+#pragma unroll
+      for (int b = 0; b < 4; ++b) {
+        if (dsts[b] < 0) break;
+        short8 v = vals[b];
+        if (!ok[b]) v = short8{};
+        if constexpr (SCALED) {
+          if (ok[b])
+            v = scale8<T16>(v, sc8_load(asc, ash, cc * BK + gs[b] * 8));
+        }
+        *reinterpret_cast<short8*>(patch + dsts[b]) = v;
+      }
     }
     __syncthreads();
 
