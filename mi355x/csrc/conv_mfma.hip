@@ -815,7 +815,6 @@ template <typename T16, int KT, bool SCALED = false, bool KMIN = false>
 __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
-    const T16* __restrict__ zpage,
     const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
     const float* __restrict__ ash,
     float* __restrict__ dw,       // [KO, R*S*CI]
@@ -884,35 +883,33 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
   };
 
   // gather + pack 4 m-rows of 8 elems into registers for m-step m0
-  // (address select via the zero page: value-ternaries made hipcc branch
-  // around each load — serialized round trips)
   short8 vdy[4], vx[4];
   auto load_m = [&](long m0) {
     if (do_dy) {
-      const T16* dsrc[4];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const long m = m0 + sm + mi;
-        dsrc[mi] = m < m_end ? dy + m * KO + k0 + sk : zpage;
+        vdy[mi] = m < m_end ? *reinterpret_cast<const short8*>(
+                                  dy + m * KO + k0 + sk)
+                            : short8{};
       }
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        vdy[mi] = *reinterpret_cast<const short8*>(dsrc[mi]);
     }
     if (do_x) {
-      const T16* xsrc[4];
-      bool okx[4];
       int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const long m = m0 + smx + mi;
         const int ih = p_ * stride - pad + r_;
         const int iw = q_ * stride - pad + s_;
-        okx[mi] = m < m_end && (unsigned)ih < (unsigned)Hi &&
-                  (unsigned)iw < (unsigned)Wi;
-        xsrc[mi] = okx[mi]
-                       ? x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + skx
-                       : zpage;
+        const bool ok = m < m_end && (unsigned)ih < (unsigned)Hi &&
+                        (unsigned)iw < (unsigned)Wi;
+        vx[mi] = ok ? *reinterpret_cast<const short8*>(
+                          x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 +
+                          skx)
+                    : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[mi] = scale8<T16>(vx[mi], wsc);
+        }
         if (mi < 3 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -920,14 +917,6 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
             ++n_;
           }
         }
-      }
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        vx[mi] = *reinterpret_cast<const short8*>(xsrc[mi]);
-      if constexpr (SCALED) {
-#pragma unroll
-        for (int mi = 0; mi < 4; ++mi)
-          if (okx[mi]) vx[mi] = scale8<T16>(vx[mi], wsc);
       }
       advance(WGM);  // position for the NEXT m-step's gather
     }
@@ -1010,7 +999,6 @@ __global__ __launch_bounds__(256, SCALED ? 4 : 2) void conv_wgrad_mfma_kernel(
 template <typename T16, bool SCALED = false, bool KMIN = false>
 __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
     const T16* __restrict__ x, const T16* __restrict__ dy,
-    const T16* __restrict__ zpage,
     const float* __restrict__ asc,  // SCALED only
     const float* __restrict__ ash,
     float* __restrict__ dw,  // chunk slabs of [KO, R*S*CI]
@@ -1067,21 +1055,22 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
 
   short8 vdy[4], vx0[4];
   auto load_m = [&](long m0) {
-    const T16* dsrc[4];
-    const T16* xsrc[4];
-    bool okx[4];
     int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const long m = m0 + sm + mi;
-      dsrc[mi] = m < m_end ? dy + m * KO + k0 + sk : zpage;
+      vdy[mi] = m < m_end ? *reinterpret_cast<const short8*>(
+                                dy + m * KO + k0 + sk)
+                          : short8{};
       const int ih = p_ * stride - pad + r_;
       const int iw = q_ * stride - pad + s_;
-      okx[mi] = m < m_end && (unsigned)ih < (unsigned)Hi &&
-                (unsigned)iw < (unsigned)Wi;
-      xsrc[mi] = okx[mi]
-                     ? x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk
-                     : zpage;
+      const bool ok = m < m_end && (unsigned)ih < (unsigned)Hi &&
+                      (unsigned)iw < (unsigned)Wi;
+      const T16* xp = x + (((long)n_ * Hi + ih) * Wi + iw) * CI + c0 + sk;
+      vx0[mi] = ok ? *reinterpret_cast<const short8*>(xp) : short8{};
+      if constexpr (SCALED) {
+        if (ok) vx0[mi] = scale8<T16>(vx0[mi], wsc);
+      }
       if (mi < 3 && ++q_ == Wo) {
         q_ = 0;
         if (++p_ == Ho) {
@@ -1089,16 +1078,6 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_t128(
           ++n_;
         }
       }
-    }
-#pragma unroll
-    for (int mi = 0; mi < 4; ++mi) {
-      vdy[mi] = *reinterpret_cast<const short8*>(dsrc[mi]);
-      vx0[mi] = *reinterpret_cast<const short8*>(xsrc[mi]);
-    }
-    if constexpr (SCALED) {
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        if (okx[mi]) vx0[mi] = scale8<T16>(vx0[mi], wsc);
     }
     advance(WGM);
   };
@@ -1182,7 +1161,6 @@ template <typename T16, int KT, bool SCALED = false, bool KMIN = false>
 __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
     const T16* __restrict__ x,    // [N, Hi, Wi, CI]
     const T16* __restrict__ dy,   // [M, KO]
-    const T16* __restrict__ zpage,
     const float* __restrict__ asc,  // SCALED only: [CI] lazy-BN scale
     const float* __restrict__ ash,
     float* __restrict__ dw,       // chunk slabs of [KO, 3*3*CI]
@@ -1248,21 +1226,15 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
   int qg[6];  // q of each gathered position (for the row-boundary masks)
   auto load_m = [&](long m0) {
     if (do_dy) {
-      // address select, not a value branch: ternary-on-value made hipcc
-      // branch around each load — 4 serialized round trips per step
-      const T16* dsrc[4];
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const long m = m0 + sm + mi;
-        dsrc[mi] = m < m_end ? dy + m * KO + k0 + sk : zpage;
+        vdy[mi] = (m < m_end) ? *reinterpret_cast<const short8*>(
+                                    dy + m * KO + k0 + sk)
+                              : short8{};
       }
-#pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        vdy[mi] = *reinterpret_cast<const short8*>(dsrc[mi]);
     }
     if (do_x) {
-      const T16* xsrc[6];
-      bool okx[6];
       int n_ = dn, p_ = dp, q_ = dq;  // position m0 + smx (center start)
       {
         // i = 0 halo: column q-1 of the SAME row; when q == 0 the only
@@ -1270,20 +1242,29 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
         // zero stands in
         const int ih = p_ + r_ - 1;  // stride 1, pad 1
         qg[0] = q_ - 1;
-        okx[0] = m0 + smx - 1 >= 0 && q_ > 0 && (unsigned)ih < (unsigned)Hi;
-        xsrc[0] = okx[0] ? x + (((long)n_ * Hi + ih) * Wi + (q_ - 1)) * CI +
-                               c0 + skx
-                         : zpage;
+        const bool ok = m0 + smx - 1 >= 0 && q_ > 0 &&
+                        (unsigned)ih < (unsigned)Hi;
+        vx[0] = ok ? *reinterpret_cast<const short8*>(
+                         x + (((long)n_ * Hi + ih) * Wi + (q_ - 1)) * CI +
+                         c0 + skx)
+                   : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[0] = scale8<T16>(vx[0], wsc);
+        }
       }
 #pragma unroll
       for (int i = 1; i < 6; ++i) {
         const long m = m0 + smx - 1 + i;
         const int ih = p_ + r_ - 1;
         qg[i] = q_;
-        okx[i] = m < Mtot && (unsigned)ih < (unsigned)Hi;
-        xsrc[i] = okx[i]
-                      ? x + (((long)n_ * Hi + ih) * Wi + q_) * CI + c0 + skx
-                      : zpage;
+        const bool ok = m < Mtot && (unsigned)ih < (unsigned)Hi;
+        vx[i] = ok ? *reinterpret_cast<const short8*>(
+                         x + (((long)n_ * Hi + ih) * Wi + q_) * CI + c0 +
+                         skx)
+                   : short8{};
+        if constexpr (SCALED) {
+          if (ok) vx[i] = scale8<T16>(vx[i], wsc);
+        }
         if (i < 5 && ++q_ == Wo) {
           q_ = 0;
           if (++p_ == Ho) {
@@ -1291,14 +1272,6 @@ __global__ __launch_bounds__(256, SCALED ? 3 : 2) void conv_wgrad_mfma_s3(
             ++n_;
           }
         }
-      }
-#pragma unroll
-      for (int i = 0; i < 6; ++i)
-        vx[i] = *reinterpret_cast<const short8*>(xsrc[i]);
-      if constexpr (SCALED) {
-#pragma unroll
-        for (int i = 0; i < 6; ++i)
-          if (okx[i]) vx[i] = scale8<T16>(vx[i], wsc);
       }
       advance(WGM);
     }
@@ -1782,7 +1755,6 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
                                                     : nullptr;
   const float* ash_p = ash.defined() && ash.numel() ? ash.data_ptr<float>()
                                                     : nullptr;
-  at::Tensor zpw = conv_zero_page(x);  // OOB gather target (address select)
   const int N = x.size(0), Hi = x.size(1), Wi = x.size(2), CI = x.size(3);
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   const long M = (long)N * Ho * Wo;
@@ -1855,31 +1827,27 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
       if (KT == 128)
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 128>), grid, dim3(256),
                            0, cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
       else if (kminor && !asc_p)
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64, false, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
       else if (asc_p)
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           asc_p, ash_p,
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
       else
         hipLaunchKernelGGL((conv_wgrad_mfma_s3<T16, 64>), grid, dim3(256),
                            0, cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, m_per_chunk, nchunks);
     });
@@ -1912,8 +1880,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           asc_p, ash_p,
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1921,16 +1888,14 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16, false, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
       else
         hipLaunchKernelGGL((conv_wgrad_mfma_t128<T16>), grid, dim3(256), 0,
                            cur_stream(), (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1968,8 +1933,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           asc_p, ash_p,
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1977,8 +1941,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128, false, true>),
                            grid, dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1986,8 +1949,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 128>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -1996,8 +1958,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64, true>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           asc_p, ash_p,
+                           (const T16*)dy.data_ptr(), asc_p, ash_p,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -2005,8 +1966,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64, false, true>),
                            grid, dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
@@ -2014,8 +1974,7 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
         hipLaunchKernelGGL((conv_wgrad_mfma_kernel<T16, 64>), grid,
                            dim3(256), 0, cur_stream(),
                            (const T16*)x.data_ptr(),
-                           (const T16*)dy.data_ptr(), (const T16*)zpw.data_ptr(),
-                           nullptr, nullptr,
+                           (const T16*)dy.data_ptr(), nullptr, nullptr,
                            part.data_ptr<float>(),
                            N, Hi, Wi, CI, KO, Ho, Wo, (int)R, (int)S,
                            (int)stride, (int)pad, m_per_chunk, nchunks);
