@@ -833,9 +833,9 @@ void conv_wgrad_mfma_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
 // stem_mfma.hip — MFMA GEMM stem (7x7/s2/C=3), replaces the dot2 kernels
 bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
                                at::Tensor y, long pad, long act,
-                               at::Tensor stats);
+                               at::Tensor stats, long R, long stride);
 void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
-                                 long pad);
+                                 long pad, long R, long stride);
 
 static bool stem_gemm_on() {
   static const bool v = [] {
@@ -861,7 +861,7 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
   const bool stem7 = K <= 64 && C == 3 && R == 7 && S == 7 && stride == 2 &&
                      pad <= 3;
   if (stem7 && w.dim() == 2 && K % 64 == 0 && stem_gemm_on()) {
-    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act, at::Tensor());
+    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act, at::Tensor(), 7, 2);
     return y;
   }
   if (stem7) {
@@ -883,6 +883,13 @@ at::Tensor conv2d_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
                          (T16*)y.data_ptr(), N, H, W, K, P, Q, (int)pad,
                          wrow, (int)act, has_bias);
     });
+    return y;
+  }
+  // CIFAR stem 3x3/s1 with K % 64 == 0 (dim-2 padded weight): MFMA GEMM
+  // (stem_mfma.hip) — the dot2 LDS kernels below are the K < 64 fallback
+  if (C == 3 && R == 3 && S == 3 && stride == 1 && pad <= 1 &&
+      w.dim() == 2 && K % 64 == 0 && stem_gemm_on()) {
+    conv_fwd_stem_gemm_launch(x, w, bias, y, pad, act, at::Tensor(), 3, 1);
     return y;
   }
   // CIFAR stem 3x3/s1/p1: LDS-staged variant (conv_fwd_stem3_lds)
@@ -971,16 +978,18 @@ std::vector<at::Tensor> conv2d_fwd_stats(at::Tensor x, at::Tensor w,
   CHECK_CONTIG(w);
   const int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   const int K = w.size(0);
-  // ImageNet stem (7x7/s2/C=3, dim-2 padded weight): the strip kernel's
-  // epilogue can emit stats too
-  if (w.dim() == 2 && C == 3 && kR == 7 && kS == 7 && stride == 2 &&
-      pad <= 3 && K % 64 == 0 && stem_gemm_on()) {
-    const int P = out_dim(H, 7, stride, pad), Q = out_dim(W, 7, stride, pad);
+  // stems (C=3, dim-2 padded weight): 7x7/s2 ImageNet and 3x3/s1 CIFAR —
+  // the strip kernel's epilogue can emit stats too
+  if (w.dim() == 2 && C == 3 && K % 64 == 0 && stem_gemm_on() &&
+      ((kR == 7 && kS == 7 && stride == 2 && pad <= 3) ||
+       (kR == 3 && kS == 3 && stride == 1 && pad <= 1))) {
+    const int P = out_dim(H, kR, stride, pad);
+    const int Q = out_dim(W, kS, stride, pad);
     auto y = at::empty({N, P, Q, K}, x.options());
     auto stats = at::zeros({2, K}, x.options().dtype(at::kFloat));
     auto empty_bias = at::empty(0, x.options().dtype(at::kFloat));
-    const bool got =
-        conv_fwd_stem_gemm_launch(x, w, empty_bias, y, pad, 0, stats);
+    const bool got = conv_fwd_stem_gemm_launch(x, w, empty_bias, y, pad, 0,
+                                               stats, kR, stride);
     return {y, got ? stats : at::Tensor()};
   }
   if (w.dim() != 4 || !conv_mfma_supported(C, K)) {
@@ -1089,6 +1098,12 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
                            at::Tensor());
     return dw;
   }
+  if (K % 64 == 0 && C == 3 && R == 3 && S == 3 && stride == 1 &&
+      pad <= 1 && stem_gemm_on()) {
+    auto dw = at::empty({K, 3L, 3L, 3L}, x.options().dtype(at::kFloat));
+    conv_wgrad_stem_gemm_launch(x, dy, dw, pad, 3, 1);
+    return dw;
+  }
   if (K <= 64 && C == 3 && R == 3 && S == 3 && stride == 1 && pad == 1 &&
       P == H && Q == W) {
     const long nrows = (long)N * P;
@@ -1122,7 +1137,7 @@ at::Tensor conv2d_wgrad(at::Tensor x, at::Tensor dy, long R, long S,
     auto dw = at::empty({K, (long)C, R, S}, x.options().dtype(at::kFloat));
     if (C == 3 && S == 7 && stride == 2 && pad <= 4) {
       if (K % 64 == 0 && R == 7 && pad <= 3 && stem_gemm_on()) {
-        conv_wgrad_stem_gemm_launch(x, dy, dw, pad);
+        conv_wgrad_stem_gemm_launch(x, dy, dw, pad, 7, 2);
         return dw;
       }
       const long nrows = (long)N * P;

@@ -1,15 +1,17 @@
-// MFMA GEMM kernels for the ImageNet stem conv (7x7/stride-2/C=3 —
-// BASELINE config 4's first layer; reference conv surface
-// /root/reference/cifar_example.py:20-29 generalized to ResNet-50).
+// MFMA GEMM kernels for the C=3 stem convs — the 7x7/stride-2 ImageNet
+// stem (BASELINE config 4's first layer) and the 3x3/stride-1 CIFAR stem
+// (ResNet-18/CIFAR's first layer; reference conv surface
+// /root/reference/cifar_example.py:20-29 generalized to ResNet).
 //
 // The dot2 stem kernels (conv.hip) are VALU/LDS-bound at ~30-64 TF/s;
 // these route the same im2col GEMM through the matrix cores. The im2col
-// k-axis uses a ROW-PADDED layout: kg = r*24 + (s*3 + c), rows padded
-// 21 -> 24 so every 8-element k-group stays inside one filter row
-// (24 % 8 == 0 — a group is then 8 CONTIGUOUS elements of one input
+// k-axis uses a ROW-PADDED layout: kg = r*SROW + (s*3 + c), rows padded
+// 3R -> SROW (a multiple of 8) so every 8-element k-group stays inside
+// one filter row (a group is then 8 CONTIGUOUS elements of one input
 // image row, loadable as one 16-byte read away from the edges), total
-// padded K = 192 = 3 BK=64 chunks (utilization 147/192). The weight is
-// re-laid [KO, 192] on the host per call (27 KB).
+// padded K = SKG (R=7: 24-rows -> 192 = 3 BK=64 chunks, 147/192 used;
+// R=3: 16-rows -> 64 = 1 chunk, 27/64 used). The weight is re-laid
+// [KO, SKG] on the host per call (<= 27 KB).
 //
 // Load discipline (v2): every k-group issues ONE unconditional 16-byte
 // read from a selected address (real row or a zero page) and fixes up
@@ -29,8 +31,19 @@ typedef __attribute__((ext_vector_type(16))) float f32x16_s;
 
 namespace {
 
-constexpr int SROW = 24;   // padded elements per filter row (21 valid)
-constexpr int SKG = 192;   // padded im2col K (7*24 -> 3 chunks of 64)
+// per-filter-size geometry: R=7 (ImageNet, stride 2) and R=3 (CIFAR,
+// stride 1). SROW = 3R padded to 8; SKG = R*SROW padded to 64.
+template <int R_>
+struct StemGeo {};
+template <>
+struct StemGeo<7> {
+  static constexpr int SROW = 24, SKG = 192, VALID = 21;
+};
+template <>
+struct StemGeo<3> {
+  static constexpr int SROW = 16, SKG = 64, VALID = 9;
+};
+
 constexpr int SLDK = 64 + 8;
 constexpr int SLDM = 72;   // wgrad m-minor row length
 
@@ -60,28 +73,30 @@ struct G8 {
   unsigned char fast, row_ok;
 };
 
-template <typename T16>
+template <typename T16, int R_, int STRIDE_>
 DEV_INLINE void g8_plan(G8<T16>& g, const T16* __restrict__ x,
                         const T16* __restrict__ zpage, int n, int p, int q,
                         int kg0, int H, int W3, int pad, bool m_ok) {
+  constexpr int SROW = StemGeo<R_>::SROW;
   const int r = kg0 / SROW;
   g.jj = kg0 - r * SROW;
-  const int ih = 2 * p - pad + r;
-  g.e0 = (2 * q - pad) * 3 + g.jj;
-  g.row_ok = m_ok && r < 7 && (unsigned)ih < (unsigned)H;
+  const int ih = STRIDE_ * p - pad + r;
+  g.e0 = (STRIDE_ * q - pad) * 3 + g.jj;
+  g.row_ok = m_ok && r < R_ && (unsigned)ih < (unsigned)H;
   g.fast = g.row_ok && g.e0 >= 0 && g.e0 + 8 <= W3;
   g.base = ((long)n * H + ih) * (long)W3;
   g.src = g.fast ? x + g.base + g.e0 : zpage;
 }
 
-template <typename T16>
+template <typename T16, int R_>
 DEV_INLINE short8 g8_fix(const G8<T16>& g, short8 v,
                          const T16* __restrict__ x, int W3) {
+  constexpr int VALID = StemGeo<R_>::VALID;
   if (g.fast) {
-    if (g.jj > 21 - 8) {
+    if (g.jj > VALID - 8) {
 #pragma unroll
       for (int u = 0; u < 8; ++u)
-        if (g.jj + u >= 21) v[u] = 0;
+        if (g.jj + u >= VALID) v[u] = 0;
     }
     return v;
   }
@@ -90,7 +105,7 @@ DEV_INLINE short8 g8_fix(const G8<T16>& g, short8 v,
 #pragma unroll
     for (int u = 0; u < 8; ++u) {
       const int e = g.e0 + u;
-      if (g.jj + u < 21 && (unsigned)e < (unsigned)W3)
+      if (g.jj + u < VALID && (unsigned)e < (unsigned)W3)
         w[u] = *reinterpret_cast<const short*>(x + g.base + e);
     }
   }
@@ -100,14 +115,15 @@ DEV_INLINE short8 g8_fix(const G8<T16>& g, short8 v,
 // ---- forward ----------------------------------------------------------
 // conv_gather_gemm anatomy: 128m x 64ko block, 3 k-chunks of 64, next
 // chunk's gather rides under the MFMA phase, single-buffered LDS.
-template <typename T16>
+template <typename T16, int R_, int STRIDE_>
 __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
     const T16* __restrict__ x,    // [N, H, W, 3]
-    const T16* __restrict__ w24,  // [KO, 192] row-padded
+    const T16* __restrict__ w24,  // [KO, SKG] row-padded
     const float* __restrict__ bias, const T16* __restrict__ zpage,
     T16* __restrict__ y, const int N, const int H, const int W,
     const int KO, const int Ho, const int Wo, const int pad, const int act,
     const int has_bias) {
+  constexpr int SKG = StemGeo<R_>::SKG;
   __shared__ T16 lds[128 * SLDK + 64 * SLDK];
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -143,13 +159,13 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
     G8<T16> g[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
-      g8_plan(g[i], x, zpage, n_, p_, q_, j * 64 + sa_c + 8 * i, H, W3, pad,
-              m_ok);
+      g8_plan<T16, R_, STRIDE_>(g[i], x, zpage, n_, p_, q_,
+                                j * 64 + sa_c + 8 * i, H, W3, pad, m_ok);
 #pragma unroll
     for (int i = 0; i < 4; ++i)
       sa[i] = *reinterpret_cast<const short8*>(g[i].src);
 #pragma unroll
-    for (int i = 0; i < 4; ++i) sa[i] = g8_fix(g[i], sa[i], x, W3);
+    for (int i = 0; i < 4; ++i) sa[i] = g8_fix<T16, R_>(g[i], sa[i], x, W3);
     const T16* wp = w24 + (long)(k0 + sb_n) * SKG + j * 64 + sb_c;
     sb[0] = *reinterpret_cast<const short8*>(wp);
     sb[1] = *reinterpret_cast<const short8*>(wp + 8);
@@ -211,16 +227,19 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
 // spans DIRECTLY from the strip (strip idx 6q + jj + u — contiguous,
 // 4B-aligned) instead of v2's per-m scattered 16-byte global gathers
 // (~45 KB of uncoalesced reads per block). Weights [64,192] live in LDS.
-template <typename T16>
+template <typename T16, int R_, int STRIDE_>
 __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
     const T16* __restrict__ x,    // [N, H, W, 3]
-    const T16* __restrict__ w24,  // [KO, 192] row-padded
+    const T16* __restrict__ w24,  // [KO, SKG] row-padded
     const float* __restrict__ bias,
     float* __restrict__ stats_slab,  // null, or per-block (sum,sumsq) rows
     T16* __restrict__ y, const int N,
     const int H, const int W, const int KO, const int Ho, const int Wo,
     const int pad, const int act, const int has_bias, const int nrows,
     const int sstride) {
+  constexpr int SROW = StemGeo<R_>::SROW;
+  constexpr int SKG = StemGeo<R_>::SKG;
+  constexpr int VALID = StemGeo<R_>::VALID;
   extern __shared__ __attribute__((aligned(16))) char ssmem[];
   T16* strip = reinterpret_cast<T16*>(ssmem);  // [nrows+1][sstride]
   T16* ldsB = strip + (long)(nrows + 1) * sstride;  // [64][200]
@@ -231,11 +250,11 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
   const int n = (int)(bm0 / Mimg);
   const long bmi = bm0 - (long)n * Mimg;
   const int p0 = (int)(bmi / Wo);
-  const int ih0 = 2 * p0 - pad;  // strip row 0 = input row ih0
+  const int ih0 = STRIDE_ * p0 - pad;  // strip row 0 = input row ih0
   const int k0 = blockIdx.y * 64;
 
   // ---- strip fill: rows ih0 .. ih0+nrows-1, each 3*pad zeros | row |
-  // zeros; row nrows is the all-zero stub for r >= 7 k-pad reads ----
+  // zeros; row nrows is the all-zero stub for r >= R_ k-pad reads ----
   const int off3 = 3 * pad;
   const int nv8 = (W3 + 7) / 8;
   for (int t = tid; t < nrows * nv8; t += 256) {
@@ -284,8 +303,8 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
   const long mi = bmi + wm + li;
   const int p_lane = (int)(mi / Wo);
   const int q_lane = (int)(mi - (long)p_lane * Wo);
-  const int rr_base = 2 * (p_lane - p0);  // strip row of tap r=0
-  const int ebase = 6 * q_lane;           // strip elem of (jj=0) tap col
+  const int rr_base = STRIDE_ * (p_lane - p0);  // strip row of tap r=0
+  const int ebase = 3 * STRIDE_ * q_lane;  // strip elem of (jj=0) tap col
   f32x16_s acc[2] = {};
 
 #pragma unroll
@@ -293,7 +312,7 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
     const int kg0 = slice * 16 + kh * 8;  // wave-uniform per kh
     const int r = kg0 / SROW;
     const int jj = kg0 - r * SROW;
-    const int rr = r < 7 ? rr_base + r : nrows;  // stub row for k-pad
+    const int rr = r < R_ ? rr_base + r : nrows;  // stub row for k-pad
     const short* ap = reinterpret_cast<const short*>(
         strip + (long)rr * sstride + ebase + jj);
     short8 af;
@@ -303,12 +322,12 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
       __builtin_memcpy(&v32, ap + 2 * u2, 4);
       __builtin_memcpy(reinterpret_cast<char*>(&af) + 4 * u2, &v32, 4);
     }
-    // zero the row-pad lanes (jj+u >= 21) — weights there are zero too,
-    // but the strip span can run past the zero pads near the row end
-    if (jj > 21 - 8) {
+    // zero the row-pad lanes (jj+u >= VALID) — weights there are zero
+    // too, but the strip span can run past the zero pads at the row end
+    if (jj > VALID - 8) {
 #pragma unroll
       for (int u = 0; u < 8; ++u)
-        if (jj + u >= 21) af[u] = 0;
+        if (jj + u >= VALID) af[u] = 0;
     }
     const short8 b0 = *reinterpret_cast<const short8*>(
         ldsB + li * (SKG + 8) + slice * 16 + kh * 8);
@@ -366,19 +385,23 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
 }
 
 // ---- wgrad ------------------------------------------------------------
-// dw slab scatter target layout: [KO, 3, 7, 7] (parameter layout), one
-// slab per m-chunk; grid = (KO/64, m-chunks). ALL 192 kg in one block:
-// dy is staged once for 3x the MFMA work (the kg-chunked version staged
+// dw slab scatter target layout: [KO, 3, R, R] (parameter layout), one
+// slab per m-chunk; grid = (KO/64, m-chunks). ALL SKG kg in one block:
+// dy is staged once for NJ x the MFMA work (the kg-chunked version staged
 // the same dy per chunk and its 4-MFMA inner loop couldn't cover the
 // scattered x-gather latency).
-template <typename T16>
+template <typename T16, int R_, int STRIDE_>
 __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     const T16* __restrict__ x,   // [N, H, W, 3]
     const T16* __restrict__ dy,  // [M, KO]
     const T16* __restrict__ zpage,
-    float* __restrict__ dw,      // chunk slabs of [KO*3*7*7]
+    float* __restrict__ dw,      // chunk slabs of [KO*3*R_*R_]
     const int N, const int H, const int W, const int KO, const int Ho,
     const int Wo, const int pad, const long m_per_chunk) {
+  constexpr int SROW = StemGeo<R_>::SROW;
+  constexpr int SKG = StemGeo<R_>::SKG;
+  constexpr int VALID = StemGeo<R_>::VALID;
+  constexpr int NJ = SKG / 64;  // kg 64-chunks = MFMA j-tiles per wave
   __shared__ T16 lds[(64 + SKG) * SLDM];
   const int tid = threadIdx.x;
   const long Mtot = (long)N * Ho * Wo;
@@ -394,7 +417,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
   const int sk = (t >> 4) * 8;
   const int tx = tid & 127;
   const int smx = (tx & 15) * 4;
-  const int gx0 = (tx >> 4) * 3;  // this thread's 3 kg-groups
+  const int gx0 = (tx >> 4) * NJ;  // this thread's NJ kg-groups
 
   const int wave = tid >> 6;
   const int lane = tid & 63;
@@ -403,7 +426,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
   const int i0 = (wave & 1) * 32;       // KO sub-tile
   const int j0 = (wave >> 1) * 32;      // kg sub-tile base (stride 64)
 
-  f32x16_s acc[3] = {};
+  f32x16_s acc[NJ] = {};
 
   int dn = 0, dp = 0, dq = 0;
   {
@@ -424,7 +447,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     }
   };
 
-  short8 vdy[4], vx[3][4];
+  short8 vdy[4], vx[NJ][4];
   auto load_m = [&](long m0) {
     if (do_dy) {
 #pragma unroll
@@ -437,14 +460,14 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     }
     if (do_x) {
 #pragma unroll
-      for (int ii = 0; ii < 3; ++ii) {
+      for (int ii = 0; ii < NJ; ++ii) {
         G8<T16> g[4];
         int n_ = dn, p_ = dp, q_ = dq;
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi) {
           const long m = m0 + smx + mi;
-          g8_plan(g[mi], x, zpage, n_, p_, q_, (gx0 + ii) * 8, H, W3, pad,
-                  m < m_end);
+          g8_plan<T16, R_, STRIDE_>(g[mi], x, zpage, n_, p_, q_,
+                                    (gx0 + ii) * 8, H, W3, pad, m < m_end);
           if (mi < 3 && ++q_ == Wo) {
             q_ = 0;
             if (++p_ == Ho) {
@@ -458,7 +481,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
           vx[ii][mi] = *reinterpret_cast<const short8*>(g[mi].src);
 #pragma unroll
         for (int mi = 0; mi < 4; ++mi)
-          vx[ii][mi] = g8_fix(g[mi], vx[ii][mi], x, W3);
+          vx[ii][mi] = g8_fix<T16, R_>(g[mi], vx[ii][mi], x, W3);
       }
       advance(64);
     }
@@ -475,7 +498,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     if (do_x) {
       T16* ldsT = lds + 64 * SLDM;
 #pragma unroll
-      for (int ii = 0; ii < 3; ++ii) {
+      for (int ii = 0; ii < NJ; ++ii) {
         const int krow = (gx0 + ii) * 8;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
@@ -501,7 +524,7 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
       const short8 af = *reinterpret_cast<const short8*>(
           ldsDyT + (i0 + li) * SLDM + kk + kh * 8);
 #pragma unroll
-      for (int jj = 0; jj < 3; ++jj) {
+      for (int jj = 0; jj < NJ; ++jj) {
         const short8 bf = *reinterpret_cast<const short8*>(
             ldsXT + (j0 + jj * 64 + li) * SLDM + kk + kh * 8);
         acc[jj] = SMfma<T16>::run(af, bf, acc[jj]);
@@ -509,20 +532,20 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
     }
   }
 
-  // scatter the 32(ko) x 3x32(kg) fp32 tiles into the [KO,3,7,7] slab;
-  // row-pad positions (jj >= 21) and r >= 7 are dropped
-  float* slab = dw + (long)blockIdx.y * ((long)KO * 3 * 7 * 7);
+  // scatter the 32(ko) x NJx32(kg) fp32 tiles into the [KO,3,R,R] slab;
+  // row-pad positions (jj >= VALID) and r >= R_ are dropped
+  float* slab = dw + (long)blockIdx.y * ((long)KO * 3 * R_ * R_);
 #pragma unroll
-  for (int jt = 0; jt < 3; ++jt) {
+  for (int jt = 0; jt < NJ; ++jt) {
     const int kg = j0 + jt * 64 + li;
     const int r = kg / SROW;
     const int jj = kg - r * SROW;
-    if (jj < 21 && r < 7) {
+    if (jj < VALID && r < R_) {
       const int s = jj / 3, c = jj - s * 3;
 #pragma unroll
       for (int reg = 0; reg < 16; ++reg) {
         const int i = (reg & 3) + 8 * (reg >> 2) + 4 * kh;
-        slab[(((long)(k0 + i0 + i) * 3 + c) * 7 + r) * 7 + s] =
+        slab[(((long)(k0 + i0 + i) * 3 + c) * R_ + r) * R_ + s] =
             acc[jt][reg];
       }
     }
@@ -535,29 +558,33 @@ __global__ __launch_bounds__(256, 2) void conv_wgrad_stem_gemm(
 
 at::Tensor conv_zero_page(const at::Tensor& like);  // conv_mfma.hip
 
-// w: [KO, KGP] tap-major padded ([kg = (r*7+s)*3+c]) -> [KO, 192]
-// row-padded (kg = r*24 + s*3 + c)
-static at::Tensor stem_w24(const at::Tensor& w) {
+// w: [KO, KGP] tap-major padded ([kg = (r*R+s)*3+c]) -> [KO, SKG]
+// row-padded (kg = r*SROW + s*3 + c)
+template <int R_>
+static at::Tensor stem_wpad(const at::Tensor& w) {
+  constexpr int SROW = StemGeo<R_>::SROW;
+  constexpr int SKG = StemGeo<R_>::SKG;
   const long KO = w.size(0);
-  auto valid = w.narrow(1, 0, 147).reshape({KO, 7, 21});
-  auto padded = at::constant_pad_nd(valid, {0, 3}, 0);  // [KO,7,24]
-  return at::constant_pad_nd(padded.reshape({KO, 168}), {0, SKG - 168}, 0)
+  auto valid = w.narrow(1, 0, 3 * R_ * R_).reshape({KO, R_, 3 * R_});
+  auto padded = at::constant_pad_nd(valid, {0, SROW - 3 * R_}, 0);
+  return at::constant_pad_nd(padded.reshape({KO, R_ * SROW}),
+                             {0, SKG - R_ * SROW}, 0)
       .contiguous();
 }
 
 void stats_slab_reduce(at::Tensor slab, at::Tensor stats, int gx, int bnt2,
                        int C);  // conv_mfma.hip
 
-// returns true when `stats` ([2,KO], zeroed) was filled from the conv
-// epilogue (strip path only; the v2 fallback leaves it to the caller)
-bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+template <int R_, int STRIDE_>
+static bool fwd_stem_gemm_impl(at::Tensor x, at::Tensor w, at::Tensor bias,
                                at::Tensor y, long pad, long act,
                                at::Tensor stats) {
+  constexpr int SKG = StemGeo<R_>::SKG;
   const int N = x.size(0), H = x.size(1), W = x.size(2);
   const int KO = w.size(0);
   const int Ho = y.size(1), Wo = y.size(2);
   TORCH_CHECK(KO % 64 == 0, "stem GEMM expects KO % 64 == 0");
-  auto w24 = stem_w24(w);
+  auto w24 = stem_wpad<R_>(w);
   const long M = (long)N * Ho * Wo;
   dim3 grid((unsigned)cdiv_l(M, 128), KO / 64);
   const int has_bias = bias.numel() > 0;
@@ -570,11 +597,11 @@ bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   const long Mimg = (long)Ho * Wo;
   if (strip_on && Mimg % 128 == 0) {
     // rows spanned: 128 outputs cover <= ceil(127/Wo)+1 p-rows; input
-    // rows 2*p0-pad .. 2*p_last+6-pad
+    // rows STRIDE*p0-pad .. STRIDE*p_last+R-1-pad
     const int pspan = (int)((127 / Wo) + 1);
-    const int nrows = 2 * (pspan - 1) + 7;
+    const int nrows = STRIDE_ * (pspan - 1) + R_;
     // row length: left pads (3*pad) + payload (3W) + right slack for the
-    // widest tap span (max strip idx = 3W + 6*pad + 2), rounded to 4
+    // widest 8-element tap read (max strip idx <= 3W + 6*pad + 6)
     const int sstride = (3 * W + 6 * (int)pad + 8 + 3) & ~3;
     const size_t smem =
         ((size_t)(nrows + 1) * sstride + 64 * (SKG + 8)) * x.element_size();
@@ -588,8 +615,9 @@ bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
         slab_p = slab.data_ptr<float>();
       }
       DISPATCH_16(x, T16, {
-        hipLaunchKernelGGL((conv_fwd_stem_strip<T16>), grid, dim3(256),
-                           smem, cur_stream(), (const T16*)x.data_ptr(),
+        hipLaunchKernelGGL((conv_fwd_stem_strip<T16, R_, STRIDE_>), grid,
+                           dim3(256), smem, cur_stream(),
+                           (const T16*)x.data_ptr(),
                            (const T16*)w24.data_ptr(),
                            has_bias ? bias.data_ptr<float>() : nullptr,
                            slab_p, (T16*)y.data_ptr(), N, H, W, KO, Ho, Wo,
@@ -602,8 +630,8 @@ bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   }
   at::Tensor zp = conv_zero_page(x);
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL((conv_fwd_stem_gemm<T16>), grid, dim3(256), 0,
-                       cur_stream(), (const T16*)x.data_ptr(),
+    hipLaunchKernelGGL((conv_fwd_stem_gemm<T16, R_, STRIDE_>), grid,
+                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)w24.data_ptr(),
                        has_bias ? bias.data_ptr<float>() : nullptr,
                        (const T16*)zp.data_ptr(), (T16*)y.data_ptr(), N, H,
@@ -612,16 +640,27 @@ bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
   return false;
 }
 
+// returns true when `stats` ([2,KO], zeroed) was filled from the conv
+// epilogue (strip path only; the v2 fallback leaves it to the caller)
+bool conv_fwd_stem_gemm_launch(at::Tensor x, at::Tensor w, at::Tensor bias,
+                               at::Tensor y, long pad, long act,
+                               at::Tensor stats, long R, long stride) {
+  if (R == 7 && stride == 2)
+    return fwd_stem_gemm_impl<7, 2>(x, w, bias, y, pad, act, stats);
+  TORCH_CHECK(R == 3 && stride == 1, "stem GEMM: unsupported (R, stride)");
+  return fwd_stem_gemm_impl<3, 1>(x, w, bias, y, pad, act, stats);
+}
+
 void wgrad_reduce_launch(at::Tensor part, at::Tensor dw, long E, long nz);
 
-// dw [KO,3,7,7] fp32
-void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+template <int R_, int STRIDE_>
+static void wgrad_stem_gemm_impl(at::Tensor x, at::Tensor dy, at::Tensor dw,
                                  long pad) {
   const int N = x.size(0), H = x.size(1), W = x.size(2);
   const int Ho = dy.size(1), Wo = dy.size(2), KO = dy.size(3);
   TORCH_CHECK(KO % 64 == 0, "stem wgrad GEMM expects KO % 64 == 0");
   const long M = (long)N * Ho * Wo;
-  const long E = (long)KO * 3 * 7 * 7;
+  const long E = (long)KO * 3 * R_ * R_;
   long nchunks = std::min<long>(512, cdiv_l(M, 4096));
   const long m_per_chunk = cdiv_l(M, std::max<long>(nchunks, 1));
   nchunks = cdiv_l(M, m_per_chunk);
@@ -629,11 +668,20 @@ void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
   at::Tensor zp = conv_zero_page(x);
   dim3 grid(KO / 64, (unsigned)nchunks);
   DISPATCH_16(x, T16, {
-    hipLaunchKernelGGL((conv_wgrad_stem_gemm<T16>), grid, dim3(256), 0,
-                       cur_stream(), (const T16*)x.data_ptr(),
+    hipLaunchKernelGGL((conv_wgrad_stem_gemm<T16, R_, STRIDE_>), grid,
+                       dim3(256), 0, cur_stream(), (const T16*)x.data_ptr(),
                        (const T16*)dy.data_ptr(), (const T16*)zp.data_ptr(),
                        part.data_ptr<float>(), N, H, W, KO, Ho, Wo, (int)pad,
                        m_per_chunk);
   });
   wgrad_reduce_launch(part, dw, E, nchunks);
+}
+
+// dw [KO,3,R,R] fp32
+void conv_wgrad_stem_gemm_launch(at::Tensor x, at::Tensor dy, at::Tensor dw,
+                                 long pad, long R, long stride) {
+  if (R == 7 && stride == 2)
+    return wgrad_stem_gemm_impl<7, 2>(x, dy, dw, pad);
+  TORCH_CHECK(R == 3 && stride == 1, "stem wgrad GEMM: unsupported (R, stride)");
+  wgrad_stem_gemm_impl<3, 1>(x, dy, dw, pad);
 }

@@ -139,10 +139,13 @@ def conv_bn(conv: "Conv2d", bn: "BatchNorm2d", x, residual=None):
     if (_fuse_stats() and x.is_cuda and bn.training and conv.bias is None
             and conv.act is None
             # shapes with a stats-emitting MFMA path: C % 64 == 0 convs
-            # and the 7x7/s2 ImageNet stem (strip-kernel epilogue stats)
+            # and the C=3 stems (7x7/s2 ImageNet, 3x3/s1 CIFAR — strip
+            # kernel epilogue stats, mi355x/csrc/stem_mfma.hip)
             and (conv.weight.shape[1] % 64 == 0
                  or (conv.weight.shape[1] == 3
-                     and conv.weight.shape[2] == 7 and conv.stride == 2))):
+                     and ((conv.weight.shape[2] == 7 and conv.stride == 2)
+                          or (conv.weight.shape[2] == 3
+                              and conv.stride == 1))))):
         from mi355x.ops import functional as F_
 
         bn._nbt += 1

@@ -49,7 +49,10 @@ def _close(gpu_t, cpu_t, rtol=RTOL, atol=ATOL):
     ((2, 16, 16, 128), 256, 1, 2, 0, False, None), # t128 1x1 stride-2 (r50 ds)
     ((2, 15, 15, 128), 128, 3, 2, 1, False, None), # t128 3x3/s2/p1 (r50 conv2)
     ((3, 7, 7, 64), 192, 3, 2, 1, False, None),   # MFMA stride-2, odd M
-    ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # GENC stem (CIFAR)
+    ((2, 32, 32, 3), 64, 3, 1, 1, False, None),    # stem3 GEMM strip (CIFAR)
+    ((2, 32, 32, 3), 64, 3, 1, 1, True, "relu"),   # stem3 GEMM strip, bias+relu
+    ((2, 30, 30, 3), 64, 3, 1, 1, False, None),    # stem3 GEMM v2 fallback (900 % 128 != 0)
+    ((2, 34, 34, 3), 64, 3, 1, 0, False, None),    # stem3 GEMM pad=0 (32x32 out)
     ((2, 32, 32, 3), 64, 7, 2, 3, True, "relu"),   # stem7 GEMM (strip: Ho*Wo%128==0)
     ((2, 30, 30, 3), 64, 7, 2, 3, False, None),    # stem7 GEMM v2 fallback (15x15 out)
     ((2, 29, 29, 3), 32, 7, 2, 3, False, None),    # stem7 odd W, K<64
@@ -324,4 +327,16 @@ def test_stem_epilogue_stats_match_bn_stats():
     # the epilogue accumulates PRE-bf16-rounding fp32 values; bn_stats
     # reads the rounded tensor — a sqrt(M)-scaled rounding-noise gap
     # (observed ~0.3 abs on 4096-element channel sums), not an error
+    torch.testing.assert_close(stats, ref, rtol=3e-2, atol=1.0)
+
+
+def test_stem3_epilogue_stats_match_bn_stats():
+    """Same check for the 3x3/s1 CIFAR stem geometry of the strip kernel
+    (conv->BN fusion of ResNet-18/CIFAR's first layer)."""
+    g = torch.Generator().manual_seed(12)
+    x = torch.randn(4, 32, 32, 3, generator=g).cuda().to(torch.bfloat16)
+    w = (torch.randn(64, 3, 3, 3, generator=g) * 0.2).cuda()
+    y, stats = fn.conv2d_with_stats(x, w, stride=1, padding=1)
+    assert stats is not None and stats.numel() == 128, "stem3 stats missing"
+    ref = ops.ext().bn_stats(y.detach().contiguous())
     torch.testing.assert_close(stats, ref, rtol=3e-2, atol=1.0)
