@@ -37,6 +37,24 @@ from torch import nn
 from .flat import FlatState
 
 
+def _device_slot(device: torch.device) -> int:
+    """Globally-unique device slot for duplicate-device detection:
+    (host, ordinal) for GPUs; CPU ranks get -(rank+1), which never
+    collides. Two ranks reporting the same GPU slot (oversubscribed
+    rehearsal layouts — more ranks than GPUs) cannot build a native RCCL
+    communicator: ncclCommInitRank refuses duplicate devices, and the
+    rank-0 uid server would be left hanging."""
+    import socket
+    import zlib
+
+    if device.type == "cuda":
+        host = zlib.crc32(socket.gethostname().encode()) & 0x3FFFFF
+        ordinal = (device.index if device.index is not None
+                   else torch.cuda.current_device())
+        return host * 512 + ordinal
+    return -(dist.get_rank() + 1)
+
+
 class _TorchComm:
     def __init__(self, process_group):
         self.pg = process_group
@@ -129,9 +147,13 @@ class DistributedDataParallel(nn.Module):
         self._reset_bucket_state()
 
     def _native_feasible_everywhere(self) -> bool:
-        """All-rank MIN-reduce of local native-comm feasibility (extension
-        importable + RCCL symbols present) over the already-initialized
-        torch.distributed group, so every rank takes the same comm path."""
+        """Collective agreement on native-comm feasibility over the
+        already-initialized torch.distributed group, so every rank takes
+        the same comm path. Feasible iff on EVERY rank the extension is
+        importable with the RCCL symbols present, AND no two ranks sit on
+        the same physical GPU (all-gathered _device_slot values must be
+        unique — RCCL cannot build a communicator for an oversubscribed
+        rehearsal layout; those fall back to torch.distributed/gloo)."""
         ok = 1
         try:
             from mi355x.ops import ext
@@ -140,11 +162,16 @@ class DistributedDataParallel(nn.Module):
         except Exception:
             ok = 0
         backend = dist.get_backend(self.process_group)
-        flag = torch.tensor(
-            [ok], dtype=torch.int32,
-            device=self.flat.flat_param.device if backend == "nccl" else "cpu")
-        dist.all_reduce(flag, op=dist.ReduceOp.MIN, group=self.process_group)
-        return bool(int(flag.item()))
+        dev = (self.flat.flat_param.device if backend == "nccl" else
+               torch.device("cpu"))
+        me = torch.tensor([ok, _device_slot(self.flat.flat_param.device)],
+                          dtype=torch.int64, device=dev)
+        world = dist.get_world_size(self.process_group)
+        gathered = [torch.zeros_like(me) for _ in range(world)]
+        dist.all_gather(gathered, me, group=self.process_group)
+        ok_all = all(int(g[0]) for g in gathered)
+        slots = [int(g[1]) for g in gathered]
+        return ok_all and len(set(slots)) == len(slots)
 
     # -- init-time sync ----------------------------------------------------
     def _broadcast_initial_state(self):

@@ -174,3 +174,35 @@ def test_batched_buffer_broadcast_syncs_all_dtypes():
         torch.testing.assert_close(f, ref.stat_f, rtol=0, atol=0)
         torch.testing.assert_close(g, ref.stat_g, rtol=0, atol=0)
         assert torch.equal(c, ref.count)
+
+
+def _dup_device_fallback(rank):
+    """Feasibility agreement: unique device slots -> feasible (on CPU the
+    extension-symbol check decides); a forced duplicate slot -> all ranks
+    agree native RCCL is infeasible (the oversubscribed-rehearsal case:
+    two ranks on one GPU would hang ncclCommInitRank)."""
+    from mi355x.parallel import DistributedDataParallel, comm
+    from mi355x.parallel import ddp as ddp_mod
+
+    comm.init_process_group(backend="gloo")
+    torch.manual_seed(0)
+    net = nn.Linear(8, 10)
+    ddp = DistributedDataParallel(net)
+    unique = ddp._native_feasible_everywhere()
+
+    orig = ddp_mod._device_slot
+    ddp_mod._device_slot = lambda device: 7  # every rank reports one GPU
+    try:
+        dup = ddp._native_feasible_everywhere()
+    finally:
+        ddp_mod._device_slot = orig
+    return unique, dup
+
+
+def test_duplicate_device_blocks_native_comm():
+    results = run_distributed(_dup_device_fallback)
+    for rank, (unique, dup) in results.items():
+        # CPU ranks always present distinct slots; with the extension
+        # built its RCCL symbols exist, so the unique case is feasible
+        assert unique is True, f"rank {rank}: unique slots deemed infeasible"
+        assert dup is False, f"rank {rank}: duplicate slots not detected"
