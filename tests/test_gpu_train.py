@@ -196,3 +196,25 @@ def test_lazy_bn_grad_parity(model, size):
     (both all-consumer and gather-only modes)."""
     _grad_parity(model, size, "MI355X_LAZY_BN", "0", "1")
     _grad_parity(model, size, "MI355X_LAZY_BN", "0", "g")
+
+
+def test_ddp_world2_single_gpu_fallback():
+    """Oversubscribed layout (2 ranks sharing cuda:0 over gloo): the
+    collective feasibility agreement must detect the duplicate device and
+    fall back to torch.distributed instead of hanging in native RCCL
+    init (parallel/ddp.py::_native_feasible_everywhere; the torch-DDP
+    contract of /root/reference/cifar_example_ddp.py:83)."""
+    import os
+    import subprocess
+    import sys
+
+    worker = os.path.join(os.path.dirname(__file__), "_ddp_gpu_worker.py")
+    env = dict(os.environ, MI355X_BACKEND="gloo",
+               HSA_ENABLE_IPC_MODE_LEGACY="0")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--standalone",
+         "--local-addr", "127.0.0.1", "--nproc-per-node", "2", worker],
+        capture_output=True, text=True, timeout=300, env=env,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, (r.stdout[-1500:] + "\n" + r.stderr[-2500:])
+    assert "DDP_GPU_OK native=False" in r.stdout
