@@ -106,3 +106,33 @@ def test_serve_exactly_world_minus_one_then_closes():
     assert not t.is_alive()
     with pytest.raises(TimeoutError):
         _fetch_id(ADDR, port, timeout_s=1.0)
+
+
+def test_roundtrip_world8_burst():
+    """8-rank fan-in, the driver's SCALE-run shape: seven peers connect
+    SIMULTANEOUSLY (barrier-released) against one rank-0 listen socket
+    whose backlog must absorb the burst; every uid arrives intact."""
+    port = _free_port()
+    payload = os.urandom(UID_BYTES)
+    srv = _bind_server(ADDR, port, world=8)
+    t = threading.Thread(target=_serve_id, args=(srv, payload, 8), daemon=True)
+    t.start()
+    got = []
+    errs = []
+    release = threading.Barrier(7)
+
+    def client():
+        try:
+            release.wait(timeout=20)
+            got.append(_fetch_id(ADDR, port, timeout_s=20))
+        except Exception as e:
+            errs.append(e)
+
+    clients = [threading.Thread(target=client) for _ in range(7)]
+    for c in clients:
+        c.start()
+    for c in clients:
+        c.join(timeout=30)
+    t.join(timeout=30)
+    assert not errs, errs
+    assert got == [payload] * 7
