@@ -1,6 +1,6 @@
 """bench.py's own distributed branch, exercised exactly the way the driver
 launches it (python -m torch.distributed.run --nnodes=1 --nproc-per-node N
-... bench.py --gpus N), on CPU/gloo at world 2 and 4 — so the round-end
+... bench.py --gpus N), on CPU/gloo at world 2, 4 and 8 — so the round-end
 SCALE run's code path (rank/env plumbing, DDP construction, barrier +
 max-over-ranks timing, single JSON line from rank 0) is covered before it
 ever meets an 8-GPU node."""
@@ -31,7 +31,7 @@ def _run_bench_dist(world, port, extra=()):
     return json.loads(lines[0])
 
 
-@pytest.mark.parametrize("world,port", [(2, 29651), (4, 29655)])
+@pytest.mark.parametrize("world,port", [(2, 29651), (4, 29655), (8, 29663)])
 def test_bench_distributed_contract(world, port):
     out = _run_bench_dist(world, port)
     assert out["config"]["parallelism"] == f"dp{world}"
