@@ -221,12 +221,13 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_stem_gemm(
 
 // ---- forward v3: strip-staged -----------------------------------------
 // One block = 128 consecutive outputs of ONE image (requires Ho*Wo % 128
-// == 0 so blocks never straddle images). The 7-11 input rows the block's
+// == 0 so blocks never straddle images). The nrows input rows the block's
 // receptive fields span are staged ONCE into an LDS strip (coalesced row
 // copies with zero pads); the MFMA A-fragments then read 8-element tap
-// spans DIRECTLY from the strip (strip idx 6q + jj + u — contiguous,
-// 4B-aligned) instead of v2's per-m scattered 16-byte global gathers
-// (~45 KB of uncoalesced reads per block). Weights [64,192] live in LDS.
+// spans DIRECTLY from the strip (strip idx 3*STRIDE*q + jj + u —
+// contiguous, 4B-aligned) instead of v2's per-m scattered 16-byte global
+// gathers (~45 KB of uncoalesced reads per block for the 7x7 stem).
+// Weights [64, SKG] live in LDS.
 template <typename T16, int R_, int STRIDE_>
 __global__ __launch_bounds__(256, 2) void conv_fwd_stem_strip(
     const T16* __restrict__ x,    // [N, H, W, 3]
