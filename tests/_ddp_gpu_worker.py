@@ -5,6 +5,11 @@ fall back to torch.distributed collectives on every rank, and still
 train. Launched under torch.distributed.run --nproc-per-node 2."""
 
 import os
+import sys
+
+# launched by script path, so sys.path[0] is tests/ — make the repo root
+# (the mi355x package location) importable first
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
