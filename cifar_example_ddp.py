@@ -152,7 +152,9 @@ def main(args):
                 break
     acc = accuracy.compute()
     if args.rank == 0:
-        print(f"Accuracy on the test set: {100 * acc:.1f} %")
+        # reference print format (/root/reference/cifar_example_ddp.py:135)
+        print("Accuracy of the network on the 10000 test images: %d %%" %
+              (100 * acc))
     comm.destroy()
 
 
