@@ -9,7 +9,9 @@ same --world_size CLI flag (env wins, as in the reference), per-rank batch
 RCCL reducer), 'module.'-prefixed checkpoint keys. Conscious fixes of
 reference bugs: eval tensors are moved to the device
 (quirk 6), the checkpoint write is rank-0-gated (quirk 7), and the dead
-`dataiter.next()` call is dropped (quirk 1).
+`dataiter.next()` call is dropped (quirk 1), and a launcher-less run
+works as world-1 instead of crashing in DistributedSampler (quirk 2 —
+comm.init_process_group builds a 1-rank group).
 
 Launch:  python -m mi355x.launcher --nproc-per-node 8 cifar_example_ddp.py
    (or)  torchrun --nproc-per-node 8 cifar_example_ddp.py
