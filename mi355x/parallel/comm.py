@@ -55,8 +55,22 @@ def init_process_group(backend: str | None = None, timeout_s: int = 300):
     if backend == "nccl":
         torch.cuda.set_device(local % max(torch.cuda.device_count(), 1))
     if not dist.is_initialized():
-        dist.init_process_group(backend=backend, rank=rank, world_size=world,
-                                timeout=datetime.timedelta(seconds=timeout_s))
+        # MI355X_RDZV_FILE: filesystem-store rendezvous instead of the
+        # env:// TCPStore. Used by the multi-process CPU tests — freeing a
+        # probed port and rebinding it races with the OS handing the same
+        # ephemeral port to any other socket (observed ~1-in-10
+        # ConnectionError flakes under repeated suite runs); a file store
+        # has no port to lose.
+        store_path = os.environ.get("MI355X_RDZV_FILE")
+        if store_path:
+            store = dist.FileStore(store_path, world)
+            dist.init_process_group(backend=backend, store=store, rank=rank,
+                                    world_size=world,
+                                    timeout=datetime.timedelta(seconds=timeout_s))
+        else:
+            dist.init_process_group(backend=backend, rank=rank,
+                                    world_size=world,
+                                    timeout=datetime.timedelta(seconds=timeout_s))
     return rank, world, local
 
 

@@ -18,9 +18,12 @@ REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 def _run_bench_dist(world, port, extra=()):
     env = dict(os.environ)
     env.update({"MI355X_BENCH_MODEL": "net"})
-    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-           "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
-           "--master-port", str(port), "bench.py", "--gpus", str(world),
+    # --standalone: the rendezvous port is OS-assigned at bind time — a
+    # fixed --master-port left TIME_WAIT sockets that collided across
+    # back-to-back suite runs (the `port` arg is kept but unused)
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--standalone",
+           "--local-addr", "127.0.0.1",
+           "--nproc-per-node", str(world), "bench.py", "--gpus", str(world),
            "--steps", "2", "--warmup", "1", "--batch", "8", "--model", "net",
            *extra]
     r = subprocess.run(cmd, cwd=REPO, env=env, capture_output=True,

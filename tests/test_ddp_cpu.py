@@ -18,11 +18,14 @@ from mi355x.parallel.flat import FlatState
 WORLD = 2
 
 
-def _run_worker(rank, fn, port, q):
+def _run_worker(rank, fn, store_path, q):
     os.environ.update({
         "RANK": str(rank), "LOCAL_RANK": str(rank),
         "WORLD_SIZE": str(WORLD), "MASTER_ADDR": "127.0.0.1",
-        "MASTER_PORT": str(port),
+        # file-store rendezvous (comm.init_process_group): no TCP port to
+        # race over — probing a free port and rebinding it flaked ~1/10
+        # suite runs when the OS re-issued the port in between
+        "MI355X_RDZV_FILE": store_path,
     })
     try:
         res = fn(rank)
@@ -36,23 +39,16 @@ def _run_worker(rank, fn, port, q):
             dist.destroy_process_group()
 
 
-def _free_port():
-    import socket
-    s = socket.socket()
-    s.bind(("127.0.0.1", 0))
-    port = s.getsockname()[1]
-    s.close()
-    return port
-
-
 def run_distributed(fn, port=None):
-    # an OS-assigned free port per call: fixed ports collided with
-    # TIME_WAIT sockets left by earlier subprocess tests (observed as a
-    # transient ConnectionError flake in full-suite runs)
-    port = _free_port()
+    # a fresh FileStore path per call (the `port` arg is kept for caller
+    # compatibility but unused — TCP rendezvous raced on ephemeral ports)
+    import tempfile
+
+    d = tempfile.mkdtemp(prefix="mi355x_rdzv_")
+    store_path = os.path.join(d, "store")
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_run_worker, args=(r, fn, port, q))
+    procs = [ctx.Process(target=_run_worker, args=(r, fn, store_path, q))
              for r in range(WORLD)]
     for p in procs:
         p.start()
@@ -63,6 +59,9 @@ def run_distributed(fn, port=None):
         results[rank] = res
     for p in procs:
         p.join(timeout=60)
+    import shutil
+
+    shutil.rmtree(d, ignore_errors=True)
     return results
 
 
