@@ -29,7 +29,14 @@ def _run_worker(rank, fn, store_path, q):
     })
     try:
         res = fn(rank)
-        q.put((rank, "ok", res))
+        # serialize to BYTES: putting live tensors on a SimpleQueue shares
+        # them by fd-passing over an AF_UNIX socket, and a worker that
+        # exits before the parent has received the fds resets the socket
+        # (observed as a ~1-in-10 ConnectionResetError in parent q.get())
+        import io
+        buf = io.BytesIO()
+        torch.save(res, buf)
+        q.put((rank, "ok", buf.getvalue()))
     except Exception as e:  # surface tracebacks
         import traceback
         q.put((rank, "err", traceback.format_exc() + str(e)))
@@ -56,7 +63,8 @@ def run_distributed(fn, port=None):
     for _ in range(WORLD):
         rank, status, res = q.get()
         assert status == "ok", f"rank {rank} failed:\n{res}"
-        results[rank] = res
+        import io
+        results[rank] = torch.load(io.BytesIO(res), weights_only=False)
     for p in procs:
         p.join(timeout=60)
     import shutil
